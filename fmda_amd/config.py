@@ -1,0 +1,123 @@
+"""Configuration layer.
+
+Replaces the reference's module-global config (reference config.py:6-65) with
+dataclasses + the same knob set: order-book depth, indicator event list,
+moving-average periods, Bollinger / stochastic settings, and the
+(synthetic-time) session parameters that stood behind the Kafka topics.
+
+API tokens, Kafka brokers and MySQL credentials have no equivalent here:
+transport is an in-process queue and storage is an in-memory feature table.
+"""
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+
+# Number of order book price levels to include (reference config.py:36-37).
+BID_LEVELS = 7
+ASK_LEVELS = 7
+
+# Moving-average periods (reference config.py:40-42).
+VOLUME_MA_PERIODS = [6, 20]
+PRICE_MA_PERIODS = [20]
+DELTA_MA_PERIODS = [12]
+
+# Bollinger Bands (reference config.py:45-46).
+BOLLINGER_BANDS_PERIOD = 20
+BOLLINGER_BANDS_STD = 2
+
+# Stochastic oscillator on/off (reference config.py:49).
+STOCHASTIC_OSCILLATOR = True
+
+# Economic-indicator event list (reference config.py:52-54).
+EVENT_LIST = [
+    'Crude Oil Inventories', 'ISM Non-Manufacturing PMI',
+    'ISM Non-Manufacturing Employment', 'Services PMI',
+    'ADP Nonfarm Employment Change', 'Core CPI', 'Fed Interest Rate Decision',
+    'Building Permits', 'Core Retail Sales', 'Retail Sales',
+    'JOLTs Job Openings', 'Nonfarm Payrolls', 'Unemployment Rate',
+]
+EVENT_LIST_REPL = [e.replace(" ", "_").replace("-", "_") for e in EVENT_LIST]
+EVENT_VALUES = ["Actual", "Prev_actual_diff", "Forc_actual_diff"]
+
+# ATR factors for the 4-label target rule (reference create_database.py:176-177).
+TARGET_N1 = 1.5
+TARGET_N2 = 3.0
+TARGET_LEAD_1 = 8    # bars ahead for up1/down1 (create_database.py:186)
+TARGET_LEAD_2 = 15   # bars ahead for up2/down2 (create_database.py:187)
+
+# In-process topic names mirroring the reference Kafka topics (config.py:15).
+TOPICS = ['vix', 'volume', 'cot', 'ind', 'deep', 'predict_timestamp', 'prediction']
+
+
+def empty_indicator_message() -> dict:
+    """Indicator message template (reference config.py:58-65)."""
+    msg = {"Timestamp": 0}
+    for event in EVENT_LIST_REPL:
+        msg[event] = {v: 0 for v in EVENT_VALUES}
+    return msg
+
+
+@dataclass
+class ModelConfig:
+    """Hyperparameters of the biGRU model (reference notebook cell 29 /
+    predict.py:71-81 defaults)."""
+    hidden_size: int = 32
+    n_features: int = 108
+    output_size: int = 4
+    n_layers: int = 1
+    clip: float = 50.0
+    dropout: float = 0.5
+    spatial_dropout: bool = False
+    bidirectional: bool = True
+
+
+@dataclass
+class DataConfig:
+    """Synthetic data-pipeline parameters (chunk/window semantics mirror
+    reference sql_pytorch_dataloader.py)."""
+    n_rows: int = 3980           # reference dataset size (notebook cell 14)
+    chunk_size: int = 100
+    window: int = 30
+    val_size: float = 0.1
+    test_size: float = 0.1
+    seed: int = 1234
+    n_features: int = 108        # truncate/extend the canonical 108 columns
+
+
+@dataclass
+class TrainConfig:
+    batch_size: int = 2
+    lr: float = 1e-3
+    epochs: int = 25
+    device: str = "cpu"
+    dtype: str = "fp32"          # "fp32" | "bf16" (GPU compute dtype)
+
+
+@dataclass
+class BenchConfig:
+    """One of the BASELINE.json benchmark configurations."""
+    name: str = "repo"
+    hidden_size: int = 128
+    n_layers: int = 2
+    seq_len: int = 120
+    n_features: int = 96
+    batch_per_gpu: int = 4096
+    dtype: str = "bf16"
+
+
+BENCH_CONFIGS = {
+    # BASELINE.json configs[0]: CPU plumbing check
+    "cpu": BenchConfig(name="cpu", hidden_size=32, n_layers=1, seq_len=60,
+                       n_features=64, batch_per_gpu=8, dtype="fp32"),
+    # configs[1]/[2]: repo config, 1 and 8 GPU
+    "repo": BenchConfig(name="repo", hidden_size=128, n_layers=2, seq_len=120,
+                        n_features=96, batch_per_gpu=4096, dtype="bf16"),
+    # configs[3]: stress
+    "stress": BenchConfig(name="stress", hidden_size=512, n_layers=4,
+                          seq_len=512, n_features=96, batch_per_gpu=256,
+                          dtype="bf16"),
+    # configs[4]: streaming inference
+    "predict": BenchConfig(name="predict", hidden_size=128, n_layers=2,
+                           seq_len=120, n_features=96, batch_per_gpu=1,
+                           dtype="bf16"),
+}

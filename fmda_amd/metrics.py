@@ -1,0 +1,50 @@
+"""Quality-metric formulas identical to the reference.
+
+The reference computes sklearn subset accuracy, Hamming loss and fbeta
+(beta=0.5 in train/eval loops — biGRU_model.py:215-222, 274-281). These
+implementations reproduce the sklearn formulas exactly for binary multilabel
+indicator input, with a torch fast path that can run on-GPU (no host sync in
+the training hot loop); `test_metrics.py` validates them against sklearn.
+"""
+from typing import Tuple
+
+import torch
+
+
+def subset_accuracy(target: torch.Tensor, pred: torch.Tensor) -> torch.Tensor:
+    """sklearn.metrics.accuracy_score on multilabel data: exact-match ratio."""
+    t = target.to(torch.bool)
+    p = pred.to(torch.bool)
+    return (t == p).all(dim=1).float().mean()
+
+
+def hamming(target: torch.Tensor, pred: torch.Tensor) -> torch.Tensor:
+    """sklearn.metrics.hamming_loss: fraction of wrong labels."""
+    t = target.to(torch.bool)
+    p = pred.to(torch.bool)
+    return (t != p).float().mean()
+
+
+def fbeta_per_class(target: torch.Tensor, pred: torch.Tensor,
+                    beta: float = 0.5) -> torch.Tensor:
+    """sklearn.metrics.fbeta_score(average=None) on binary indicator columns.
+
+    sklearn returns 0 for a class where precision and recall are both
+    undefined or the denominator is 0 (zero_division default).
+    """
+    t = target.to(torch.float32)
+    p = pred.to(torch.float32)
+    tp = (t * p).sum(dim=0)
+    fp = ((1 - t) * p).sum(dim=0)
+    fn = (t * (1 - p)).sum(dim=0)
+    b2 = beta * beta
+    denom = (1 + b2) * tp + b2 * fn + fp
+    fbeta = torch.where(denom > 0, (1 + b2) * tp / torch.clamp(denom, min=1e-38),
+                        torch.zeros_like(denom))
+    return fbeta
+
+
+def batch_metrics(target: torch.Tensor, pred: torch.Tensor,
+                  beta: float = 0.5) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    return (subset_accuracy(target, pred), hamming(target, pred),
+            fbeta_per_class(target, pred, beta))

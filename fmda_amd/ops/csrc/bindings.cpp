@@ -1,0 +1,104 @@
+// Torch bindings for the fmda_amd HIP kernels (MI355X / gfx950).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <vector>
+
+extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
+                                   const void* w, const float* bhh, void* out,
+                                   float* hlast, int B, int Tseq, int n_dir,
+                                   hipStream_t stream);
+extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
+                                   const void* w, const float* bhh,
+                                   const void* out, const void* dout,
+                                   const float* dhT, void* dgi, void* dgh,
+                                   float* dh0, int B, int Tseq, int n_dir,
+                                   hipStream_t stream);
+extern "C" int fmda_mfma_selftest_launch(const void* A, const void* Bm,
+                                         float* C, hipStream_t stream);
+
+namespace {
+
+void check_common(const torch::Tensor& gi, const torch::Tensor& w,
+                  const torch::Tensor& bhh, int& B, int& T, int& n_dir,
+                  int& Hp, bool& is_bf16) {
+    TORCH_CHECK(gi.is_cuda() && w.is_cuda() && bhh.is_cuda(),
+                "fmda gru: tensors must be on GPU");
+    TORCH_CHECK(gi.is_contiguous() && w.is_contiguous() && bhh.is_contiguous(),
+                "fmda gru: tensors must be contiguous");
+    TORCH_CHECK(gi.dim() == 3 && w.dim() == 3, "fmda gru: bad ranks");
+    TORCH_CHECK(w.scalar_type() == gi.scalar_type(), "fmda gru: dtype mismatch");
+    TORCH_CHECK(bhh.scalar_type() == torch::kFloat32, "bhh must be fp32");
+    n_dir = w.size(0);
+    Hp = w.size(2);
+    TORCH_CHECK(w.size(1) == 3 * Hp, "w must be (n_dir, 3H, H)");
+    B = gi.size(0);
+    T = gi.size(1);
+    TORCH_CHECK(gi.size(2) == (int64_t)n_dir * 3 * Hp, "gi last dim mismatch");
+    TORCH_CHECK(n_dir == 1 || n_dir == 2, "n_dir must be 1 or 2");
+    is_bf16 = gi.scalar_type() == torch::kBFloat16;
+    TORCH_CHECK(is_bf16 || gi.scalar_type() == torch::kFloat32,
+                "fmda gru: dtype must be bf16 or fp32");
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> gru_fwd(torch::Tensor gi, torch::Tensor w,
+                                   torch::Tensor bhh) {
+    int B, T, n_dir, Hp;
+    bool is_bf16;
+    check_common(gi, w, bhh, B, T, n_dir, Hp, is_bf16);
+    auto out = torch::empty({B, T, (int64_t)n_dir * Hp}, gi.options());
+    auto hlast = torch::empty({n_dir, B, Hp},
+                              gi.options().dtype(torch::kFloat32));
+    auto stream = at::hip::getCurrentHIPStream();
+    int rc = fmda_gru_fwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
+                                 w.data_ptr(), bhh.data_ptr<float>(),
+                                 out.data_ptr(), hlast.data_ptr<float>(), B, T,
+                                 n_dir, stream.stream());
+    TORCH_CHECK(rc == 0, "fmda gru_fwd launch failed rc=", rc, " Hp=", Hp);
+    return {out, hlast};
+}
+
+std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
+                                   torch::Tensor bhh, torch::Tensor out,
+                                   torch::Tensor dout, torch::Tensor dhT) {
+    int B, T, n_dir, Hp;
+    bool is_bf16;
+    check_common(gi, w, bhh, B, T, n_dir, Hp, is_bf16);
+    TORCH_CHECK(out.is_contiguous() && dout.is_contiguous() &&
+                dhT.is_contiguous(), "fmda gru_bwd: tensors must be contiguous");
+    TORCH_CHECK(dhT.scalar_type() == torch::kFloat32, "dhT must be fp32");
+    auto dgi = torch::empty_like(gi);
+    auto dgh = torch::empty_like(gi);
+    auto dh0 = torch::empty({n_dir, B, Hp},
+                            gi.options().dtype(torch::kFloat32));
+    auto stream = at::hip::getCurrentHIPStream();
+    int rc = fmda_gru_bwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
+                                 w.data_ptr(), bhh.data_ptr<float>(),
+                                 out.data_ptr(), dout.data_ptr(),
+                                 dhT.data_ptr<float>(), dgi.data_ptr(),
+                                 dgh.data_ptr(), dh0.data_ptr<float>(), B, T,
+                                 n_dir, stream.stream());
+    TORCH_CHECK(rc == 0, "fmda gru_bwd launch failed rc=", rc, " Hp=", Hp);
+    return {dgi, dgh, dh0};
+}
+
+torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor Bm) {
+    TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16);
+    TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}) &&
+                Bm.sizes() == torch::IntArrayRef({32, 16}));
+    auto C = torch::empty({16, 16}, A.options().dtype(torch::kFloat32));
+    auto stream = at::hip::getCurrentHIPStream();
+    int rc = fmda_mfma_selftest_launch(A.contiguous().data_ptr(),
+                                       Bm.contiguous().data_ptr(),
+                                       C.data_ptr<float>(), stream.stream());
+    TORCH_CHECK(rc == 0, "mfma selftest launch failed");
+    return C;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("gru_fwd", &gru_fwd, "fused biGRU recurrence forward (HIP/CDNA4)");
+    m.def("gru_bwd", &gru_bwd, "fused biGRU recurrence backward (HIP/CDNA4)");
+    m.def("mfma_selftest", &mfma_selftest, "mfma fragment layout self-test");
+}

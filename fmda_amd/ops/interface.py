@@ -1,0 +1,161 @@
+"""Autograd integration of the HIP biGRU recurrence.
+
+Composition (MI355X-first): the time-batched input projections
+gi = x @ W_ih^T + b_ih for both directions run as ONE rocBLAS MFMA GEMM in
+torch (plain library GEMM); the sequential recurrence runs in the
+hand-written persistent HIP kernel (`csrc/gru_kernels.hip`); dW_hh/db_hh are
+reduced from the kernel's dGh output by one more plain GEMM. Everything
+stays inside torch autograd, so dropout, pooling, head, loss and DDP
+compose naturally.
+
+Replaces the reference's `self.gru(input_seq)` cuDNN RNN call site
+(biGRU_model.py:102).
+"""
+from typing import List, Optional, Tuple
+
+import torch
+
+from . import load_extension
+
+_ALLOWED_HP = [16, 32, 64, 128, 256, 512]
+
+
+def _pad_h(H: int) -> int:
+    for hp in _ALLOWED_HP:
+        if H <= hp:
+            return hp
+    raise ValueError(f"hidden size {H} > {_ALLOWED_HP[-1]} not supported")
+
+
+def _pad_gate_rows(w: torch.Tensor, H: int, Hp: int) -> torch.Tensor:
+    """(3H, ...) -> (3Hp, ...): zero-pad each of the r/z/n row blocks."""
+    if H == Hp:
+        return w
+    shape = list(w.shape)
+    shape[0] = 3 * Hp
+    out = w.new_zeros(shape)
+    for g in range(3):
+        out[g * Hp:g * Hp + H] = w[g * H:(g + 1) * H]
+    return out
+
+
+def _pad_cols(w: torch.Tensor, H: int, Hp: int) -> torch.Tensor:
+    """(..., H) -> (..., Hp) zero-pad."""
+    if H == Hp:
+        return w
+    shape = list(w.shape)
+    shape[-1] = Hp
+    out = w.new_zeros(shape)
+    out[..., :H] = w
+    return out
+
+
+class _GRURecurrence(torch.autograd.Function):
+    """Custom op: (gi, w_hh, b_hh) -> (out, h_last) for 1 or 2 directions.
+
+    gi:  (B, T, n_dir*3Hp) input projections incl. b_ih
+    w:   (n_dir, 3Hp, Hp) recurrent weights
+    bhh: (n_dir, 3Hp) fp32
+    out: (B, T, n_dir*Hp) hidden states (direction-concat layout)
+    h_last: (n_dir, B, Hp) fp32
+    """
+
+    @staticmethod
+    def forward(ctx, gi, w, bhh):
+        ext = load_extension()
+        gi = gi.contiguous()
+        w = w.contiguous()
+        bhh32 = bhh.to(torch.float32).contiguous()
+        out, h_last = ext.gru_fwd(gi, w, bhh32)
+        ctx.save_for_backward(gi, w, bhh32, out)
+        return out, h_last
+
+    @staticmethod
+    def backward(ctx, d_out, d_hlast):
+        ext = load_extension()
+        gi, w, bhh32, out = ctx.saved_tensors
+        B, T, _ = gi.shape
+        n_dir, threeHp, Hp = w.shape
+        d_out = d_out.contiguous().to(gi.dtype)
+        d_hlast = d_hlast.contiguous().to(torch.float32)
+        dgi, dgh, _dh0 = ext.gru_bwd(gi, w, bhh32, out, d_out, d_hlast)
+
+        # dW_hh[n, k] = sum_{b,t} dGh[b,t,n] * h_prev[b,t,k]  (plain GEMM).
+        # h_prev is the stored bf16 out, shifted by one step in each
+        # direction's processing order (dir 0 forward, dir 1 reversed).
+        dw = torch.empty_like(w, dtype=torch.float32)
+        dbhh = torch.empty_like(bhh32)
+        for d in range(n_dir):
+            dgh_d = dgh[:, :, d * threeHp:(d + 1) * threeHp]
+            out_d = out[:, :, d * Hp:(d + 1) * Hp]
+            h_prev = torch.zeros_like(out_d)
+            if d == 0:
+                h_prev[:, 1:] = out_d[:, :-1]
+            else:
+                h_prev[:, :-1] = out_d[:, 1:]
+            dw[d] = torch.matmul(
+                dgh_d.reshape(B * T, threeHp).t().float(),
+                h_prev.reshape(B * T, Hp).float())
+            dbhh[d] = dgh_d.float().sum(dim=(0, 1))
+        return dgi, dw.to(w.dtype), dbhh
+
+
+def gru_directions(gi: torch.Tensor, w: torch.Tensor, bhh: torch.Tensor
+                   ) -> Tuple[torch.Tensor, torch.Tensor]:
+    return _GRURecurrence.apply(gi, w, bhh)
+
+
+def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
+                bidirectional: bool, dropout_p: float, training: bool,
+                hidden: Optional[torch.Tensor] = None
+                ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """nn.GRU(batch_first=True)-equivalent stacked biGRU on the HIP engine.
+
+    Weights are read from the nn.GRU parameter container (reference
+    state_dict layout) each call; masters stay fp32 and are cast to the
+    compute dtype (x.dtype) so bf16 training keeps fp32 master weights.
+    """
+    assert hidden is None, "initial hidden state not supported on GPU path"
+    D = 2 if bidirectional else 1
+    H = gru_module.hidden_size
+    Hp = _pad_h(H)
+    dtype = x.dtype
+    B, T, _ = x.shape
+
+    h_n_parts: List[torch.Tensor] = []
+    inp = x
+    for layer in range(n_layers):
+        w_ihs, w_hhs, b_ihs, b_hhs = [], [], [], []
+        for d in range(D):
+            sfx = f"l{layer}" + ("_reverse" if d == 1 else "")
+            w_ihs.append(_pad_gate_rows(getattr(gru_module, f"weight_ih_{sfx}"), H, Hp))
+            w_hh = getattr(gru_module, f"weight_hh_{sfx}")
+            w_hhs.append(_pad_cols(_pad_gate_rows(w_hh, H, Hp), H, Hp))
+            b_ihs.append(_pad_gate_rows(getattr(gru_module, f"bias_ih_{sfx}"), H, Hp))
+            b_hhs.append(_pad_gate_rows(getattr(gru_module, f"bias_hh_{sfx}"), H, Hp))
+
+        w_ih_cat = torch.cat(w_ihs, dim=0).to(dtype)      # (D*3Hp, F_in)
+        b_ih_cat = torch.cat(b_ihs, dim=0).to(dtype)      # (D*3Hp,)
+        w_hh_cat = torch.stack(w_hhs, dim=0).to(dtype)    # (D, 3Hp, Hp)
+        b_hh_cat = torch.stack(b_hhs, dim=0).float()      # (D, 3Hp)
+
+        # One MFMA GEMM for all timesteps and both directions.
+        gi = torch.addmm(b_ih_cat, inp.reshape(B * T, -1), w_ih_cat.t())
+        gi = gi.view(B, T, D * 3 * Hp)
+
+        out_pad, h_last = gru_directions(gi, w_hh_cat, b_hh_cat)
+
+        if Hp == H:
+            out = out_pad
+        elif D == 2:
+            out = torch.cat([out_pad[..., :H], out_pad[..., Hp:Hp + H]], dim=-1)
+        else:
+            out = out_pad[..., :H]
+        h_n_parts.append(h_last[:, :, :H].to(dtype))
+
+        inp = out
+        if training and dropout_p > 0 and layer < n_layers - 1:
+            inp = torch.nn.functional.dropout(inp, p=dropout_p, training=True)
+
+    h_n = torch.cat(h_n_parts, dim=0)  # (L*D, B, H)
+    return inp, h_n
