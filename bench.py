@@ -1,0 +1,187 @@
+#!/usr/bin/env python3
+"""fmda_amd benchmark driver.
+
+Measures the headline metric of BASELINE.json: train seq/sec for the biGRU
+on synthetic order-book-shaped data (random-init weights), on N GPUs of one
+node (weak scaling: per-GPU batch fixed).
+
+    python bench.py --gpus N --steps K --warmup W [--config repo]
+
+Launched by torchrun for N > 1 (reads RANK/LOCAL_RANK/WORLD_SIZE/MASTER_*).
+Rank 0 prints exactly one JSON line.
+
+The timed step is the full reference training step (biGRU_model.py:162-210):
+dropout -> biGRU forward -> 3-way pooled head -> BCEWithLogitsLoss(weight,
+pos_weight) -> backward -> gradient all-reduce (N > 1) -> clip_grad_norm(50)
+-> Adam step. Nothing is skipped or cached inside the timed region.
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.nn as nn
+
+from fmda_amd.config import BENCH_CONFIGS
+from fmda_amd.data.generator import synthetic_batch
+from fmda_amd.metrics import subset_accuracy
+from fmda_amd.models import BiGRU
+
+
+def three_class_accuracy(target: torch.Tensor, pred: torch.Tensor) -> float:
+    """Derived up/down/stall accuracy (SURVEY.md section 6 labeling note)."""
+    def to3(lbl):
+        up = lbl[:, 0:2].any(dim=1)
+        dn = lbl[:, 2:4].any(dim=1)
+        cls = torch.zeros(lbl.shape[0], dtype=torch.long, device=lbl.device)
+        cls[up & ~dn] = 1
+        cls[dn & ~up] = 2
+        return cls
+    return float((to3(target.bool()) == to3(pred.bool())).float().mean())
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--config", default=None,
+                    help="repo|cpu|stress|predict (default: repo on GPU, cpu otherwise)")
+    ap.add_argument("--batch", type=int, default=None,
+                    help="override per-GPU batch")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_cuda = torch.cuda.is_available()
+
+    cfg_name = args.config or ("repo" if use_cuda else "cpu")
+    cfg = BENCH_CONFIGS[cfg_name]
+    batch = args.batch or cfg.batch_per_gpu
+    dtype = torch.bfloat16 if (cfg.dtype == "bf16" and use_cuda) else torch.float32
+
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        from fmda_amd.parallel import GradAllReduce, init_distributed
+        init_distributed()
+        dist = tdist
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        device = torch.device(f"cuda:{local_rank}")
+    else:
+        device = torch.device("cpu")
+        if args.config is None:
+            batch = min(batch, 8)
+
+    torch.manual_seed(1234)  # same init on every rank
+    model = BiGRU(cfg.hidden_size, cfg.n_features, 4, n_layers=cfg.n_layers,
+                  clip=50, dropout=0.5, spatial_dropout=False,
+                  bidirectional=True).to(device)
+
+    engine = None
+    if world > 1:
+        from fmda_amd.parallel import GradAllReduce
+        engine = GradAllReduce(model)
+
+    # Class weighting like the training notebook (cell 16): weight =
+    # total/positives, pos_weight = negatives/positives, from the synthetic
+    # class rates (~reference rates).
+    rates = torch.tensor([0.24, 0.16, 0.25, 0.16], device=device)
+    weight = 1.0 / rates
+    pos_weight = (1.0 - rates) / rates
+    loss_fn = nn.BCEWithLogitsLoss(weight=weight, pos_weight=pos_weight)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+
+    # Pre-generate a small rotating pool of synthetic batches per rank.
+    pool = []
+    for i in range(2):
+        x, y = synthetic_batch(batch, cfg.seq_len, cfg.n_features,
+                               seed=1000 + 7 * rank + i)
+        pool.append((x.to(device=device, dtype=dtype), y.to(device)))
+
+    acc_sum = torch.zeros((), device=device)
+    acc3_sum = 0.0
+    n_acc = 0
+
+    def step(i, measure_quality=False):
+        nonlocal acc3_sum, n_acc
+        x, y = pool[i % len(pool)]
+        opt.zero_grad(set_to_none=True)
+        logits = model(x)
+        loss = loss_fn(logits.float(), y)
+        loss.backward()
+        if engine is not None:
+            engine.finalize()
+        nn.utils.clip_grad_norm_(model.parameters(), 50.0)
+        opt.step()
+        pred = torch.sigmoid(logits.detach().float()) > 0.5
+        acc_sum.add_(subset_accuracy(y, pred))  # stays on device, no sync
+        if measure_quality:
+            acc3_sum += three_class_accuracy(y, pred)
+        n_acc += 1
+
+    model.train()
+    for i in range(args.warmup):
+        step(i)
+
+    if dist is not None:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    if dist is not None:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+
+    dt = torch.tensor([t1 - t0], dtype=torch.float64,
+                      device=device if dist is not None and use_cuda else "cpu")
+    if dist is not None:
+        dist.all_reduce(dt, op=dist.ReduceOp.MAX)
+    dt_max = float(dt.item())
+
+    # quality snapshot (outside the timed region)
+    step(0, measure_quality=True)
+    acc = float(acc_sum.item()) / max(n_acc, 1)
+
+    if rank == 0:
+        n_gpus = world if use_cuda else 1
+        seq_per_sec = n_gpus * batch * args.steps / dt_max
+        result = {
+            "metric": "train seq/sec, biGRU on synthetic order-book",
+            "value": seq_per_sec,
+            "unit": "seq/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": dt_max / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"biGRU L{cfg.n_layers} H{cfg.hidden_size} bidirectional",
+                "global_batch": n_gpus * batch,
+                "seq_len": cfg.seq_len,
+                "n_features": cfg.n_features,
+                "parallelism": f"dp{n_gpus}",
+                "bench_config": cfg_name,
+                "train_subset_acc": acc,
+                "train_3class_acc": acc3_sum / max(1, 1),
+            },
+        }
+        print(json.dumps(result))
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,1 @@
+from .ddp import GradAllReduce, init_distributed  # noqa: F401

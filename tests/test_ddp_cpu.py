@@ -1,0 +1,76 @@
+"""Data-parallel gradient equivalence on CPU (gloo, world_size=2).
+
+All-reduced per-rank half-batch gradients must equal single-process
+full-batch gradients (the correctness contract of fmda_amd.parallel;
+SURVEY.md section 4 test (e) run at CPU scale).
+"""
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+import torch.nn as nn
+
+
+def _build_model(seed=0):
+    torch.manual_seed(seed)
+    from fmda_amd.models import BiGRU
+    return BiGRU(16, 12, 4, n_layers=1, spatial_dropout=False, dropout=0.0)
+
+
+def _full_batch():
+    g = torch.Generator().manual_seed(42)
+    x = torch.randn(8, 10, 12, generator=g)
+    y = (torch.rand(8, 4, generator=g) < 0.3).float()
+    return x, y
+
+
+def _worker(rank, world, port, out_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from fmda_amd.parallel import GradAllReduce
+
+    model = _build_model()
+    engine = GradAllReduce(model, bucket_bytes=4096)  # force several buckets
+    x, y = _full_batch()
+    shard = slice(rank * 4, (rank + 1) * 4)
+    model.train()
+    logits = model(x[shard])
+    loss = nn.functional.binary_cross_entropy_with_logits(logits, y[shard])
+    loss.backward()
+    engine.finalize()
+    grads = {n: p.grad.clone() for n, p in model.named_parameters()}
+    if rank == 0:
+        out_q.put(grads)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_allreduced_grads_match_single_process():
+    # single-process full-batch reference (loss mean over full batch equals
+    # the average of per-shard means when shards are equal-sized)
+    model = _build_model()
+    x, y = _full_batch()
+    logits = model(x)
+    loss = nn.functional.binary_cross_entropy_with_logits(logits, y)
+    loss.backward()
+    ref = {n: p.grad.clone() for n, p in model.named_parameters()}
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_worker, args=(r, 2, 29571, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = q.get()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+
+    for n in ref:
+        assert torch.allclose(ref[n], got[n], atol=1e-6), n
