@@ -104,6 +104,10 @@ class BiGRU(nn.Module):
         avg_pool = gru_out.sum(dim=1) / float(input_length)
 
         concat_out = torch.cat([last_hidden, max_pool, avg_pool], dim=1)
+        if concat_out.dtype != self.linear.weight.dtype:
+            # bf16 compute path with fp32 master weights: cast the head.
+            return F.linear(concat_out, self.linear.weight.to(concat_out.dtype),
+                            self.linear.bias.to(concat_out.dtype))
         return self.linear(concat_out)
 
     def _gru_hip(self, x: torch.Tensor, hidden: Optional[torch.Tensor]):
