@@ -1,0 +1,3 @@
+from .bus import MessageBus, Topic  # noqa: F401
+from .streaming import FeatureRing, StreamingPredictor  # noqa: F401
+from .session import MarketCalendar, MarketSession  # noqa: F401

@@ -1,0 +1,151 @@
+"""Real-time streaming inference engine (the reference predict.py path).
+
+Semantics of predict.py:124-197, MI355X-native:
+- consume timestamp messages from the `predict_timestamp` topic (in-process
+  bus instead of Kafka);
+- drop stale messages older than `stale_after` (predict.py:135);
+- assemble the window of the latest `window` feature rows from a ring buffer
+  (replacing the SQL `SELECT ... WHERE sd.ID IN (...)` fetch,
+  predict.py:162-166);
+- min-max normalize with the norm_params scaling table (predict.py:175);
+- forward -> sigmoid -> threshold 0.5 -> publish labels to the `prediction`
+  topic (predict.py:177-197).
+
+On GPU the batch-1 step (normalize -> biGRU -> head -> sigmoid) is
+hipGraph-captured (torch.cuda.CUDAGraph == hipGraph on ROCm): one graph
+replay per tick instead of tens of small kernel launches.
+"""
+import time
+from typing import Dict, List, Optional
+
+import torch
+
+from ..features import FEATURE_NAMES, TARGET_NAMES
+from ..models.bigru import BiGRU
+from .bus import MessageBus
+
+
+class FeatureRing:
+    """Ring buffer of the latest feature rows (replaces the warehouse's
+    last-N-rows query)."""
+
+    def __init__(self, window: int, n_features: int):
+        self.window = window
+        self.buf = torch.zeros(window, n_features)
+        self.count = 0
+
+    def push(self, row: torch.Tensor) -> None:
+        self.buf = torch.roll(self.buf, -1, dims=0)
+        self.buf[-1] = row
+        self.count += 1
+
+    @property
+    def full(self) -> bool:
+        return self.count >= self.window
+
+
+class StreamingPredictor:
+    def __init__(self, model: BiGRU, x_min: torch.Tensor, x_max: torch.Tensor,
+                 window: int, bus: Optional[MessageBus] = None,
+                 prob_threshold: float = 0.5, stale_after: float = 240.0,
+                 device: str = "cpu", dtype: torch.dtype = torch.float32,
+                 use_graph: Optional[bool] = None):
+        self.model = model.eval().to(device)
+        self.window = window
+        self.n_features = model.n_features
+        self.bus = bus or MessageBus()
+        self.prob_threshold = prob_threshold
+        self.stale_after = stale_after
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self.ring = FeatureRing(window, self.n_features)
+        self.x_min = x_min.clone()
+        self.x_rng = (x_max - x_min).clone()
+        self.y_fields: List[str] = list(TARGET_NAMES)
+        self._graph = None
+        self._use_graph = (self.device.type == "cuda"
+                           if use_graph is None else use_graph)
+        self._static_in = torch.zeros(1, window, self.n_features,
+                                      device=self.device, dtype=self.dtype)
+        self._static_out = None
+        self.n_predictions = 0
+
+    # ---------------- feature ingestion ----------------
+
+    def push_row(self, row: torch.Tensor) -> None:
+        """Ingest one raw (unnormalized) 108-feature row."""
+        self.ring.push(row)
+
+    # ---------------- the batch-1 step ----------------
+
+    def _capture_graph(self):
+        g = torch.cuda.CUDAGraph()
+        # warmup on a side stream (required before capture)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s), torch.no_grad():
+            for _ in range(3):
+                out = torch.sigmoid(self.model(self._static_in))
+        torch.cuda.current_stream().wait_stream(s)
+        with torch.cuda.graph(g), torch.no_grad():
+            self._static_out = torch.sigmoid(self.model(self._static_in))
+        self._graph = g
+
+    def _forward_probs(self, x_norm: torch.Tensor) -> torch.Tensor:
+        if self._use_graph:
+            if self._graph is None:
+                self._capture_graph()
+            self._static_in.copy_(x_norm.to(self.device, self.dtype))
+            self._graph.replay()
+            return self._static_out.float()
+        with torch.no_grad():
+            x = x_norm.to(self.device, self.dtype)
+            return torch.sigmoid(self.model(x)).float()
+
+    def predict_window(self) -> Dict:
+        """Run inference on the current window; returns the prediction dict
+        (shape of predict.py:193-194)."""
+        x = self.ring.buf.unsqueeze(0)  # (1, window, F)
+        x_norm = (x - self.x_min) / self.x_rng
+        probs = self._forward_probs(x_norm).squeeze(0).cpu()
+        idx = (probs > self.prob_threshold).nonzero().flatten().tolist()
+        labels = [self.y_fields[i] for i in idx]
+        self.n_predictions += 1
+        return {"probabilities": probs.tolist(),
+                "prob_threshold": self.prob_threshold,
+                "pred_indices": idx, "pred_labels": labels}
+
+    # ---------------- message loop ----------------
+
+    def handle_timestamp(self, msg: Dict, now: Optional[float] = None) -> Optional[Dict]:
+        """Process one predict_timestamp message; returns the published
+        prediction dict, or None if stale/insufficient data."""
+        ts = float(msg["Timestamp"])
+        now = time.time() if now is None else now
+        if ts <= now - self.stale_after:    # stale filter (predict.py:135)
+            return None
+        if not self.ring.full:
+            return None
+        pred = self.predict_window()
+        pred["timestamp"] = ts
+        self.bus.publish("prediction", pred)
+        return pred
+
+    def run(self, max_messages: Optional[int] = None,
+            timeout: Optional[float] = 1.0) -> int:
+        """Consume predict_timestamp messages until the topic drains (or
+        max_messages). The consumer offset persists across calls."""
+        topic = self.bus.topic("predict_timestamp")
+        if not hasattr(self, "_offset"):
+            self._offset = 0
+        n = 0
+        while True:
+            msg = topic.read(self._offset, timeout=timeout)
+            if msg is None:
+                break
+            self._offset += 1
+            self.handle_timestamp(msg)
+            n += 1
+            if max_messages is not None and n >= max_messages:
+                break
+        return n

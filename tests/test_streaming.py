@@ -1,0 +1,95 @@
+"""Streaming inference path: bus, ring buffer, stale filter, end-to-end
+session -> predictor -> prediction topic (reference predict.py semantics)."""
+import time
+
+import pytest
+import torch
+
+from fmda_amd.data import SyntheticMarket, ChunkLoader, load_norm_params
+from fmda_amd.models import BiGRU
+from fmda_amd.runtime import (FeatureRing, MarketCalendar, MarketSession,
+                              MessageBus, StreamingPredictor)
+
+
+def _predictor(window=5, bus=None, **kw):
+    torch.manual_seed(0)
+    m = BiGRU(8, 108, 4, spatial_dropout=False, dropout=0.2)
+    x_min = torch.zeros(108)
+    x_max = torch.ones(108)
+    return StreamingPredictor(m, x_min, x_max, window, bus=bus,
+                              use_graph=False, **kw)
+
+
+def test_bus_publish_consume():
+    bus = MessageBus()
+    for i in range(5):
+        bus.publish("vix", {"VIX": float(i)})
+    got = list(bus.consume("vix", from_end=False, timeout=0.01))
+    assert [m["VIX"] for m in got] == [0.0, 1.0, 2.0, 3.0, 4.0]
+    # seek-to-end semantics (predict.py:30)
+    assert list(bus.consume("vix", from_end=True, timeout=0.01)) == []
+
+
+def test_ring_buffer_window():
+    ring = FeatureRing(3, 4)
+    for i in range(5):
+        ring.push(torch.full((4,), float(i)))
+    assert ring.full
+    assert torch.equal(ring.buf[:, 0], torch.tensor([2.0, 3.0, 4.0]))
+
+
+def test_stale_message_filter():
+    p = _predictor()
+    for i in range(6):
+        p.push_row(torch.rand(108))
+    now = time.time()
+    assert p.handle_timestamp({"Timestamp": now - 300.0}, now=now) is None
+    assert p.handle_timestamp({"Timestamp": now - 10.0}, now=now) is not None
+
+
+def test_prediction_schema_and_threshold():
+    p = _predictor()
+    for i in range(6):
+        p.push_row(torch.rand(108))
+    pred = p.predict_window()
+    assert set(pred) == {"probabilities", "prob_threshold", "pred_indices",
+                        "pred_labels"}
+    probs = torch.tensor(pred["probabilities"])
+    assert probs.shape == (4,)
+    expect_idx = (probs > 0.5).nonzero().flatten().tolist()
+    assert pred["pred_indices"] == expect_idx
+    assert pred["pred_labels"] == [["up1", "up2", "down1", "down2"][i]
+                                   for i in expect_idx]
+
+
+def test_session_end_to_end(tmp_path):
+    """Session publishes bars; predictor consumes timestamps and publishes
+    predictions with real norm_params scaling."""
+    mk = SyntheticMarket(260, seed=5)
+    path = str(tmp_path / "norm_params")
+    ChunkLoader(mk.X, chunk_size=100, window=5, norm_params_path=path)
+    names, x_min, x_max = load_norm_params(path)
+
+    bus = MessageBus()
+    torch.manual_seed(1)
+    model = BiGRU(8, 108, 4, spatial_dropout=False)
+    pred = StreamingPredictor(model, x_min, x_max, window=5, bus=bus,
+                              use_graph=False, stale_after=1e12)
+    sess = MarketSession(mk, bus=bus, predictor=pred,
+                         calendar=MarketCalendar(forex_fallback=True))
+    # interleave ticks with message handling (live operation)
+    for _ in range(50):
+        assert sess.tick()
+        pred.run(max_messages=1, timeout=0.01)
+    preds = list(bus.consume("prediction", from_end=False, timeout=0.01))
+    # first window-1 bars have no full window yet
+    assert len(preds) == 50 - 4
+    assert all(torch.isfinite(torch.tensor(p["probabilities"])).all()
+               for p in preds)
+
+
+def test_calendar_gate():
+    cal = MarketCalendar()
+    assert cal.is_open(0)            # Monday
+    assert not cal.is_open(5 * 78)   # Saturday
+    assert MarketCalendar(forex_fallback=True).is_open(5 * 78)
