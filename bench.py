@@ -41,6 +41,54 @@ def three_class_accuracy(target: torch.Tensor, pred: torch.Tensor) -> float:
     return float((to3(target.bool()) == to3(pred.bool())).float().mean())
 
 
+def bench_predict(cfg, args, use_cuda):
+    """BASELINE.json configs[4]: batch=1 streaming inference p50 latency,
+    hipGraph-captured step on GPU."""
+    import numpy as np
+    from fmda_amd.runtime import StreamingPredictor
+    device = "cuda:0" if use_cuda else "cpu"
+    dtype = torch.bfloat16 if use_cuda else torch.float32
+    torch.manual_seed(0)
+    model = BiGRU(cfg.hidden_size, cfg.n_features, 4, n_layers=cfg.n_layers,
+                  spatial_dropout=False, dropout=0.2)
+    if dtype == torch.bfloat16:
+        model = model.to(dtype)
+    pred = StreamingPredictor(model, torch.zeros(cfg.n_features),
+                              torch.ones(cfg.n_features), cfg.seq_len,
+                              device=device, dtype=dtype)
+    g = torch.Generator().manual_seed(1)
+    for _ in range(cfg.seq_len):
+        pred.push_row(torch.rand(cfg.n_features, generator=g))
+    for _ in range(max(args.warmup, 3)):
+        pred.predict_window()
+    if use_cuda:
+        torch.cuda.synchronize()
+    lats = []
+    for i in range(max(args.steps, 50)):
+        pred.push_row(torch.rand(cfg.n_features, generator=g))
+        t0 = time.perf_counter()
+        pred.predict_window()
+        if use_cuda:
+            torch.cuda.synchronize()
+        lats.append((time.perf_counter() - t0) * 1000.0)
+    p50 = float(np.percentile(lats, 50))
+    result = {
+        "metric": "p50 batch=1 streaming predict latency",
+        "value": p50, "unit": "ms", "n_gpus": 1 if use_cuda else 0,
+        "steps": len(lats), "warmup": max(args.warmup, 3),
+        "ms_per_step": p50, "higher_is_better": False, "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
+        "data": "synthetic",
+        "config": {"model": f"biGRU L{cfg.n_layers} H{cfg.hidden_size}",
+                   "global_batch": 1, "seq_len": cfg.seq_len,
+                   "parallelism": "single",
+                   "p90_ms": float(np.percentile(lats, 90)),
+                   "hipgraph": bool(use_cuda)},
+    }
+    print(json.dumps(result))
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -59,6 +107,10 @@ def main():
 
     cfg_name = args.config or ("repo" if use_cuda else "cpu")
     cfg = BENCH_CONFIGS[cfg_name]
+
+    if cfg_name == "predict":
+        bench_predict(cfg, args, use_cuda)
+        return
     batch = args.batch or cfg.batch_per_gpu
     dtype = torch.bfloat16 if (cfg.dtype == "bf16" and use_cuda) else torch.float32
 

@@ -1,0 +1,118 @@
+"""Training driver — the reference notebook (`biGRU_model_training.ipynb`)
+as a reproducible script.
+
+Per-epoch flow matches the notebook's cell 29: chunk-level re-split,
+per-chunk DataLoader over sliding windows, train_model / evaluate_model
+with sklearn-formula metrics, class-imbalance loss weighting (cell 16:
+weight = total/positives, pos_weight = negatives/positives), Adam lr=1e-3,
+clip=50, and `torch.save(state_dict)` of the best model (cell 39).
+
+Run: python -m fmda_amd.train [--epochs 25] [--device cuda]
+"""
+import argparse
+import json
+import time
+from typing import Optional
+
+import torch
+import torch.nn as nn
+from torch.utils.data import DataLoader
+
+from .config import DataConfig, ModelConfig, TrainConfig
+from .data import BatchLoader, ChunkLoader, SyntheticMarket, TrainValTestSplit
+from .models import BiGRU, save_checkpoint
+
+
+def class_weights(Y: torch.Tensor):
+    """Loss weights from label counts (notebook cell 16)."""
+    total = Y.shape[0]
+    pos = Y.sum(dim=0).clamp(min=1.0)
+    weight = total / pos
+    pos_weight = (total - pos) / pos
+    return weight, pos_weight
+
+
+def make_epoch_sets(market: SyntheticMarket, dcfg: DataConfig,
+                    norm_params_path: Optional[str] = None):
+    chunks = ChunkLoader(market.X[:, :dcfg.n_features], dcfg.chunk_size,
+                         dcfg.window, norm_params_path=norm_params_path)
+    split = TrainValTestSplit(chunks, dcfg.val_size, dcfg.test_size)
+    return split.get_sets()
+
+
+def chunk_batches(market, chunk_set, dcfg, batch_size):
+    """Yield (x, y) batches over every chunk in the set (notebook epoch
+    inner loop)."""
+    for indices, norms in chunk_set:
+        bl = BatchLoader(indices, norms, market.X[:, :dcfg.n_features],
+                         market.Y, dcfg.window)
+        dl = DataLoader(bl, batch_size=batch_size, shuffle=False)
+        for x, y in dl:
+            yield x, y
+
+
+def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
+          tcfg: TrainConfig = None, checkpoint_path: str = "model_params.pt",
+          norm_params_path: str = "norm_params", log=print):
+    mcfg = mcfg or ModelConfig()
+    dcfg = dcfg or DataConfig(n_features=mcfg.n_features)
+    tcfg = tcfg or TrainConfig()
+
+    device = torch.device(tcfg.device)
+    torch.manual_seed(dcfg.seed)
+
+    market = SyntheticMarket(dcfg.n_rows, seed=dcfg.seed)
+    weight, pos_weight = class_weights(market.Y)
+
+    model = BiGRU(mcfg.hidden_size, mcfg.n_features, mcfg.output_size,
+                  n_layers=mcfg.n_layers, clip=mcfg.clip,
+                  dropout=mcfg.dropout, spatial_dropout=mcfg.spatial_dropout,
+                  bidirectional=mcfg.bidirectional).to(device)
+    model.add_loss_fn(nn.BCEWithLogitsLoss(weight=weight.to(device),
+                                           pos_weight=pos_weight.to(device)))
+    model.add_optimizer(torch.optim.Adam(model.parameters(), lr=tcfg.lr))
+    model.add_device(device)
+
+    history = []
+    best_val_acc = -1.0
+    for epoch in range(1, tcfg.epochs + 1):
+        t0 = time.time()
+        train_set, val_set, _ = make_epoch_sets(market, dcfg,
+                                                norm_params_path)
+        tr = model.train_model(
+            chunk_batches(market, train_set, dcfg, tcfg.batch_size))
+        va = model.evaluate_model(
+            chunk_batches(market, val_set, dcfg, tcfg.batch_size))
+        rec = {"epoch": epoch, "train_acc": float(tr[0]),
+               "train_hamming": float(tr[1]), "train_loss": float(tr[2]),
+               "val_acc": float(va[0]), "val_hamming": float(va[1]),
+               "sec": time.time() - t0}
+        history.append(rec)
+        log(json.dumps(rec))
+        if rec["val_acc"] >= best_val_acc:
+            best_val_acc = rec["val_acc"]
+            save_checkpoint(model, checkpoint_path)
+    return model, history
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--epochs", type=int, default=25)
+    ap.add_argument("--hidden", type=int, default=32)
+    ap.add_argument("--layers", type=int, default=1)
+    ap.add_argument("--rows", type=int, default=3980)
+    ap.add_argument("--window", type=int, default=30)
+    ap.add_argument("--batch", type=int, default=2)
+    ap.add_argument("--device", default="cpu")
+    ap.add_argument("--checkpoint", default="model_params.pt")
+    args = ap.parse_args()
+    mcfg = ModelConfig(hidden_size=args.hidden, n_layers=args.layers,
+                       spatial_dropout=False, dropout=0.5)
+    dcfg = DataConfig(n_rows=args.rows, window=args.window)
+    tcfg = TrainConfig(batch_size=args.batch, epochs=args.epochs,
+                       device=args.device)
+    train(mcfg, dcfg, tcfg, checkpoint_path=args.checkpoint)
+
+
+if __name__ == "__main__":
+    main()

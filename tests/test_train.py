@@ -1,0 +1,47 @@
+"""Training driver: one short epoch end-to-end on CPU (notebook cell 29
+semantics at reduced scale) + checkpoint emission."""
+import os
+
+import torch
+
+from fmda_amd.config import DataConfig, ModelConfig, TrainConfig
+from fmda_amd.train import class_weights, train
+
+
+def test_class_weights():
+    Y = torch.tensor([[1, 0, 0, 0], [1, 1, 0, 0], [0, 0, 1, 0],
+                      [0, 0, 0, 0]], dtype=torch.float32)
+    w, pw = class_weights(Y)
+    assert torch.allclose(w, torch.tensor([2.0, 4.0, 4.0, 4.0]))
+    # positives are clamped to >=1 before both ratios (all-zero class)
+    assert torch.allclose(pw, torch.tensor([1.0, 3.0, 3.0, 3.0]))
+
+
+def test_train_one_epoch(tmp_path):
+    ckpt = str(tmp_path / "model_params.pt")
+    npp = str(tmp_path / "norm_params")
+    mcfg = ModelConfig(hidden_size=8, spatial_dropout=False, dropout=0.2)
+    dcfg = DataConfig(n_rows=420, chunk_size=100, window=10, seed=3)
+    tcfg = TrainConfig(batch_size=4, epochs=1)
+    model, hist = train(mcfg, dcfg, tcfg, checkpoint_path=ckpt,
+                        norm_params_path=npp, log=lambda s: None)
+    assert len(hist) == 1
+    assert 0.0 <= hist[0]["train_acc"] <= 1.0
+    assert os.path.exists(ckpt) and os.path.exists(npp)
+    sd = torch.load(ckpt)
+    assert "gru.weight_ih_l0" in sd and "linear.bias" in sd
+    # norm_params is loadable in the reference pickle format
+    from fmda_amd.data import load_norm_params
+    names, x_min, x_max = load_norm_params(npp)
+    assert len(names) == 108
+
+
+def test_loss_decreases_over_epochs(tmp_path):
+    mcfg = ModelConfig(hidden_size=16, spatial_dropout=False, dropout=0.0)
+    dcfg = DataConfig(n_rows=420, chunk_size=100, window=10, seed=4)
+    tcfg = TrainConfig(batch_size=8, epochs=3)
+    _, hist = train(mcfg, dcfg, tcfg,
+                    checkpoint_path=str(tmp_path / "m.pt"),
+                    norm_params_path=str(tmp_path / "np"),
+                    log=lambda s: None)
+    assert hist[-1]["train_loss"] < hist[0]["train_loss"]
