@@ -12,8 +12,8 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const float* bhh,
                                    const void* out, const void* dout,
                                    const float* dhT, void* dgi, void* dgh,
-                                   float* dh0, int B, int Tseq, int n_dir,
-                                   hipStream_t stream);
+                                   float* dh0, float* dbhh, int B, int Tseq,
+                                   int n_dir, hipStream_t stream);
 extern "C" int fmda_mfma_selftest_launch(const void* A, const void* Bm,
                                          float* C, hipStream_t stream);
 
@@ -73,15 +73,18 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
     auto dgh = torch::empty_like(gi);
     auto dh0 = torch::empty({n_dir, B, Hp},
                             gi.options().dtype(torch::kFloat32));
+    auto dbhh = torch::zeros({n_dir, 3 * Hp},
+                             gi.options().dtype(torch::kFloat32));
     auto stream = at::hip::getCurrentHIPStream();
     int rc = fmda_gru_bwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
                                  w.data_ptr(), bhh.data_ptr<float>(),
                                  out.data_ptr(), dout.data_ptr(),
                                  dhT.data_ptr<float>(), dgi.data_ptr(),
-                                 dgh.data_ptr(), dh0.data_ptr<float>(), B, T,
+                                 dgh.data_ptr(), dh0.data_ptr<float>(),
+                                 dbhh.data_ptr<float>(), B, T,
                                  n_dir, stream.stream());
     TORCH_CHECK(rc == 0, "fmda gru_bwd launch failed rc=", rc, " Hp=", Hp);
-    return {dgi, dgh, dh0};
+    return {dgi, dgh, dh0, dbhh};
 }
 
 torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor Bm) {

@@ -64,14 +64,15 @@ struct alignas(16) chunk16 { unsigned int u[4]; };
 // LDS tile [rows][cols] <- global rows at row_stride (elements).
 template <typename T, int ROWS>
 FMDA_DEV void stage_tile(T* __restrict__ lds, const T* __restrict__ gp,
-                         int cols, long row_stride, int rows_valid, int tid) {
+                         int cols, int lds_pitch, long row_stride,
+                         int rows_valid, int tid) {
     const int cpr = (cols * (int)sizeof(T)) / 16;  // chunks per row
     const int total = ROWS * cpr;
     for (int c = tid; c < total; c += 256) {
         const int r = c / cpr;
         const int jc = c % cpr;
         if (r < rows_valid) {
-            ((chunk16*)((char*)lds + (long)r * cols * sizeof(T)))[jc] =
+            ((chunk16*)((char*)lds + (long)r * lds_pitch * sizeof(T)))[jc] =
                 ((const chunk16*)((const char*)gp +
                                   (long)r * row_stride * sizeof(T)))[jc];
         }
@@ -80,7 +81,8 @@ FMDA_DEV void stage_tile(T* __restrict__ lds, const T* __restrict__ gp,
 
 template <typename T, int ROWS>
 FMDA_DEV void store_tile(const T* __restrict__ lds, T* __restrict__ gp,
-                         int cols, long row_stride, int rows_valid, int tid) {
+                         int cols, int lds_pitch, long row_stride,
+                         int rows_valid, int tid) {
     const int cpr = (cols * (int)sizeof(T)) / 16;
     const int total = ROWS * cpr;
     for (int c = tid; c < total; c += 256) {
@@ -89,26 +91,28 @@ FMDA_DEV void store_tile(const T* __restrict__ lds, T* __restrict__ gp,
         if (r < rows_valid) {
             ((chunk16*)((char*)gp + (long)r * row_stride * sizeof(T)))[jc] =
                 ((const chunk16*)((const char*)lds +
-                                  (long)r * cols * sizeof(T)))[jc];
+                                  (long)r * lds_pitch * sizeof(T)))[jc];
         }
     }
 }
 
 template <typename T, int ROWS>
-FMDA_DEV void zero_tile(T* __restrict__ lds, int cols, int tid) {
-    for (int c = tid; c < ROWS * cols; c += 256) lds[c] = from_f32<T>(0.0f);
+FMDA_DEV void zero_tile(T* __restrict__ lds, int cols, int lds_pitch, int tid) {
+    for (int c = tid; c < ROWS * lds_pitch; c += 256) lds[c] = from_f32<T>(0.0f);
+    (void)cols;
 }
 
 // fp32 LDS tile += bf16/f32 global tile (used for dh += dOut[t]).
 template <typename T, int ROWS>
 FMDA_DEV void accum_tile_f32(float* __restrict__ lds, const T* __restrict__ gp,
-                             int cols, long row_stride, int rows_valid, int tid) {
+                             int cols, int lds_pitch, long row_stride,
+                             int rows_valid, int tid) {
     const int total = ROWS * cols;
     for (int c = tid; c < total; c += 256) {
         const int r = c / cols;
         const int j = c % cols;
         if (r < rows_valid)
-            lds[c] += to_f32<T>(gp[(long)r * row_stride + j]);
+            lds[r * lds_pitch + j] += to_f32<T>(gp[(long)r * row_stride + j]);
     }
 }
 
@@ -128,7 +132,7 @@ FMDA_DEV void accum_tile_f32(float* __restrict__ lds, const T* __restrict__ gp,
 // hb: LDS [BT][Hp] fragment source (bf16 shadow of h, or fp32 h directly).
 // w_row(n) returns pointer to row n (length Hp) of W_hh (LDS or global).
 template <int BT, int Hp>
-FMDA_DEV void gemm_ct_bf16(const __hip_bfloat16* __restrict__ hb,
+FMDA_DEV void gemm_ct_bf16(const __hip_bfloat16* __restrict__ hb, int hb_pitch,
                            const __hip_bfloat16* __restrict__ wbase,
                            long w_row_stride, int ct, int lane,
                            f32x4_t acc[3][BT / 16]) {
@@ -142,7 +146,7 @@ FMDA_DEV void gemm_ct_bf16(const __hip_bfloat16* __restrict__ hb,
         bf16x8_t a[MT];
 #pragma unroll
         for (int m = 0; m < MT; ++m)
-            a[m] = *(const bf16x8_t*)&hb[(16 * m + arow) * Hp + kbase];
+            a[m] = *(const bf16x8_t*)&hb[(16 * m + arow) * hb_pitch + kbase];
 #pragma unroll
         for (int g = 0; g < 3; ++g) {
             const long n = g * Hp + jcol;
@@ -156,7 +160,7 @@ FMDA_DEV void gemm_ct_bf16(const __hip_bfloat16* __restrict__ hb,
 }
 
 template <int BT, int Hp>
-FMDA_DEV void gemm_ct_f32(const float* __restrict__ hf,
+FMDA_DEV void gemm_ct_f32(const float* __restrict__ hf, int hb_pitch,
                           const float* __restrict__ wbase,
                           long w_row_stride, int ct, int lane,
                           f32x4_t acc[3][BT / 16]) {
@@ -172,7 +176,7 @@ FMDA_DEV void gemm_ct_f32(const float* __restrict__ hf,
                 const int row = 16 * m + 4 * (lane >> 4) + e;
                 float s = acc[g][m][e];
                 for (int k = 0; k < Hp; ++k)
-                    s += hf[row * Hp + k] * wrow[k];
+                    s += hf[row * hb_pitch + k] * wrow[k];
                 acc[g][m][e] = s;
             }
         }
@@ -182,7 +186,7 @@ FMDA_DEV void gemm_ct_f32(const float* __restrict__ hf,
 // dh GEMM: dh[b][j] += sum_n dgh[b][n] * W[n][j]  (K = 3*Hp, N = Hp).
 // A = dgh (LDS, row-major, contiguous k). B[k=n][j] = W[n][j]: strided reads.
 template <int BT, int Hp>
-FMDA_DEV void gemm_dh_bf16(const __hip_bfloat16* __restrict__ dgh,
+FMDA_DEV void gemm_dh_bf16(const __hip_bfloat16* __restrict__ dgh, int dgh_pitch,
                            const __hip_bfloat16* __restrict__ wbase,
                            long w_row_stride, int ct, int lane,
                            f32x4_t acc[BT / 16]) {
@@ -196,7 +200,7 @@ FMDA_DEV void gemm_dh_bf16(const __hip_bfloat16* __restrict__ dgh,
         bf16x8_t a[MT];
 #pragma unroll
         for (int m = 0; m < MT; ++m)
-            a[m] = *(const bf16x8_t*)&dgh[(16 * m + arow) * (3 * Hp) + kbase];
+            a[m] = *(const bf16x8_t*)&dgh[(16 * m + arow) * dgh_pitch + kbase];
         bf16x8_t b;
         const __bf16* wb = (const __bf16*)wbase;
 #pragma unroll
@@ -210,7 +214,7 @@ FMDA_DEV void gemm_dh_bf16(const __hip_bfloat16* __restrict__ dgh,
 }
 
 template <int BT, int Hp>
-FMDA_DEV void gemm_dh_f32(const float* __restrict__ dgh,
+FMDA_DEV void gemm_dh_f32(const float* __restrict__ dgh, int dgh_pitch,
                           const float* __restrict__ wbase,
                           long w_row_stride, int ct, int lane,
                           f32x4_t acc[BT / 16]) {
@@ -223,7 +227,7 @@ FMDA_DEV void gemm_dh_f32(const float* __restrict__ dgh,
             const int row = 16 * m + 4 * (lane >> 4) + e;
             float s = acc[m][e];
             for (int n = 0; n < 3 * Hp; ++n)
-                s += dgh[row * 3 * Hp + n] * wbase[(long)n * w_row_stride + jcol];
+                s += dgh[row * dgh_pitch + n] * wbase[(long)n * w_row_stride + jcol];
             acc[m][e] = s;
         }
     }
@@ -248,6 +252,14 @@ __global__ __launch_bounds__(256) void gru_fwd_kernel(
     constexpr int NCT = Hp / 16;          // column tiles per gate
     constexpr int CPW = (NCT + 3) / 4;    // column tiles per wave (max)
     constexpr bool IS_BF16 = !__is_same(T, float);
+    // Padded LDS row pitches (+16 B per row): the fragment reads hit 16
+    // distinct rows at one 16-B column slot; an even power-of-two row pitch
+    // puts the whole lane group on one LDS bank slot (16-way conflict), the
+    // odd-slot pitch spreads it over all 16 slots (conflict-free).
+    constexpr int PADE = 16 / (int)sizeof(T);
+    constexpr int WP = Hp + PADE;           // pitch of Hp-wide T rows
+    constexpr int GP3 = 3 * Hp + PADE;      // pitch of 3Hp-wide T rows
+    constexpr int HFP = Hp + 4;             // pitch of Hp-wide fp32 rows
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -260,24 +272,22 @@ __global__ __launch_bounds__(256) void gru_fwd_kernel(
     extern __shared__ __attribute__((aligned(16))) char smem[];
     char* p = smem;
     T* w_s = nullptr;
-    if (WLDS) { w_s = (T*)p; p += sizeof(T) * 3 * Hp * Hp; }
-    float* hf_s = (float*)p; p += sizeof(float) * BT * Hp;
+    if (WLDS) { w_s = (T*)p; p += sizeof(T) * 3 * Hp * WP; }
+    float* hf_s = (float*)p; p += sizeof(float) * BT * HFP;
     __hip_bfloat16* hb_s = nullptr;
-    if (IS_BF16) { hb_s = (__hip_bfloat16*)p; p += 2 * BT * Hp; }
-    T* gi_s = (T*)p; p += sizeof(T) * BT * 3 * Hp;
+    if (IS_BF16) { hb_s = (__hip_bfloat16*)p; p += 2 * BT * WP; }
+    T* gi_s = (T*)p; p += sizeof(T) * BT * GP3;
     float* bhh_s = (float*)p;
 
     const T* wdir = w + (long)dir * 3 * Hp * Hp;
-    if (WLDS) {
-        for (int c = tid; c < (3 * Hp * Hp) / 8; c += 256)
-            *(short8_t*)&w_s[c * 8] = *(const short8_t*)&wdir[c * 8];
-    }
+    if (WLDS)
+        stage_tile<T, 3 * Hp>(w_s, wdir, Hp, WP, Hp, 3 * Hp, tid);
     for (int c = tid; c < 3 * Hp; c += 256)
         bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
-    for (int c = tid; c < BT * Hp; c += 256) {
-        hf_s[c] = 0.0f;
-        if (IS_BF16) hb_s[c] = __float2bfloat16(0.0f);
-    }
+    for (int c = tid; c < BT * HFP; c += 256) hf_s[c] = 0.0f;
+    if (IS_BF16)
+        for (int c = tid; c < BT * WP; c += 256)
+            hb_s[c] = __float2bfloat16(0.0f);
 
     const long gi_row = (long)Tseq * n_dir * 3 * Hp;   // per batch row
     const long out_row = (long)Tseq * n_dir * Hp;
@@ -287,13 +297,13 @@ __global__ __launch_bounds__(256) void gru_fwd_kernel(
     // stage gi for step 0
     {
         const int tt = rev ? (Tseq - 1) : 0;
-        stage_tile<T, BT>(gi_s, gi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp,
+        stage_tile<T, BT>(gi_s, gi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp, GP3,
                           gi_row, rows_valid, tid);
     }
     __syncthreads();
 
     const T* wfrag = WLDS ? w_s : wdir;
-    const long wstride = Hp;  // row stride in elements for both cases
+    const long wstride = WLDS ? WP : Hp;  // row stride in elements
 
     for (int u = 0; u < Tseq; ++u) {
         const int tt = rev ? (Tseq - 1 - u) : u;
@@ -310,10 +320,10 @@ __global__ __launch_bounds__(256) void gru_fwd_kernel(
             const int ct = wave + 4 * i;
             if (ct < NCT) {
                 if constexpr (IS_BF16)
-                    gemm_ct_bf16<BT, Hp>(hb_s, wfrag, wstride, ct, lane,
+                    gemm_ct_bf16<BT, Hp>(hb_s, WP, wfrag, wstride, ct, lane,
                                          acc[i]);
                 else
-                    gemm_ct_f32<BT, Hp>(hf_s, wfrag, wstride, ct, lane,
+                    gemm_ct_f32<BT, Hp>(hf_s, HFP, wfrag, wstride, ct, lane,
                                         acc[i]);
             }
         }
@@ -333,17 +343,17 @@ __global__ __launch_bounds__(256) void gru_fwd_kernel(
                     const float gr = acc[i][0][m][e] + bhh_s[j];
                     const float gz = acc[i][1][m][e] + bhh_s[Hp + j];
                     const float hn = acc[i][2][m][e] + bhh_s[2 * Hp + j];
-                    const float ir = to_f32<T>(gi_s[b * 3 * Hp + j]);
-                    const float iz = to_f32<T>(gi_s[b * 3 * Hp + Hp + j]);
-                    const float in_ = to_f32<T>(gi_s[b * 3 * Hp + 2 * Hp + j]);
+                    const float ir = to_f32<T>(gi_s[b * GP3 + j]);
+                    const float iz = to_f32<T>(gi_s[b * GP3 + Hp + j]);
+                    const float in_ = to_f32<T>(gi_s[b * GP3 + 2 * Hp + j]);
                     const float r = sigmoidf(ir + gr);
                     const float z = sigmoidf(iz + gz);
                     const float n = tanhf(in_ + r * hn);
-                    const float hprev = hf_s[b * Hp + j];
+                    const float hprev = hf_s[b * HFP + j];
                     const float hnew = (1.0f - z) * n + z * hprev;
-                    hf_s[b * Hp + j] = hnew;
+                    hf_s[b * HFP + j] = hnew;
                     if constexpr (IS_BF16)
-                        hb_s[b * Hp + j] = __float2bfloat16(hnew);
+                        hb_s[b * WP + j] = __float2bfloat16(hnew);
                 }
             }
         }
@@ -352,16 +362,16 @@ __global__ __launch_bounds__(256) void gru_fwd_kernel(
         // ---- phase C: write out[t]; stage gi for t+1
         if constexpr (IS_BF16) {
             store_tile<__hip_bfloat16, BT>(
-                hb_s, (__hip_bfloat16*)(out_b + (long)tt * n_dir * Hp), Hp,
+                hb_s, (__hip_bfloat16*)(out_b + (long)tt * n_dir * Hp), Hp, WP,
                 out_row, rows_valid, tid);
         } else {
             store_tile<float, BT>(hf_s, (float*)(out_b + (long)tt * n_dir * Hp),
-                                  Hp, out_row, rows_valid, tid);
+                                  Hp, HFP, out_row, rows_valid, tid);
         }
         if (u + 1 < Tseq) {
             const int ttn = rev ? (Tseq - 2 - u) : (u + 1);
             stage_tile<T, BT>(gi_s, gi_b + (long)ttn * n_dir * 3 * Hp, 3 * Hp,
-                              gi_row, rows_valid, tid);
+                              GP3, gi_row, rows_valid, tid);
         }
         __syncthreads();
     }
@@ -370,7 +380,7 @@ __global__ __launch_bounds__(256) void gru_fwd_kernel(
     float* hl = hlast + ((long)dir * B + b0) * Hp;
     for (int c = tid; c < BT * Hp; c += 256) {
         const int r = c / Hp;
-        if (r < rows_valid) hl[(long)r * Hp + (c % Hp)] = hf_s[c];
+        if (r < rows_valid) hl[(long)r * Hp + (c % Hp)] = hf_s[r * HFP + (c % Hp)];
     }
 }
 
@@ -387,11 +397,15 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
     const float* __restrict__ bhh, const T* __restrict__ out,
     const T* __restrict__ dout, const float* __restrict__ dhT,
     T* __restrict__ dgi, T* __restrict__ dgh, float* __restrict__ dh0,
-    int B, int Tseq, int n_dir) {
+    float* __restrict__ dbhh, int B, int Tseq, int n_dir) {
     constexpr int MT = BT / 16;
     constexpr int NCT = Hp / 16;
     constexpr int CPW = (NCT + 3) / 4;
     constexpr bool IS_BF16 = !__is_same(T, float);
+    constexpr int PADE = 16 / (int)sizeof(T);
+    constexpr int WP = Hp + PADE;
+    constexpr int GP3 = 3 * Hp + PADE;
+    constexpr int HFP = Hp + 4;
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -404,18 +418,16 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
     extern __shared__ __attribute__((aligned(16))) char smem[];
     char* p = smem;
     T* w_s = nullptr;
-    if (WLDS) { w_s = (T*)p; p += sizeof(T) * 3 * Hp * Hp; }
-    float* dh_s = (float*)p; p += sizeof(float) * BT * Hp;
-    T* hb_s = (T*)p; p += sizeof(T) * BT * Hp;      // h_{t-1} (from out)
-    T* gi_s = (T*)p; p += sizeof(T) * BT * 3 * Hp;  // gi[t], then overwritten by dGi
-    T* dgh_s = (T*)p; p += sizeof(T) * BT * 3 * Hp;
+    if (WLDS) { w_s = (T*)p; p += sizeof(T) * 3 * Hp * WP; }
+    float* dh_s = (float*)p; p += sizeof(float) * BT * HFP;
+    T* hb_s = (T*)p; p += sizeof(T) * BT * WP;      // h_{t-1} (from out)
+    T* gi_s = (T*)p; p += sizeof(T) * BT * GP3;  // gi[t], then overwritten by dGi
+    T* dgh_s = (T*)p; p += sizeof(T) * BT * GP3;
     float* bhh_s = (float*)p;
 
     const T* wdir = w + (long)dir * 3 * Hp * Hp;
-    if (WLDS) {
-        for (int c = tid; c < (3 * Hp * Hp) / 8; c += 256)
-            *(short8_t*)&w_s[c * 8] = *(const short8_t*)&wdir[c * 8];
-    }
+    if (WLDS)
+        stage_tile<T, 3 * Hp>(w_s, wdir, Hp, WP, Hp, 3 * Hp, tid);
     for (int c = tid; c < 3 * Hp; c += 256)
         bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
 
@@ -432,7 +444,8 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
         const float* hT = dhT + ((long)dir * B + b0) * Hp;
         for (int c = tid; c < BT * Hp; c += 256) {
             const int r = c / Hp;
-            dh_s[c] = (r < rows_valid) ? hT[(long)r * Hp + (c % Hp)] : 0.0f;
+            dh_s[r * HFP + (c % Hp)] =
+                (r < rows_valid) ? hT[(long)r * Hp + (c % Hp)] : 0.0f;
         }
     }
 
@@ -440,22 +453,28 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
     {
         const int u = Tseq - 1;
         const int tt = rev ? (Tseq - 1 - u) : u;
-        stage_tile<T, BT>(gi_s, gi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp,
+        stage_tile<T, BT>(gi_s, gi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp, GP3,
                           gi_row, rows_valid, tid);
         if (u > 0) {
             const int ttp = rev ? (Tseq - u) : (u - 1);
-            stage_tile<T, BT>(hb_s, out_b + (long)ttp * n_dir * Hp, Hp,
+            stage_tile<T, BT>(hb_s, out_b + (long)ttp * n_dir * Hp, Hp, WP,
                               out_row, rows_valid, tid);
         } else {
-            zero_tile<T, BT>(hb_s, Hp, tid);
+            zero_tile<T, BT>(hb_s, Hp, WP, tid);
         }
-        accum_tile_f32<T, BT>(dh_s, dout_b + (long)tt * n_dir * Hp, Hp,
+        accum_tile_f32<T, BT>(dh_s, dout_b + (long)tt * n_dir * Hp, Hp, HFP,
                               out_row, rows_valid, tid);
     }
     __syncthreads();
 
     const T* wfrag = WLDS ? w_s : wdir;
-    const long wstride = Hp;
+    const long wstride = WLDS ? WP : Hp;
+    // per-lane db_hh partials accumulated across all T (summed over b rows)
+    float dbacc[CPW][3];
+#pragma unroll
+    for (int i = 0; i < CPW; ++i)
+#pragma unroll
+        for (int g = 0; g < 3; ++g) dbacc[i][g] = 0.0f;
 
     for (int u = Tseq - 1; u >= 0; --u) {
         const int tt = rev ? (Tseq - 1 - u) : u;
@@ -471,9 +490,9 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
 #pragma unroll
                 for (int m = 0; m < MT; ++m) acc[g][m] = f32x4_t{0.f};
             if constexpr (IS_BF16)
-                gemm_ct_bf16<BT, Hp>(hb_s, wfrag, wstride, ct, lane, acc);
+                gemm_ct_bf16<BT, Hp>(hb_s, WP, wfrag, wstride, ct, lane, acc);
             else
-                gemm_ct_f32<BT, Hp>((const float*)hb_s, wfrag, wstride, ct,
+                gemm_ct_f32<BT, Hp>((const float*)hb_s, WP, wfrag, wstride, ct,
                                     lane, acc);
             const int j = ct * 16 + (lane & 15);
 #pragma unroll
@@ -484,28 +503,33 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
                     const float gr = acc[0][m][e] + bhh_s[j];
                     const float gz = acc[1][m][e] + bhh_s[Hp + j];
                     const float hn = acc[2][m][e] + bhh_s[2 * Hp + j];
-                    const float ir = to_f32<T>(gi_s[b * 3 * Hp + j]);
-                    const float iz = to_f32<T>(gi_s[b * 3 * Hp + Hp + j]);
-                    const float in_ = to_f32<T>(gi_s[b * 3 * Hp + 2 * Hp + j]);
+                    const float ir = to_f32<T>(gi_s[b * GP3 + j]);
+                    const float iz = to_f32<T>(gi_s[b * GP3 + Hp + j]);
+                    const float in_ = to_f32<T>(gi_s[b * GP3 + 2 * Hp + j]);
                     const float r = sigmoidf(ir + gr);
                     const float z = sigmoidf(iz + gz);
                     const float n = tanhf(in_ + r * hn);
-                    const float hprev = to_f32<T>(hb_s[b * Hp + j]);
-                    const float dht = dh_s[b * Hp + j];
-                    const float dz_pre = dht * (hprev - n) * z * (1.0f - z);
-                    const float dn_pre = dht * (1.0f - z) * (1.0f - n * n);
+                    const float hprev = to_f32<T>(hb_s[b * WP + j]);
+                    const float dht = dh_s[b * HFP + j];
+                    const bool live = (b < rows_valid);
+                    const float dz_pre = live ? dht * (hprev - n) * z * (1.0f - z) : 0.0f;
+                    const float dn_pre = live ? dht * (1.0f - z) * (1.0f - n * n) : 0.0f;
                     const float dr_pre = dn_pre * hn * r * (1.0f - r);
                     const float dhn = dn_pre * r;
                     // dGi (packed r,z,n pre-activation grads) -> reuse gi_s
-                    gi_s[b * 3 * Hp + j] = from_f32<T>(dr_pre);
-                    gi_s[b * 3 * Hp + Hp + j] = from_f32<T>(dz_pre);
-                    gi_s[b * 3 * Hp + 2 * Hp + j] = from_f32<T>(dn_pre);
+                    gi_s[b * GP3 + j] = from_f32<T>(dr_pre);
+                    gi_s[b * GP3 + Hp + j] = from_f32<T>(dz_pre);
+                    gi_s[b * GP3 + 2 * Hp + j] = from_f32<T>(dn_pre);
                     // dGh differs in the n block only
-                    dgh_s[b * 3 * Hp + j] = from_f32<T>(dr_pre);
-                    dgh_s[b * 3 * Hp + Hp + j] = from_f32<T>(dz_pre);
-                    dgh_s[b * 3 * Hp + 2 * Hp + j] = from_f32<T>(dhn);
+                    dgh_s[b * GP3 + j] = from_f32<T>(dr_pre);
+                    dgh_s[b * GP3 + Hp + j] = from_f32<T>(dz_pre);
+                    dgh_s[b * GP3 + 2 * Hp + j] = from_f32<T>(dhn);
                     // direct part of dh_{t-1}; GEMM part added in phase B
-                    dh_s[b * Hp + j] = dht * z;
+                    dh_s[b * HFP + j] = dht * z;
+                    // db_hh partials (summed in fp32 across b and t)
+                    dbacc[i][0] += dr_pre;
+                    dbacc[i][1] += dz_pre;
+                    dbacc[i][2] += dhn;
                 }
             }
         }
@@ -520,41 +544,41 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
 #pragma unroll
             for (int m = 0; m < MT; ++m) acc2[m] = f32x4_t{0.f};
             if constexpr (IS_BF16)
-                gemm_dh_bf16<BT, Hp>((const __hip_bfloat16*)dgh_s, wfrag,
+                gemm_dh_bf16<BT, Hp>((const __hip_bfloat16*)dgh_s, GP3, wfrag,
                                      wstride, ct, lane, acc2);
             else
-                gemm_dh_f32<BT, Hp>((const float*)dgh_s, wfrag, wstride, ct,
-                                    lane, acc2);
+                gemm_dh_f32<BT, Hp>((const float*)dgh_s, GP3, wfrag, wstride,
+                                    ct, lane, acc2);
             const int j = ct * 16 + (lane & 15);
 #pragma unroll
             for (int m = 0; m < MT; ++m)
 #pragma unroll
                 for (int e = 0; e < 4; ++e) {
                     const int b = 16 * m + 4 * (lane >> 4) + e;
-                    dh_s[b * Hp + j] += acc2[m][e];
+                    dh_s[b * HFP + j] += acc2[m][e];
                 }
         }
         __syncthreads();
 
         // ---- phase C: write dGi/dGh; stage next step
-        store_tile<T, BT>(gi_s, dgi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp,
+        store_tile<T, BT>(gi_s, dgi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp, GP3,
                           gi_row, rows_valid, tid);
         store_tile<T, BT>(dgh_s, dgh_b + (long)tt * n_dir * 3 * Hp, 3 * Hp,
-                          gi_row, rows_valid, tid);
+                          GP3, gi_row, rows_valid, tid);
         if (u > 0) {
             const int un = u - 1;
             const int ttn = rev ? (Tseq - 1 - un) : un;
             stage_tile<T, BT>(gi_s, gi_b + (long)ttn * n_dir * 3 * Hp, 3 * Hp,
-                              gi_row, rows_valid, tid);
+                              GP3, gi_row, rows_valid, tid);
             if (un > 0) {
                 const int ttp = rev ? (Tseq - un) : (un - 1);
-                stage_tile<T, BT>(hb_s, out_b + (long)ttp * n_dir * Hp, Hp,
+                stage_tile<T, BT>(hb_s, out_b + (long)ttp * n_dir * Hp, Hp, WP,
                                   out_row, rows_valid, tid);
             } else {
-                zero_tile<T, BT>(hb_s, Hp, tid);
+                zero_tile<T, BT>(hb_s, Hp, WP, tid);
             }
             accum_tile_f32<T, BT>(dh_s, dout_b + (long)ttn * n_dir * Hp, Hp,
-                                  out_row, rows_valid, tid);
+                                  HFP, out_row, rows_valid, tid);
         }
         __syncthreads();
     }
@@ -563,7 +587,24 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
     float* d0 = dh0 + ((long)dir * B + b0) * Hp;
     for (int c = tid; c < BT * Hp; c += 256) {
         const int r = c / Hp;
-        if (r < rows_valid) d0[(long)r * Hp + (c % Hp)] = dh_s[c];
+        if (r < rows_valid) d0[(long)r * Hp + (c % Hp)] = dh_s[r * HFP + (c % Hp)];
+    }
+
+    // db_hh: reduce the 4 row-quarter lanes (same j = lane&15), then one
+    // atomicAdd per (gate, j) per block.
+#pragma unroll
+    for (int i = 0; i < CPW; ++i) {
+        const int ct = wave + 4 * i;
+        if (ct >= NCT) continue;
+#pragma unroll
+        for (int g = 0; g < 3; ++g) {
+            float v = dbacc[i][g];
+            v += __shfl_xor(v, 16);
+            v += __shfl_xor(v, 32);
+            if ((lane >> 4) == 0)
+                atomicAdd(&dbhh[(long)dir * 3 * Hp + g * Hp + ct * 16 +
+                                (lane & 15)], v);
+        }
     }
 }
 
@@ -623,23 +664,28 @@ static inline LaunchCfg bwd_cfg(bool bf16, int Hp) {
     return {16, false};
 }
 
+// NOTE: must mirror the in-kernel padded pitches (WP/GP3/HFP).
 static inline size_t fwd_lds_bytes(bool bf16, int Hp, int bt, bool wlds) {
     const size_t es = bf16 ? 2 : 4;
+    const size_t pade = 16 / es;
+    const size_t wp = Hp + pade, gp3 = 3 * Hp + pade, hfp = Hp + 4;
     size_t s = 0;
-    if (wlds) s += es * 3 * Hp * Hp;
-    s += 4 * (size_t)bt * Hp;              // hf_s
-    if (bf16) s += 2 * (size_t)bt * Hp;    // hb_s
-    s += es * (size_t)bt * 3 * Hp;         // gi_s
+    if (wlds) s += es * 3 * Hp * wp;
+    s += 4 * (size_t)bt * hfp;             // hf_s
+    if (bf16) s += 2 * (size_t)bt * wp;    // hb_s
+    s += es * (size_t)bt * gp3;            // gi_s
     s += 4 * (size_t)3 * Hp;               // bhh_s
     return s;
 }
 static inline size_t bwd_lds_bytes(bool bf16, int Hp, int bt, bool wlds) {
     const size_t es = bf16 ? 2 : 4;
+    const size_t pade = 16 / es;
+    const size_t wp = Hp + pade, gp3 = 3 * Hp + pade, hfp = Hp + 4;
     size_t s = 0;
-    if (wlds) s += es * 3 * Hp * Hp;
-    s += 4 * (size_t)bt * Hp;              // dh_s
-    s += es * (size_t)bt * Hp;             // hb_s
-    s += 2 * es * (size_t)bt * 3 * Hp;     // gi_s + dgh_s
+    if (wlds) s += es * 3 * Hp * wp;
+    s += 4 * (size_t)bt * hfp;             // dh_s
+    s += es * (size_t)bt * wp;             // hb_s
+    s += 2 * es * (size_t)bt * gp3;        // gi_s + dgh_s
     s += 4 * (size_t)3 * Hp;               // bhh_s
     return s;
 }
@@ -717,8 +763,8 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const float* bhh,
                                    const void* out, const void* dout,
                                    const float* dhT, void* dgi, void* dgh,
-                                   float* dh0, int B, int Tseq, int n_dir,
-                                   hipStream_t stream) {
+                                   float* dh0, float* dbhh, int B, int Tseq,
+                                   int n_dir, hipStream_t stream) {
     if (!is_bf16 && Hp > 256) return -3;  // fp32 oracle unsupported at H=512
     const LaunchCfg cfg = bwd_cfg(is_bf16, Hp);
     const size_t lds = bwd_lds_bytes(is_bf16, Hp, cfg.bt, cfg.wlds);
@@ -733,7 +779,7 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
             hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);             \
         k<<<grid, 256, lds, stream>>>((const TY*)gi, (const TY*)w, bhh,        \
             (const TY*)out, (const TY*)dout, dhT, (TY*)dgi, (TY*)dgh, dh0,     \
-            B, Tseq, n_dir);                                                   \
+            dbhh, B, Tseq, n_dir);                                             \
     } while (0)
 
 #define BWD_CASE(HPV)                                                          \

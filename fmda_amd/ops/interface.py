@@ -78,26 +78,24 @@ class _GRURecurrence(torch.autograd.Function):
         n_dir, threeHp, Hp = w.shape
         d_out = d_out.contiguous().to(gi.dtype)
         d_hlast = d_hlast.contiguous().to(torch.float32)
-        dgi, dgh, _dh0 = ext.gru_bwd(gi, w, bhh32, out, d_out, d_hlast)
+        dgi, dgh, _dh0, dbhh = ext.gru_bwd(gi, w, bhh32, out, d_out, d_hlast)
 
-        # dW_hh[n, k] = sum_{b,t} dGh[b,t,n] * h_prev[b,t,k]  (plain GEMM).
-        # h_prev is the stored bf16 out, shifted by one step in each
-        # direction's processing order (dir 0 forward, dir 1 reversed).
-        dw = torch.empty_like(w, dtype=torch.float32)
-        dbhh = torch.empty_like(bhh32)
+        # dW_hh[n, k] = sum_{b,t} dGh[b,t,n] * h_prev[b,t,k]  (plain MFMA
+        # GEMM in the compute dtype; fp32 accumulation inside rocBLAS).
+        # h_prev is the stored out shifted one step in each direction's
+        # processing order; the t=0 step's h_prev is zero, so the shifted
+        # slices drop it instead of materializing a zero-padded copy.
+        dw = torch.empty_like(w)
         for d in range(n_dir):
             dgh_d = dgh[:, :, d * threeHp:(d + 1) * threeHp]
             out_d = out[:, :, d * Hp:(d + 1) * Hp]
-            h_prev = torch.zeros_like(out_d)
             if d == 0:
-                h_prev[:, 1:] = out_d[:, :-1]
+                dw[d] = torch.einsum("btn,bth->nh", dgh_d[:, 1:],
+                                     out_d[:, :-1])
             else:
-                h_prev[:, :-1] = out_d[:, 1:]
-            dw[d] = torch.matmul(
-                dgh_d.reshape(B * T, threeHp).t().float(),
-                h_prev.reshape(B * T, Hp).float())
-            dbhh[d] = dgh_d.float().sum(dim=(0, 1))
-        return dgi, dw.to(w.dtype), dbhh
+                dw[d] = torch.einsum("btn,bth->nh", dgh_d[:, :-1],
+                                     out_d[:, 1:])
+        return dgi, dw, dbhh
 
 
 def gru_directions(gi: torch.Tensor, w: torch.Tensor, bhh: torch.Tensor
