@@ -1,22 +1,35 @@
 // Persistent biGRU recurrence kernels for MI355X (gfx950, CDNA4).
 //
-// Design (MI355X-first, not a port):
+// Design (MI355X-first, not a port of the reference's cuDNN RNN call at
+// biGRU_model.py:102):
 // - The time-batched input projections gi = x @ W_ih^T + b_ih for BOTH
 //   directions are computed outside (one rocBLAS MFMA GEMM); these kernels
-//   own the sequential part only: per timestep the recurrent GEMM
+//   own the sequential part: per timestep the recurrent GEMM
 //   gh = h_{t-1} @ W_hh^T on MFMA (v_mfma_f32_16x16x32_bf16), the gate
-//   sigmoid/tanh fusion, and the hidden-state update, with the recurrent
-//   weights staged in LDS across all T timesteps (W in LDS for H <= 128;
-//   L2-resident global reads beyond) and h kept in LDS in fp32 with a bf16
-//   shadow for MFMA fragments.
+//   sigmoid/tanh fusion, and the hidden-state update.
 // - One workgroup owns a tile of batch rows for one direction for the whole
-//   sequence: no inter-workgroup communication, no grid sync. Both
-//   directions launch in one grid (blockIdx.y = direction).
+//   sequence: batch rows are independent, so there is no inter-workgroup
+//   communication and no grid sync. Both directions launch in one grid
+//   (blockIdx.y = direction).
+// - Recurrent-weight residency by size: for the hot configs (H = 128) the
+//   MFMA B-fragments of W_hh are HOISTED INTO REGISTERS once and live there
+//   across all T timesteps (HOIST; ~0.4-0.8 KB/lane) - zero W traffic in the
+//   loop. Small sizes keep W in LDS (WLDS); at H >= 256 W streams from the
+//   per-XCD L2.
+// - The next timestep's tiles are PREFETCHED into registers during the MFMA
+//   phase and committed to LDS after the barrier (async-STAGE split,
+//   cdna_hip_programming.md T14), so HBM latency hides under compute.
+// - LDS rows are padded to an odd 16-byte-slot pitch: fragment reads touch
+//   16 distinct rows at one 16-B column slot, which at a power-of-two pitch
+//   is a 16-way bank conflict and at the padded pitch is conflict-free.
 // - The backward (BPTT) kernel recomputes the gates from gi + a fresh
-//   recurrent GEMM (cheaper than materializing activations: the op is
-//   HBM-bound), producing dGi (input-projection grads, consumed by torch
-//   autograd for dW_ih/db_ih/dx) and dGh (recurrent-gate grads, reduced to
-//   dW_hh/db_hh by one rocBLAS GEMM outside).
+//   recurrent GEMM on the stored h (cheaper than materializing activations:
+//   the op is HBM-bound), producing dGi (consumed by torch autograd for
+//   dW_ih/db_ih/dx) and dGh. dGh is stored TIME-SHIFTED so that slot t holds
+//   the gate grads whose h_prev is out[t]: dW_hh then reduces with a single
+//   contiguous rocBLAS GEMM (dgh^T @ out), no sliced copies. db_hh is
+//   accumulated in-kernel (registers across T, one atomicAdd per column per
+//   block).
 // - fp32 instantiations use a plain VALU dot-product path with the same
 //   phase structure and C-tile ownership; they are the on-GPU numerics
 //   oracle, validated against the PyTorch fp32 reference.
@@ -34,7 +47,6 @@
 
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 typedef float f32x4_t __attribute__((ext_vector_type(4)));
-typedef short short8_t __attribute__((ext_vector_type(8)));
 
 namespace fmda {
 
@@ -51,64 +63,69 @@ template <> FMDA_DEV __hip_bfloat16 from_f32<__hip_bfloat16>(float v) {
     return __float2bfloat16(v);
 }
 
-// ---------------------------------------------------------------------------
-// Cooperative tile copy helpers (256 threads), vectorized by 16-byte chunks
-// (row byte counts are multiples of 16 for every supported Hp). Global rows
-// may be masked by `rows_valid`: out-of-range rows are skipped — their LDS
-// content is stale/garbage but never observed in the final output (batch
-// rows are independent in a GRU recurrence).
-// ---------------------------------------------------------------------------
-
 struct alignas(16) chunk16 { unsigned int u[4]; };
 
-// LDS tile [rows][cols] <- global rows at row_stride (elements).
-template <typename T, int ROWS>
+// ---------------------------------------------------------------------------
+// Cooperative tile copy helpers (NT threads), vectorized by 16-byte chunks.
+// Global rows may be masked by `rows_valid`: out-of-range rows are skipped -
+// their LDS content is garbage but never observed (batch rows independent).
+// ---------------------------------------------------------------------------
+
+template <typename T, int ROWS, int NT>
 FMDA_DEV void stage_tile(T* __restrict__ lds, const T* __restrict__ gp,
                          int cols, int lds_pitch, long row_stride,
                          int rows_valid, int tid) {
-    const int cpr = (cols * (int)sizeof(T)) / 16;  // chunks per row
-    const int total = ROWS * cpr;
-    for (int c = tid; c < total; c += 256) {
+    const int cpr = (cols * (int)sizeof(T)) / 16;
+    for (int c = tid; c < ROWS * cpr; c += NT) {
         const int r = c / cpr;
         const int jc = c % cpr;
-        if (r < rows_valid) {
+        if (r < rows_valid)
             ((chunk16*)((char*)lds + (long)r * lds_pitch * sizeof(T)))[jc] =
                 ((const chunk16*)((const char*)gp +
                                   (long)r * row_stride * sizeof(T)))[jc];
-        }
     }
 }
 
-template <typename T, int ROWS>
+template <typename T, int ROWS, int NT>
 FMDA_DEV void store_tile(const T* __restrict__ lds, T* __restrict__ gp,
                          int cols, int lds_pitch, long row_stride,
                          int rows_valid, int tid) {
     const int cpr = (cols * (int)sizeof(T)) / 16;
-    const int total = ROWS * cpr;
-    for (int c = tid; c < total; c += 256) {
+    for (int c = tid; c < ROWS * cpr; c += NT) {
         const int r = c / cpr;
         const int jc = c % cpr;
-        if (r < rows_valid) {
+        if (r < rows_valid)
             ((chunk16*)((char*)gp + (long)r * row_stride * sizeof(T)))[jc] =
                 ((const chunk16*)((const char*)lds +
                                   (long)r * lds_pitch * sizeof(T)))[jc];
-        }
     }
 }
 
-template <typename T, int ROWS>
-FMDA_DEV void zero_tile(T* __restrict__ lds, int cols, int lds_pitch, int tid) {
-    for (int c = tid; c < ROWS * lds_pitch; c += 256) lds[c] = from_f32<T>(0.0f);
-    (void)cols;
+template <typename T, int ROWS, int NT>
+FMDA_DEV void store_zero_tile(T* __restrict__ gp, int cols, long row_stride,
+                              int rows_valid, int tid) {
+    const int cpr = (cols * (int)sizeof(T)) / 16;
+    const chunk16 z = {};
+    for (int c = tid; c < ROWS * cpr; c += NT) {
+        const int r = c / cpr;
+        const int jc = c % cpr;
+        if (r < rows_valid)
+            ((chunk16*)((char*)gp + (long)r * row_stride * sizeof(T)))[jc] = z;
+    }
 }
 
-// fp32 LDS tile += bf16/f32 global tile (used for dh += dOut[t]).
-template <typename T, int ROWS>
+template <typename T, int ROWS, int NT>
+FMDA_DEV void zero_tile(T* __restrict__ lds, int lds_pitch, int tid) {
+    for (int c = tid; c < ROWS * lds_pitch; c += NT)
+        lds[c] = from_f32<T>(0.0f);
+}
+
+// fp32 LDS tile += T-typed global tile (dh += dOut[t]); synchronous form.
+template <typename T, int ROWS, int NT>
 FMDA_DEV void accum_tile_f32(float* __restrict__ lds, const T* __restrict__ gp,
                              int cols, int lds_pitch, long row_stride,
                              int rows_valid, int tid) {
-    const int total = ROWS * cols;
-    for (int c = tid; c < total; c += 256) {
+    for (int c = tid; c < ROWS * cols; c += NT) {
         const int r = c / cols;
         const int j = c % cols;
         if (r < rows_valid)
@@ -116,124 +133,99 @@ FMDA_DEV void accum_tile_f32(float* __restrict__ lds, const T* __restrict__ gp,
     }
 }
 
+// Register prefetch of a tile (async-STAGE split): issue loads in one phase,
+// commit to LDS (or accumulate into an fp32 LDS tile) in a later phase.
+// PC = max 16-byte chunks per thread.
+template <typename T, int ROWS, int NT, int PC>
+struct TilePrefetch {
+    chunk16 v[PC];
+    FMDA_DEV void issue(const T* __restrict__ gp, int cols, long row_stride,
+                        int rows_valid, int tid) {
+        const int cpr = (cols * (int)sizeof(T)) / 16;
+#pragma unroll
+        for (int i = 0; i < PC; ++i) {
+            const int c = tid + i * NT;
+            if (c < ROWS * cpr) {
+                const int r = c / cpr;
+                const int jc = c % cpr;
+                if (r < rows_valid)
+                    v[i] = ((const chunk16*)((const char*)gp +
+                              (long)r * row_stride * sizeof(T)))[jc];
+            }
+        }
+    }
+    FMDA_DEV void commit(T* __restrict__ lds, int cols, int lds_pitch,
+                         int tid) {
+        const int cpr = (cols * (int)sizeof(T)) / 16;
+#pragma unroll
+        for (int i = 0; i < PC; ++i) {
+            const int c = tid + i * NT;
+            if (c < ROWS * cpr) {
+                const int r = c / cpr;
+                const int jc = c % cpr;
+                ((chunk16*)((char*)lds + (long)r * lds_pitch * sizeof(T)))[jc]
+                    = v[i];
+            }
+        }
+    }
+    // lds_f32[r][j] += (float)value for every element of the tile
+    FMDA_DEV void commit_accum_f32(float* __restrict__ lds, int cols,
+                                   int lds_pitch, int rows_valid, int tid) {
+        constexpr int EPC = 16 / (int)sizeof(T);  // elements per chunk
+        const int cpr = cols / EPC;
+#pragma unroll
+        for (int i = 0; i < PC; ++i) {
+            const int c = tid + i * NT;
+            if (c < ROWS * cpr) {
+                const int r = c / cpr;
+                const int jc = c % cpr;
+                if (r < rows_valid) {
+                    const T* e = (const T*)&v[i];
+#pragma unroll
+                    for (int k = 0; k < EPC; ++k)
+                        lds[r * lds_pitch + jc * EPC + k] += to_f32<T>(e[k]);
+                }
+            }
+        }
+    }
+};
+
 // ---------------------------------------------------------------------------
-// MFMA / VALU recurrent GEMM phase.
-//
-// C-tile ownership (v_mfma_f32_16x16x32_bf16 layout; ck_tile
-// warp_gemm_attribute_mfma_impl.hpp M16N16K32 constants):
-//   A (M=16, K=32): lane l holds A[l%16][8*(l/16) + e], e = 0..7 (contiguous)
+// MFMA fragment layout (v_mfma_f32_16x16x32_bf16; ck_tile
+// warp_gemm_attribute_mfma_impl.hpp M16N16K32 constants; verified on GPU by
+// mfma_selftest):
+//   A (M=16, K=32): lane l holds A[l%16][8*(l/16) + e], e = 0..7 contiguous
 //   B (K=32, N=16): lane l holds B[8*(l/16) + e][l%16]
 //   C (16x16):      lane l, reg v holds C[4*(l/16) + v][l%16]
 // The fp32 path computes the same owned C elements with plain dots so the
 // fused gate phase is identical for both dtypes.
 // ---------------------------------------------------------------------------
 
-// Per-wave accumulate of gh tiles for column-tile `ct` of every gate.
-// hb: LDS [BT][Hp] fragment source (bf16 shadow of h, or fp32 h directly).
-// w_row(n) returns pointer to row n (length Hp) of W_hh (LDS or global).
-template <int BT, int Hp>
-FMDA_DEV void gemm_ct_bf16(const __hip_bfloat16* __restrict__ hb, int hb_pitch,
-                           const __hip_bfloat16* __restrict__ wbase,
-                           long w_row_stride, int ct, int lane,
-                           f32x4_t acc[3][BT / 16]) {
-    constexpr int MT = BT / 16;
-    const int arow = lane & 15;
-    const int koff = 8 * (lane >> 4);
-    const int jcol = ct * 16 + (lane & 15);
-#pragma unroll
-    for (int kk = 0; kk < Hp / 32; ++kk) {
-        const int kbase = 32 * kk + koff;
-        bf16x8_t a[MT];
-#pragma unroll
-        for (int m = 0; m < MT; ++m)
-            a[m] = *(const bf16x8_t*)&hb[(16 * m + arow) * hb_pitch + kbase];
-#pragma unroll
-        for (int g = 0; g < 3; ++g) {
-            const long n = g * Hp + jcol;
-            bf16x8_t b = *(const bf16x8_t*)&wbase[n * w_row_stride + kbase];
-#pragma unroll
-            for (int m = 0; m < MT; ++m)
-                acc[g][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a[m], b, acc[g][m], 0, 0, 0);
-        }
-    }
+// B-fragment of W for the gh GEMM (N = gate output column, K = h index).
+template <int Hp>
+FMDA_DEV bf16x8_t load_wfragA(const __hip_bfloat16* __restrict__ w, long pitch,
+                              int ct, int g, int kk, int lane) {
+    const long n = (long)g * Hp + ct * 16 + (lane & 15);
+    const int kbase = 32 * kk + 8 * (lane >> 4);
+    return *(const bf16x8_t*)((const __bf16*)w + n * pitch + kbase);
 }
 
-template <int BT, int Hp>
-FMDA_DEV void gemm_ct_f32(const float* __restrict__ hf, int hb_pitch,
-                          const float* __restrict__ wbase,
-                          long w_row_stride, int ct, int lane,
-                          f32x4_t acc[3][BT / 16]) {
-    constexpr int MT = BT / 16;
+// B-fragment of W for the dh GEMM (K = gate row n, N = h column):
+// element B[n][j] = W[n][j]; per lane 8 rows n at fixed column j.
+template <int Hp>
+FMDA_DEV bf16x8_t load_wfragB(const __hip_bfloat16* __restrict__ w, long pitch,
+                              int ct, int kk, int lane) {
     const int jcol = ct * 16 + (lane & 15);
+    const int nbase = 32 * kk + 8 * (lane >> 4);
+    bf16x8_t b;
+    const __bf16* wb = (const __bf16*)w;
 #pragma unroll
-    for (int g = 0; g < 3; ++g) {
-        const float* wrow = wbase + (long)(g * Hp + jcol) * w_row_stride;
-#pragma unroll
-        for (int m = 0; m < MT; ++m) {
-#pragma unroll
-            for (int e = 0; e < 4; ++e) {
-                const int row = 16 * m + 4 * (lane >> 4) + e;
-                float s = acc[g][m][e];
-                for (int k = 0; k < Hp; ++k)
-                    s += hf[row * hb_pitch + k] * wrow[k];
-                acc[g][m][e] = s;
-            }
-        }
-    }
+    for (int e = 0; e < 8; ++e)
+        b[e] = wb[(long)(nbase + e) * pitch + jcol];
+    return b;
 }
 
-// dh GEMM: dh[b][j] += sum_n dgh[b][n] * W[n][j]  (K = 3*Hp, N = Hp).
-// A = dgh (LDS, row-major, contiguous k). B[k=n][j] = W[n][j]: strided reads.
-template <int BT, int Hp>
-FMDA_DEV void gemm_dh_bf16(const __hip_bfloat16* __restrict__ dgh, int dgh_pitch,
-                           const __hip_bfloat16* __restrict__ wbase,
-                           long w_row_stride, int ct, int lane,
-                           f32x4_t acc[BT / 16]) {
-    constexpr int MT = BT / 16;
-    const int arow = lane & 15;
-    const int koff = 8 * (lane >> 4);
-    const int jcol = ct * 16 + (lane & 15);
-#pragma unroll
-    for (int kk = 0; kk < (3 * Hp) / 32; ++kk) {
-        const int kbase = 32 * kk + koff;
-        bf16x8_t a[MT];
-#pragma unroll
-        for (int m = 0; m < MT; ++m)
-            a[m] = *(const bf16x8_t*)&dgh[(16 * m + arow) * dgh_pitch + kbase];
-        bf16x8_t b;
-        const __bf16* wb = (const __bf16*)wbase;
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-            b[e] = wb[(long)(kbase + e) * w_row_stride + jcol];
-#pragma unroll
-        for (int m = 0; m < MT; ++m)
-            acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[m], b, acc[m],
-                                                             0, 0, 0);
-    }
-}
-
-template <int BT, int Hp>
-FMDA_DEV void gemm_dh_f32(const float* __restrict__ dgh, int dgh_pitch,
-                          const float* __restrict__ wbase,
-                          long w_row_stride, int ct, int lane,
-                          f32x4_t acc[BT / 16]) {
-    constexpr int MT = BT / 16;
-    const int jcol = ct * 16 + (lane & 15);
-#pragma unroll
-    for (int m = 0; m < MT; ++m) {
-#pragma unroll
-        for (int e = 0; e < 4; ++e) {
-            const int row = 16 * m + 4 * (lane >> 4) + e;
-            float s = acc[m][e];
-            for (int n = 0; n < 3 * Hp; ++n)
-                s += dgh[row * dgh_pitch + n] * wbase[(long)n * w_row_stride + jcol];
-            acc[m][e] = s;
-        }
-    }
-}
-
-// ---------------------------------------------------------------------------
+// ===========================================================================
 // Forward kernel.
 //
 // gi:    (B, T, n_dir*3Hp)  input projections incl. b_ih (dtype T)
@@ -241,25 +233,25 @@ FMDA_DEV void gemm_dh_f32(const float* __restrict__ dgh, int dgh_pitch,
 // bhh:   (n_dir, 3Hp)       recurrent bias (fp32)
 // out:   (B, T, n_dir*Hp)   hidden states (dtype T; direction-concat layout)
 // hlast: (n_dir, B, Hp)     final hidden state (fp32)
-// grid:  (ceil(B/BT), n_dir); block: 256 threads; direction 1 runs reversed.
-// ---------------------------------------------------------------------------
-template <typename T, int BT, int Hp, bool WLDS>
-__global__ __launch_bounds__(256) void gru_fwd_kernel(
+// grid:  (ceil(B/BT), n_dir); block NT threads; direction 1 runs reversed.
+// ===========================================================================
+template <typename T, int BT, int Hp, bool WLDS, int NT, bool HOIST>
+__global__ __launch_bounds__(NT) void gru_fwd_kernel(
     const T* __restrict__ gi, const T* __restrict__ w,
     const float* __restrict__ bhh, T* __restrict__ out,
     float* __restrict__ hlast, int B, int Tseq, int n_dir) {
+    constexpr int NW = NT / 64;
     constexpr int MT = BT / 16;
-    constexpr int NCT = Hp / 16;          // column tiles per gate
-    constexpr int CPW = (NCT + 3) / 4;    // column tiles per wave (max)
+    constexpr int NCT = Hp / 16;
+    constexpr int CPW = (NCT + NW - 1) / NW;
     constexpr bool IS_BF16 = !__is_same(T, float);
-    // Padded LDS row pitches (+16 B per row): the fragment reads hit 16
-    // distinct rows at one 16-B column slot; an even power-of-two row pitch
-    // puts the whole lane group on one LDS bank slot (16-way conflict), the
-    // odd-slot pitch spreads it over all 16 slots (conflict-free).
+    constexpr int KK = Hp / 32;
     constexpr int PADE = 16 / (int)sizeof(T);
-    constexpr int WP = Hp + PADE;           // pitch of Hp-wide T rows
-    constexpr int GP3 = 3 * Hp + PADE;      // pitch of 3Hp-wide T rows
-    constexpr int HFP = Hp + 4;             // pitch of Hp-wide fp32 rows
+    constexpr int WP = Hp + PADE;
+    constexpr int GP3 = 3 * Hp + PADE;
+    constexpr int HFP = Hp + 4;
+    constexpr int GI_CHUNKS = (BT * 3 * Hp * (int)sizeof(T)) / 16;
+    constexpr int PC = (GI_CHUNKS + NT - 1) / NT;
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -281,33 +273,60 @@ __global__ __launch_bounds__(256) void gru_fwd_kernel(
 
     const T* wdir = w + (long)dir * 3 * Hp * Hp;
     if (WLDS)
-        stage_tile<T, 3 * Hp>(w_s, wdir, Hp, WP, Hp, 3 * Hp, tid);
-    for (int c = tid; c < 3 * Hp; c += 256)
+        stage_tile<T, 3 * Hp, NT>(w_s, wdir, Hp, WP, Hp, 3 * Hp, tid);
+    for (int c = tid; c < 3 * Hp; c += NT)
         bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
-    for (int c = tid; c < BT * HFP; c += 256) hf_s[c] = 0.0f;
-    if (IS_BF16)
-        for (int c = tid; c < BT * WP; c += 256)
-            hb_s[c] = __float2bfloat16(0.0f);
+    zero_tile<float, BT, NT>(hf_s, HFP, tid);
+    if (IS_BF16) zero_tile<__hip_bfloat16, BT, NT>(hb_s, WP, tid);
 
-    const long gi_row = (long)Tseq * n_dir * 3 * Hp;   // per batch row
+    const long gi_row = (long)Tseq * n_dir * 3 * Hp;
     const long out_row = (long)Tseq * n_dir * Hp;
     const T* gi_b = gi + (long)b0 * gi_row + (long)dir * 3 * Hp;
     T* out_b = out + (long)b0 * out_row + (long)dir * Hp;
 
-    // stage gi for step 0
-    {
+    {   // stage gi for step 0 synchronously
         const int tt = rev ? (Tseq - 1) : 0;
-        stage_tile<T, BT>(gi_s, gi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp, GP3,
-                          gi_row, rows_valid, tid);
+        stage_tile<T, BT, NT>(gi_s, gi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp,
+                              GP3, gi_row, rows_valid, tid);
+    }
+
+    const T* wfrag = WLDS ? w_s : wdir;
+    const long wstride = WLDS ? WP : Hp;
+
+    // Hoist W B-fragments into registers for the whole sequence (bf16 only).
+    bf16x8_t wA[HOIST ? CPW : 1][3][HOIST ? KK : 1];
+    if constexpr (HOIST && IS_BF16) {
+#pragma unroll
+        for (int i = 0; i < CPW; ++i) {
+            const int ct = wave + NW * i;
+            if (ct >= NCT) continue;
+#pragma unroll
+            for (int g = 0; g < 3; ++g)
+#pragma unroll
+                for (int kk = 0; kk < KK; ++kk)
+                    wA[i][g][kk] = load_wfragA<Hp>(
+                        (const __hip_bfloat16*)wfrag, wstride, ct, g, kk,
+                        lane);
+        }
     }
     __syncthreads();
 
-    const T* wfrag = WLDS ? w_s : wdir;
-    const long wstride = WLDS ? WP : Hp;  // row stride in elements
+    TilePrefetch<T, BT, NT, PC> pf;
 
     for (int u = 0; u < Tseq; ++u) {
         const int tt = rev ? (Tseq - 1 - u) : u;
-        // ---- phase A: recurrent GEMM gh = h @ W^T (per-wave column tiles)
+        // keep the L2-streamed W pointer opaque so the compiler does not
+        // hoist every fragment load out of the loop and spill
+        const T* wdyn = wfrag;
+        if constexpr (!WLDS && !HOIST && IS_BF16)
+            asm volatile("" : "+s"(wdyn));
+
+        // ---- phase A: issue next-step gi prefetch, then recurrent GEMM
+        if (u + 1 < Tseq) {
+            const int ttn = rev ? (Tseq - 2 - u) : (u + 1);
+            pf.issue(gi_b + (long)ttn * n_dir * 3 * Hp, 3 * Hp, gi_row,
+                     rows_valid, tid);
+        }
         f32x4_t acc[CPW][3][MT];
 #pragma unroll
         for (int i = 0; i < CPW; ++i)
@@ -315,16 +334,53 @@ __global__ __launch_bounds__(256) void gru_fwd_kernel(
             for (int g = 0; g < 3; ++g)
 #pragma unroll
                 for (int m = 0; m < MT; ++m) acc[i][g][m] = f32x4_t{0.f};
+
 #pragma unroll
         for (int i = 0; i < CPW; ++i) {
-            const int ct = wave + 4 * i;
-            if (ct < NCT) {
-                if constexpr (IS_BF16)
-                    gemm_ct_bf16<BT, Hp>(hb_s, WP, wfrag, wstride, ct, lane,
-                                         acc[i]);
-                else
-                    gemm_ct_f32<BT, Hp>(hf_s, HFP, wfrag, wstride, ct, lane,
-                                        acc[i]);
+            const int ct = wave + NW * i;
+            if (ct >= NCT) continue;
+            if constexpr (IS_BF16) {
+                const int arow = lane & 15;
+                const int koff = 8 * (lane >> 4);
+#pragma unroll
+                for (int kk = 0; kk < KK; ++kk) {
+                    const int kbase = 32 * kk + koff;
+                    bf16x8_t a[MT];
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+                        a[m] = *(const bf16x8_t*)&(
+                            (const __bf16*)hb_s)[(16 * m + arow) * WP + kbase];
+#pragma unroll
+                    for (int g = 0; g < 3; ++g) {
+                        const bf16x8_t b =
+                            HOIST ? wA[i][g][kk]
+                                  : load_wfragA<Hp>(
+                                        (const __hip_bfloat16*)wdyn, wstride,
+                                        ct, g, kk, lane);
+#pragma unroll
+                        for (int m = 0; m < MT; ++m)
+                            acc[i][g][m] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    a[m], b, acc[i][g][m], 0, 0, 0);
+                    }
+                }
+            } else {
+                const int jcol = ct * 16 + (lane & 15);
+#pragma unroll
+                for (int g = 0; g < 3; ++g) {
+                    const float* wrow = (const float*)wdyn +
+                                        (long)(g * Hp + jcol) * wstride;
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+#pragma unroll
+                        for (int e = 0; e < 4; ++e) {
+                            const int row = 16 * m + 4 * (lane >> 4) + e;
+                            float s = acc[i][g][m][e];
+                            for (int k = 0; k < Hp; ++k)
+                                s += hf_s[row * HFP + k] * wrow[k];
+                            acc[i][g][m][e] = s;
+                        }
+                }
             }
         }
         __syncthreads();  // all GEMM reads of h done
@@ -332,7 +388,7 @@ __global__ __launch_bounds__(256) void gru_fwd_kernel(
         // ---- phase B: fused gates + h update (owned (b, j) elements)
 #pragma unroll
         for (int i = 0; i < CPW; ++i) {
-            const int ct = wave + 4 * i;
+            const int ct = wave + NW * i;
             if (ct >= NCT) continue;
             const int j = ct * 16 + (lane & 15);
 #pragma unroll
@@ -359,53 +415,55 @@ __global__ __launch_bounds__(256) void gru_fwd_kernel(
         }
         __syncthreads();  // h_t complete; gi_s free
 
-        // ---- phase C: write out[t]; stage gi for t+1
+        // ---- phase C: write out[t]; commit prefetched gi
         if constexpr (IS_BF16) {
-            store_tile<__hip_bfloat16, BT>(
+            store_tile<__hip_bfloat16, BT, NT>(
                 hb_s, (__hip_bfloat16*)(out_b + (long)tt * n_dir * Hp), Hp, WP,
                 out_row, rows_valid, tid);
         } else {
-            store_tile<float, BT>(hf_s, (float*)(out_b + (long)tt * n_dir * Hp),
-                                  Hp, HFP, out_row, rows_valid, tid);
+            store_tile<float, BT, NT>(
+                hf_s, (float*)(out_b + (long)tt * n_dir * Hp), Hp, HFP,
+                out_row, rows_valid, tid);
         }
-        if (u + 1 < Tseq) {
-            const int ttn = rev ? (Tseq - 2 - u) : (u + 1);
-            stage_tile<T, BT>(gi_s, gi_b + (long)ttn * n_dir * 3 * Hp, 3 * Hp,
-                              GP3, gi_row, rows_valid, tid);
-        }
+        if (u + 1 < Tseq) pf.commit(gi_s, 3 * Hp, GP3, tid);
         __syncthreads();
     }
 
     // final hidden state (fp32)
     float* hl = hlast + ((long)dir * B + b0) * Hp;
-    for (int c = tid; c < BT * Hp; c += 256) {
+    for (int c = tid; c < BT * Hp; c += NT) {
         const int r = c / Hp;
-        if (r < rows_valid) hl[(long)r * Hp + (c % Hp)] = hf_s[r * HFP + (c % Hp)];
+        if (r < rows_valid)
+            hl[(long)r * Hp + (c % Hp)] = hf_s[r * HFP + (c % Hp)];
     }
 }
 
-// ---------------------------------------------------------------------------
-// Backward (BPTT) kernel.
-//
-// Recomputes gates from gi + fresh recurrent GEMM on the stored h (out),
-// produces dGi and dGh, and carries dh in fp32 LDS across timesteps.
-// dW_hh/db_hh are reduced outside from dGh (plain GEMM).
-// ---------------------------------------------------------------------------
-template <typename T, int BT, int Hp, bool WLDS>
-__global__ __launch_bounds__(256) void gru_bwd_kernel(
+// ===========================================================================
+// Backward (BPTT) kernel. See the header comment for the dGh time-shift and
+// the in-kernel db_hh reduction.
+// ===========================================================================
+template <typename T, int BT, int Hp, bool WLDS, int NT, bool HOIST>
+__global__ __launch_bounds__(NT) void gru_bwd_kernel(
     const T* __restrict__ gi, const T* __restrict__ w,
     const float* __restrict__ bhh, const T* __restrict__ out,
     const T* __restrict__ dout, const float* __restrict__ dhT,
     T* __restrict__ dgi, T* __restrict__ dgh, float* __restrict__ dh0,
     float* __restrict__ dbhh, int B, int Tseq, int n_dir) {
+    constexpr int NW = NT / 64;
     constexpr int MT = BT / 16;
     constexpr int NCT = Hp / 16;
-    constexpr int CPW = (NCT + 3) / 4;
+    constexpr int CPW = (NCT + NW - 1) / NW;
     constexpr bool IS_BF16 = !__is_same(T, float);
+    constexpr int KK = Hp / 32;
+    constexpr int KK2 = (3 * Hp) / 32;
     constexpr int PADE = 16 / (int)sizeof(T);
     constexpr int WP = Hp + PADE;
     constexpr int GP3 = 3 * Hp + PADE;
     constexpr int HFP = Hp + 4;
+    constexpr int GI_CHUNKS = (BT * 3 * Hp * (int)sizeof(T)) / 16;
+    constexpr int H_CHUNKS = (BT * Hp * (int)sizeof(T)) / 16;
+    constexpr int PCG = (GI_CHUNKS + NT - 1) / NT;
+    constexpr int PCH = (H_CHUNKS + NT - 1) / NT;
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -420,15 +478,15 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
     T* w_s = nullptr;
     if (WLDS) { w_s = (T*)p; p += sizeof(T) * 3 * Hp * WP; }
     float* dh_s = (float*)p; p += sizeof(float) * BT * HFP;
-    T* hb_s = (T*)p; p += sizeof(T) * BT * WP;      // h_{t-1} (from out)
-    T* gi_s = (T*)p; p += sizeof(T) * BT * GP3;  // gi[t], then overwritten by dGi
+    T* hb_s = (T*)p; p += sizeof(T) * BT * WP;
+    T* gi_s = (T*)p; p += sizeof(T) * BT * GP3;   // gi[t] -> dGi in place
     T* dgh_s = (T*)p; p += sizeof(T) * BT * GP3;
     float* bhh_s = (float*)p;
 
     const T* wdir = w + (long)dir * 3 * Hp * Hp;
     if (WLDS)
-        stage_tile<T, 3 * Hp>(w_s, wdir, Hp, WP, Hp, 3 * Hp, tid);
-    for (int c = tid; c < 3 * Hp; c += 256)
+        stage_tile<T, 3 * Hp, NT>(w_s, wdir, Hp, WP, Hp, 3 * Hp, tid);
+    for (int c = tid; c < 3 * Hp; c += NT)
         bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
 
     const long gi_row = (long)Tseq * n_dir * 3 * Hp;
@@ -439,61 +497,140 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
     const T* out_b = out + (long)b0 * out_row + (long)dir * Hp;
     const T* dout_b = dout + (long)b0 * out_row + (long)dir * Hp;
 
-    // dh carry init: dh = dhT; then += dOut[t] as each step is staged.
-    {
+    {   // dh carry init: dh = dhT
         const float* hT = dhT + ((long)dir * B + b0) * Hp;
-        for (int c = tid; c < BT * Hp; c += 256) {
+        for (int c = tid; c < BT * Hp; c += NT) {
             const int r = c / Hp;
             dh_s[r * HFP + (c % Hp)] =
                 (r < rows_valid) ? hT[(long)r * Hp + (c % Hp)] : 0.0f;
         }
     }
 
-    // Stage step u = Tseq-1: gi[tt], h_prev = out[tt_prev] or 0, dh += dout[tt]
-    {
+    {   // stage step u = Tseq-1 synchronously
         const int u = Tseq - 1;
         const int tt = rev ? (Tseq - 1 - u) : u;
-        stage_tile<T, BT>(gi_s, gi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp, GP3,
-                          gi_row, rows_valid, tid);
+        stage_tile<T, BT, NT>(gi_s, gi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp,
+                              GP3, gi_row, rows_valid, tid);
         if (u > 0) {
             const int ttp = rev ? (Tseq - u) : (u - 1);
-            stage_tile<T, BT>(hb_s, out_b + (long)ttp * n_dir * Hp, Hp, WP,
-                              out_row, rows_valid, tid);
+            stage_tile<T, BT, NT>(hb_s, out_b + (long)ttp * n_dir * Hp, Hp, WP,
+                                  out_row, rows_valid, tid);
         } else {
-            zero_tile<T, BT>(hb_s, Hp, WP, tid);
+            zero_tile<T, BT, NT>(hb_s, WP, tid);
         }
-        accum_tile_f32<T, BT>(dh_s, dout_b + (long)tt * n_dir * Hp, Hp, HFP,
-                              out_row, rows_valid, tid);
+        accum_tile_f32<T, BT, NT>(dh_s, dout_b + (long)tt * n_dir * Hp, Hp,
+                                  HFP, out_row, rows_valid, tid);
     }
-    __syncthreads();
 
     const T* wfrag = WLDS ? w_s : wdir;
     const long wstride = WLDS ? WP : Hp;
-    // per-lane db_hh partials accumulated across all T (summed over b rows)
+
+    bf16x8_t wA[HOIST ? CPW : 1][3][HOIST ? KK : 1];
+    bf16x8_t wB[HOIST ? CPW : 1][HOIST ? KK2 : 1];
+    if constexpr (HOIST && IS_BF16) {
+#pragma unroll
+        for (int i = 0; i < CPW; ++i) {
+            const int ct = wave + NW * i;
+            if (ct >= NCT) continue;
+#pragma unroll
+            for (int g = 0; g < 3; ++g)
+#pragma unroll
+                for (int kk = 0; kk < KK; ++kk)
+                    wA[i][g][kk] = load_wfragA<Hp>(
+                        (const __hip_bfloat16*)wfrag, wstride, ct, g, kk, lane);
+#pragma unroll
+            for (int kk = 0; kk < KK2; ++kk)
+                wB[i][kk] = load_wfragB<Hp>((const __hip_bfloat16*)wfrag,
+                                            wstride, ct, kk, lane);
+        }
+    }
+    __syncthreads();
+
     float dbacc[CPW][3];
 #pragma unroll
     for (int i = 0; i < CPW; ++i)
 #pragma unroll
         for (int g = 0; g < 3; ++g) dbacc[i][g] = 0.0f;
 
+    TilePrefetch<T, BT, NT, PCG> pf_gi;
+    TilePrefetch<T, BT, NT, PCH> pf_hb;
+    TilePrefetch<T, BT, NT, PCH> pf_do;
+
     for (int u = Tseq - 1; u >= 0; --u) {
         const int tt = rev ? (Tseq - 1 - u) : u;
+        const bool have_next = (u > 0);
+        const int un = u - 1;
+        const int ttn = rev ? (Tseq - 1 - un) : un;
+        const T* wdyn = wfrag;
+        if constexpr (!WLDS && !HOIST && IS_BF16)
+            asm volatile("" : "+s"(wdyn));
+
+        if (have_next) {
+            pf_gi.issue(gi_b + (long)ttn * n_dir * 3 * Hp, 3 * Hp, gi_row,
+                        rows_valid, tid);
+            if (un > 0) {
+                const int ttp = rev ? (Tseq - un) : (un - 1);
+                pf_hb.issue(out_b + (long)ttp * n_dir * Hp, Hp, out_row,
+                            rows_valid, tid);
+            }
+            pf_do.issue(dout_b + (long)ttn * n_dir * Hp, Hp, out_row,
+                        rows_valid, tid);
+        }
 
         // ---- phase A: recompute gh, fuse gate grads (per owned column tile)
 #pragma unroll
         for (int i = 0; i < CPW; ++i) {
-            const int ct = wave + 4 * i;
+            const int ct = wave + NW * i;
             if (ct >= NCT) continue;
             f32x4_t acc[3][MT];
 #pragma unroll
             for (int g = 0; g < 3; ++g)
 #pragma unroll
                 for (int m = 0; m < MT; ++m) acc[g][m] = f32x4_t{0.f};
-            if constexpr (IS_BF16)
-                gemm_ct_bf16<BT, Hp>(hb_s, WP, wfrag, wstride, ct, lane, acc);
-            else
-                gemm_ct_f32<BT, Hp>((const float*)hb_s, WP, wfrag, wstride, ct,
-                                    lane, acc);
+            if constexpr (IS_BF16) {
+                const int arow = lane & 15;
+                const int koff = 8 * (lane >> 4);
+#pragma unroll
+                for (int kk = 0; kk < KK; ++kk) {
+                    const int kbase = 32 * kk + koff;
+                    bf16x8_t a[MT];
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+                        a[m] = *(const bf16x8_t*)&(
+                            (const __bf16*)hb_s)[(16 * m + arow) * WP + kbase];
+#pragma unroll
+                    for (int g = 0; g < 3; ++g) {
+                        const bf16x8_t b =
+                            HOIST ? wA[i][g][kk]
+                                  : load_wfragA<Hp>(
+                                        (const __hip_bfloat16*)wdyn, wstride,
+                                        ct, g, kk, lane);
+#pragma unroll
+                        for (int m = 0; m < MT; ++m)
+                            acc[g][m] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    a[m], b, acc[g][m], 0, 0, 0);
+                    }
+                }
+            } else {
+                const int jcol = ct * 16 + (lane & 15);
+#pragma unroll
+                for (int g = 0; g < 3; ++g) {
+                    const float* wrow = (const float*)wdyn +
+                                        (long)(g * Hp + jcol) * wstride;
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+#pragma unroll
+                        for (int e = 0; e < 4; ++e) {
+                            const int row = 16 * m + 4 * (lane >> 4) + e;
+                            float s = acc[g][m][e];
+                            for (int k = 0; k < Hp; ++k)
+                                s += to_f32<T>(hb_s[row * WP + k]) * wrow[k];
+                            acc[g][m][e] = s;
+                        }
+                }
+            }
+
             const int j = ct * 16 + (lane & 15);
 #pragma unroll
             for (int m = 0; m < MT; ++m) {
@@ -512,21 +649,19 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
                     const float hprev = to_f32<T>(hb_s[b * WP + j]);
                     const float dht = dh_s[b * HFP + j];
                     const bool live = (b < rows_valid);
-                    const float dz_pre = live ? dht * (hprev - n) * z * (1.0f - z) : 0.0f;
-                    const float dn_pre = live ? dht * (1.0f - z) * (1.0f - n * n) : 0.0f;
+                    const float dz_pre =
+                        live ? dht * (hprev - n) * z * (1.0f - z) : 0.0f;
+                    const float dn_pre =
+                        live ? dht * (1.0f - z) * (1.0f - n * n) : 0.0f;
                     const float dr_pre = dn_pre * hn * r * (1.0f - r);
                     const float dhn = dn_pre * r;
-                    // dGi (packed r,z,n pre-activation grads) -> reuse gi_s
                     gi_s[b * GP3 + j] = from_f32<T>(dr_pre);
                     gi_s[b * GP3 + Hp + j] = from_f32<T>(dz_pre);
                     gi_s[b * GP3 + 2 * Hp + j] = from_f32<T>(dn_pre);
-                    // dGh differs in the n block only
                     dgh_s[b * GP3 + j] = from_f32<T>(dr_pre);
                     dgh_s[b * GP3 + Hp + j] = from_f32<T>(dz_pre);
                     dgh_s[b * GP3 + 2 * Hp + j] = from_f32<T>(dhn);
-                    // direct part of dh_{t-1}; GEMM part added in phase B
                     dh_s[b * HFP + j] = dht * z;
-                    // db_hh partials (summed in fp32 across b and t)
                     dbacc[i][0] += dr_pre;
                     dbacc[i][1] += dz_pre;
                     dbacc[i][2] += dhn;
@@ -538,17 +673,46 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
         // ---- phase B: dh_{t-1} += dGh @ W
 #pragma unroll
         for (int i = 0; i < CPW; ++i) {
-            const int ct = wave + 4 * i;
+            const int ct = wave + NW * i;
             if (ct >= NCT) continue;
             f32x4_t acc2[MT];
 #pragma unroll
             for (int m = 0; m < MT; ++m) acc2[m] = f32x4_t{0.f};
-            if constexpr (IS_BF16)
-                gemm_dh_bf16<BT, Hp>((const __hip_bfloat16*)dgh_s, GP3, wfrag,
-                                     wstride, ct, lane, acc2);
-            else
-                gemm_dh_f32<BT, Hp>((const float*)dgh_s, GP3, wfrag, wstride,
-                                    ct, lane, acc2);
+            if constexpr (IS_BF16) {
+                const int arow = lane & 15;
+                const int koff = 8 * (lane >> 4);
+#pragma unroll
+                for (int kk = 0; kk < KK2; ++kk) {
+                    const int kbase = 32 * kk + koff;
+                    bf16x8_t a[MT];
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+                        a[m] = *(const bf16x8_t*)&(
+                            (const __bf16*)dgh_s)[(16 * m + arow) * GP3 +
+                                                  kbase];
+                    const bf16x8_t b =
+                        HOIST ? wB[i][kk]
+                              : load_wfragB<Hp>((const __hip_bfloat16*)wdyn,
+                                                wstride, ct, kk, lane);
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+                        acc2[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[m], b, acc2[m], 0, 0, 0);
+                }
+            } else {
+                const int jcol = ct * 16 + (lane & 15);
+#pragma unroll
+                for (int m = 0; m < MT; ++m)
+#pragma unroll
+                    for (int e = 0; e < 4; ++e) {
+                        const int row = 16 * m + 4 * (lane >> 4) + e;
+                        float s = acc2[m][e];
+                        for (int n = 0; n < 3 * Hp; ++n)
+                            s += to_f32<T>(dgh_s[row * GP3 + n]) *
+                                 ((const float*)wdyn)[(long)n * wstride + jcol];
+                        acc2[m][e] = s;
+                    }
+            }
             const int j = ct * 16 + (lane & 15);
 #pragma unroll
             for (int m = 0; m < MT; ++m)
@@ -560,41 +724,44 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
         }
         __syncthreads();
 
-        // ---- phase C: write dGi/dGh; stage next step
-        store_tile<T, BT>(gi_s, dgi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp, GP3,
-                          gi_row, rows_valid, tid);
-        store_tile<T, BT>(dgh_s, dgh_b + (long)tt * n_dir * 3 * Hp, 3 * Hp,
-                          GP3, gi_row, rows_valid, tid);
-        if (u > 0) {
-            const int un = u - 1;
-            const int ttn = rev ? (Tseq - 1 - un) : un;
-            stage_tile<T, BT>(gi_s, gi_b + (long)ttn * n_dir * 3 * Hp, 3 * Hp,
+        // ---- phase C: write dGi and time-shifted dGh; commit prefetches
+        store_tile<T, BT, NT>(gi_s, dgi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp,
                               GP3, gi_row, rows_valid, tid);
-            if (un > 0) {
-                const int ttp = rev ? (Tseq - un) : (un - 1);
-                stage_tile<T, BT>(hb_s, out_b + (long)ttp * n_dir * Hp, Hp, WP,
-                                  out_row, rows_valid, tid);
-            } else {
-                zero_tile<T, BT>(hb_s, Hp, WP, tid);
-            }
-            accum_tile_f32<T, BT>(dh_s, dout_b + (long)ttn * n_dir * Hp, Hp,
-                                  HFP, out_row, rows_valid, tid);
+        {
+            const int slot = rev ? (tt + 1) : (tt - 1);
+            if (slot >= 0 && slot < Tseq)
+                store_tile<T, BT, NT>(dgh_s,
+                                      dgh_b + (long)slot * n_dir * 3 * Hp,
+                                      3 * Hp, GP3, gi_row, rows_valid, tid);
+            else
+                store_zero_tile<T, BT, NT>(
+                    dgh_b + (long)(rev ? 0 : (Tseq - 1)) * n_dir * 3 * Hp,
+                    3 * Hp, gi_row, rows_valid, tid);
+        }
+        if (have_next) {
+            pf_gi.commit(gi_s, 3 * Hp, GP3, tid);
+            if (un > 0)
+                pf_hb.commit(hb_s, Hp, WP, tid);
+            else
+                zero_tile<T, BT, NT>(hb_s, WP, tid);
+            pf_do.commit_accum_f32(dh_s, Hp, HFP, rows_valid, tid);
         }
         __syncthreads();
     }
 
     // dh0 (fp32)
     float* d0 = dh0 + ((long)dir * B + b0) * Hp;
-    for (int c = tid; c < BT * Hp; c += 256) {
+    for (int c = tid; c < BT * Hp; c += NT) {
         const int r = c / Hp;
-        if (r < rows_valid) d0[(long)r * Hp + (c % Hp)] = dh_s[r * HFP + (c % Hp)];
+        if (r < rows_valid)
+            d0[(long)r * Hp + (c % Hp)] = dh_s[r * HFP + (c % Hp)];
     }
 
     // db_hh: reduce the 4 row-quarter lanes (same j = lane&15), then one
-    // atomicAdd per (gate, j) per block.
+    // atomicAdd per (gate, column) per block.
 #pragma unroll
     for (int i = 0; i < CPW; ++i) {
-        const int ct = wave + 4 * i;
+        const int ct = wave + NW * i;
         if (ct >= NCT) continue;
 #pragma unroll
         for (int g = 0; g < 3; ++g) {
@@ -608,11 +775,9 @@ __global__ __launch_bounds__(256) void gru_bwd_kernel(
     }
 }
 
-// ---------------------------------------------------------------------------
-// MFMA layout self-test: C (16x16) = A (16x32) @ B (32x16) in bf16 with the
-// fragment layout assumed above. A GPU test compares it against torch matmul
-// so a layout regression fails loudly instead of silently transposing.
-// ---------------------------------------------------------------------------
+// ===========================================================================
+// MFMA layout self-test.
+// ===========================================================================
 __global__ void mfma_selftest_kernel(const __hip_bfloat16* __restrict__ A,
                                      const __hip_bfloat16* __restrict__ Bm,
                                      float* __restrict__ C) {
@@ -622,8 +787,8 @@ __global__ void mfma_selftest_kernel(const __hip_bfloat16* __restrict__ A,
     const __bf16* bp = (const __bf16*)Bm;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
-        a[e] = ap[(lane & 15) * 32 + 8 * (lane >> 4) + e];   // A[i][k] row-major
-        b[e] = bp[(8 * (lane >> 4) + e) * 16 + (lane & 15)]; // B[k][j] row-major
+        a[e] = ap[(lane & 15) * 32 + 8 * (lane >> 4) + e];
+        b[e] = bp[(8 * (lane >> 4) + e) * 16 + (lane & 15)];
     }
     f32x4_t acc = f32x4_t{0.f};
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
@@ -632,49 +797,45 @@ __global__ void mfma_selftest_kernel(const __hip_bfloat16* __restrict__ A,
         C[(4 * (lane >> 4) + v) * 16 + (lane & 15)] = acc[v];
 }
 
-}  // namespace fmda
+// ===========================================================================
+// Host launchers.
+// ===========================================================================
 
-// ---------------------------------------------------------------------------
-// Host launchers (exported; bound to torch in bindings.cpp).
-// ---------------------------------------------------------------------------
+struct LaunchCfg { int bt; bool wlds; int nt; bool hoist; };
 
-namespace fmda {
-
-struct LaunchCfg { int bt; bool wlds; };
-
-// Tile/LDS policy per (dtype, Hp): chosen so total LDS stays under 160 KiB
-// with the recurrent weights resident whenever they fit.
+// Tile/LDS/residency policy per (dtype, Hp); see header comment.
 static inline LaunchCfg fwd_cfg(bool bf16, int Hp) {
     if (bf16) {
-        if (Hp <= 128) return {32, true};
-        if (Hp == 256) return {32, false};
-        return {16, false};
+        if (Hp <= 64) return {32, true, 256, false};
+        if (Hp == 128) return {32, false, 512, true};
+        if (Hp == 256) return {32, false, 512, false};
+        return {16, false, 512, false};
     }
-    if (Hp <= 128) return {32, false};
-    return {16, false};
+    if (Hp <= 128) return {32, false, 256, false};
+    return {16, false, 256, false};
 }
 static inline LaunchCfg bwd_cfg(bool bf16, int Hp) {
     if (bf16) {
-        if (Hp <= 64) return {32, true};
-        if (Hp == 128) return {16, true};
-        if (Hp == 256) return {32, false};
-        return {16, false};
+        if (Hp <= 64) return {32, true, 256, false};
+        if (Hp == 128) return {32, false, 512, true};
+        if (Hp == 256) return {32, false, 512, false};
+        return {16, false, 512, false};
     }
-    if (Hp <= 128) return {32, false};
-    return {16, false};
+    if (Hp <= 128) return {32, false, 256, false};
+    return {16, false, 256, false};
 }
 
-// NOTE: must mirror the in-kernel padded pitches (WP/GP3/HFP).
+// Mirrors the in-kernel padded pitches (WP/GP3/HFP).
 static inline size_t fwd_lds_bytes(bool bf16, int Hp, int bt, bool wlds) {
     const size_t es = bf16 ? 2 : 4;
     const size_t pade = 16 / es;
     const size_t wp = Hp + pade, gp3 = 3 * Hp + pade, hfp = Hp + 4;
     size_t s = 0;
     if (wlds) s += es * 3 * Hp * wp;
-    s += 4 * (size_t)bt * hfp;             // hf_s
-    if (bf16) s += 2 * (size_t)bt * wp;    // hb_s
-    s += es * (size_t)bt * gp3;            // gi_s
-    s += 4 * (size_t)3 * Hp;               // bhh_s
+    s += 4 * (size_t)bt * hfp;
+    if (bf16) s += 2 * (size_t)bt * wp;
+    s += es * (size_t)bt * gp3;
+    s += 4 * (size_t)3 * Hp;
     return s;
 }
 static inline size_t bwd_lds_bytes(bool bf16, int Hp, int bt, bool wlds) {
@@ -683,80 +844,76 @@ static inline size_t bwd_lds_bytes(bool bf16, int Hp, int bt, bool wlds) {
     const size_t wp = Hp + pade, gp3 = 3 * Hp + pade, hfp = Hp + 4;
     size_t s = 0;
     if (wlds) s += es * 3 * Hp * wp;
-    s += 4 * (size_t)bt * hfp;             // dh_s
-    s += es * (size_t)bt * wp;             // hb_s
-    s += 2 * es * (size_t)bt * gp3;        // gi_s + dgh_s
-    s += 4 * (size_t)3 * Hp;               // bhh_s
+    s += 4 * (size_t)bt * hfp;
+    s += es * (size_t)bt * wp;
+    s += 2 * es * (size_t)bt * gp3;
+    s += 4 * (size_t)3 * Hp;
     return s;
 }
 
-#define FMDA_DISPATCH_HP(HP_VAL, FN)                                          \
-    switch (HP_VAL) {                                                         \
-        case 16: FN(16); break;                                               \
-        case 32: FN(32); break;                                               \
-        case 64: FN(64); break;                                               \
-        case 128: FN(128); break;                                             \
-        case 256: FN(256); break;                                             \
-        case 512: FN(512); break;                                             \
-        default: return -1;                                                   \
-    }
+template <typename T, int BT, int Hp, bool WLDS, int NT, bool HOIST>
+static int launch_fwd(const void* gi, const void* w, const float* bhh,
+                      void* out, float* hlast, int B, int Tseq, int n_dir,
+                      size_t lds, hipStream_t stream) {
+    auto k = gru_fwd_kernel<T, BT, Hp, WLDS, NT, HOIST>;
+    (void)hipFuncSetAttribute((const void*)k,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    const dim3 grid((B + BT - 1) / BT, n_dir);
+    k<<<grid, NT, lds, stream>>>((const T*)gi, (const T*)w, bhh, (T*)out,
+                                 hlast, B, Tseq, n_dir);
+    return 0;
+}
+
+template <typename T, int BT, int Hp, bool WLDS, int NT, bool HOIST>
+static int launch_bwd(const void* gi, const void* w, const float* bhh,
+                      const void* out, const void* dout, const float* dhT,
+                      void* dgi, void* dgh, float* dh0, float* dbhh, int B,
+                      int Tseq, int n_dir, size_t lds, hipStream_t stream) {
+    auto k = gru_bwd_kernel<T, BT, Hp, WLDS, NT, HOIST>;
+    (void)hipFuncSetAttribute((const void*)k,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    const dim3 grid((B + BT - 1) / BT, n_dir);
+    k<<<grid, NT, lds, stream>>>((const T*)gi, (const T*)w, bhh,
+                                 (const T*)out, (const T*)dout, dhT, (T*)dgi,
+                                 (T*)dgh, dh0, dbhh, B, Tseq, n_dir);
+    return 0;
+}
+
+using bf16_t = __hip_bfloat16;
 
 extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const float* bhh, void* out,
                                    float* hlast, int B, int Tseq, int n_dir,
                                    hipStream_t stream) {
-    const LaunchCfg cfg = fwd_cfg(is_bf16, Hp);
-    const size_t lds = fwd_lds_bytes(is_bf16, Hp, cfg.bt, cfg.wlds);
+    const LaunchCfg c = fwd_cfg(is_bf16, Hp);
+    const size_t lds = fwd_lds_bytes(is_bf16, Hp, c.bt, c.wlds);
     if (lds > 160 * 1024) return -2;
-    const dim3 grid((B + cfg.bt - 1) / cfg.bt, n_dir);
-    hipError_t err = hipSuccess;
-
-#define FWD_CASE(HPV)                                                          \
-    do {                                                                       \
-        if (is_bf16) {                                                         \
-            if (cfg.bt == 32 && cfg.wlds) {                                    \
-                auto k = gru_fwd_kernel<__hip_bfloat16, 32, HPV, true>;        \
-                hipFuncSetAttribute((const void*)k,                            \
-                    hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);     \
-                k<<<grid, 256, lds, stream>>>((const __hip_bfloat16*)gi,       \
-                    (const __hip_bfloat16*)w, bhh, (__hip_bfloat16*)out,       \
-                    hlast, B, Tseq, n_dir);                                    \
-            } else if (cfg.bt == 32) {                                         \
-                auto k = gru_fwd_kernel<__hip_bfloat16, 32, HPV, false>;       \
-                hipFuncSetAttribute((const void*)k,                            \
-                    hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);     \
-                k<<<grid, 256, lds, stream>>>((const __hip_bfloat16*)gi,       \
-                    (const __hip_bfloat16*)w, bhh, (__hip_bfloat16*)out,       \
-                    hlast, B, Tseq, n_dir);                                    \
-            } else {                                                           \
-                auto k = gru_fwd_kernel<__hip_bfloat16, 16, HPV, false>;       \
-                hipFuncSetAttribute((const void*)k,                            \
-                    hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);     \
-                k<<<grid, 256, lds, stream>>>((const __hip_bfloat16*)gi,       \
-                    (const __hip_bfloat16*)w, bhh, (__hip_bfloat16*)out,       \
-                    hlast, B, Tseq, n_dir);                                    \
-            }                                                                  \
-        } else {                                                               \
-            if (cfg.bt == 32) {                                                \
-                auto k = gru_fwd_kernel<float, 32, HPV, false>;                \
-                hipFuncSetAttribute((const void*)k,                            \
-                    hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);     \
-                k<<<grid, 256, lds, stream>>>((const float*)gi,                \
-                    (const float*)w, bhh, (float*)out, hlast, B, Tseq, n_dir); \
-            } else {                                                           \
-                auto k = gru_fwd_kernel<float, 16, HPV, false>;                \
-                hipFuncSetAttribute((const void*)k,                            \
-                    hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);     \
-                k<<<grid, 256, lds, stream>>>((const float*)gi,                \
-                    (const float*)w, bhh, (float*)out, hlast, B, Tseq, n_dir); \
-            }                                                                  \
-        }                                                                      \
-    } while (0)
-
-    FMDA_DISPATCH_HP(Hp, FWD_CASE)
-#undef FWD_CASE
-    err = hipGetLastError();
-    return err == hipSuccess ? 0 : (int)err;
+#define F(TY, BTV, HPV, WL, NTV, HO)                                           \
+    launch_fwd<TY, BTV, HPV, WL, NTV, HO>(gi, w, bhh, out, hlast, B, Tseq,     \
+                                          n_dir, lds, stream)
+    if (is_bf16) {
+        switch (Hp) {
+            case 16: F(bf16_t, 32, 16, true, 256, false); break;
+            case 32: F(bf16_t, 32, 32, true, 256, false); break;
+            case 64: F(bf16_t, 32, 64, true, 256, false); break;
+            case 128: F(bf16_t, 32, 128, false, 512, true); break;
+            case 256: F(bf16_t, 32, 256, false, 512, false); break;
+            case 512: F(bf16_t, 16, 512, false, 512, false); break;
+            default: return -1;
+        }
+    } else {
+        switch (Hp) {
+            case 16: F(float, 32, 16, false, 256, false); break;
+            case 32: F(float, 32, 32, false, 256, false); break;
+            case 64: F(float, 32, 64, false, 256, false); break;
+            case 128: F(float, 32, 128, false, 256, false); break;
+            case 256: F(float, 16, 256, false, 256, false); break;
+            case 512: F(float, 16, 512, false, 256, false); break;
+            default: return -1;
+        }
+    }
+#undef F
+    return hipGetLastError() == hipSuccess ? 0 : -4;
 }
 
 extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
@@ -765,45 +922,36 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    const float* dhT, void* dgi, void* dgh,
                                    float* dh0, float* dbhh, int B, int Tseq,
                                    int n_dir, hipStream_t stream) {
-    if (!is_bf16 && Hp > 256) return -3;  // fp32 oracle unsupported at H=512
-    const LaunchCfg cfg = bwd_cfg(is_bf16, Hp);
-    const size_t lds = bwd_lds_bytes(is_bf16, Hp, cfg.bt, cfg.wlds);
+    if (!is_bf16 && Hp > 256) return -3;
+    const LaunchCfg c = bwd_cfg(is_bf16, Hp);
+    const size_t lds = bwd_lds_bytes(is_bf16, Hp, c.bt, c.wlds);
     if (lds > 160 * 1024) return -2;
-    const dim3 grid((B + cfg.bt - 1) / cfg.bt, n_dir);
-    hipError_t err = hipSuccess;
-
-#define BWD_KERNEL(TY, BTV, HPV, WL)                                           \
-    do {                                                                       \
-        auto k = gru_bwd_kernel<TY, BTV, HPV, WL>;                             \
-        hipFuncSetAttribute((const void*)k,                                    \
-            hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);             \
-        k<<<grid, 256, lds, stream>>>((const TY*)gi, (const TY*)w, bhh,        \
-            (const TY*)out, (const TY*)dout, dhT, (TY*)dgi, (TY*)dgh, dh0,     \
-            dbhh, B, Tseq, n_dir);                                             \
-    } while (0)
-
-#define BWD_CASE(HPV)                                                          \
-    do {                                                                       \
-        if (is_bf16) {                                                         \
-            if (cfg.bt == 32 && cfg.wlds)                                      \
-                BWD_KERNEL(__hip_bfloat16, 32, HPV, true);                     \
-            else if (cfg.bt == 16 && cfg.wlds)                                 \
-                BWD_KERNEL(__hip_bfloat16, 16, HPV, true);                     \
-            else if (cfg.bt == 32)                                             \
-                BWD_KERNEL(__hip_bfloat16, 32, HPV, false);                    \
-            else                                                               \
-                BWD_KERNEL(__hip_bfloat16, 16, HPV, false);                    \
-        } else {                                                               \
-            if (cfg.bt == 32) BWD_KERNEL(float, 32, HPV, false);               \
-            else BWD_KERNEL(float, 16, HPV, false);                            \
-        }                                                                      \
-    } while (0)
-
-    FMDA_DISPATCH_HP(Hp, BWD_CASE)
-#undef BWD_CASE
-#undef BWD_KERNEL
-    err = hipGetLastError();
-    return err == hipSuccess ? 0 : (int)err;
+#define G(TY, BTV, HPV, WL, NTV, HO)                                           \
+    launch_bwd<TY, BTV, HPV, WL, NTV, HO>(gi, w, bhh, out, dout, dhT, dgi,     \
+                                          dgh, dh0, dbhh, B, Tseq, n_dir,      \
+                                          lds, stream)
+    if (is_bf16) {
+        switch (Hp) {
+            case 16: G(bf16_t, 32, 16, true, 256, false); break;
+            case 32: G(bf16_t, 32, 32, true, 256, false); break;
+            case 64: G(bf16_t, 32, 64, true, 256, false); break;
+            case 128: G(bf16_t, 32, 128, false, 512, true); break;
+            case 256: G(bf16_t, 32, 256, false, 512, false); break;
+            case 512: G(bf16_t, 16, 512, false, 512, false); break;
+            default: return -1;
+        }
+    } else {
+        switch (Hp) {
+            case 16: G(float, 32, 16, false, 256, false); break;
+            case 32: G(float, 32, 32, false, 256, false); break;
+            case 64: G(float, 32, 64, false, 256, false); break;
+            case 128: G(float, 32, 128, false, 256, false); break;
+            case 256: G(float, 16, 256, false, 256, false); break;
+            default: return -1;
+        }
+    }
+#undef G
+    return hipGetLastError() == hipSuccess ? 0 : -4;
 }
 
 extern "C" int fmda_mfma_selftest_launch(const void* A, const void* Bm,

@@ -80,21 +80,16 @@ class _GRURecurrence(torch.autograd.Function):
         d_hlast = d_hlast.contiguous().to(torch.float32)
         dgi, dgh, _dh0, dbhh = ext.gru_bwd(gi, w, bhh32, out, d_out, d_hlast)
 
-        # dW_hh[n, k] = sum_{b,t} dGh[b,t,n] * h_prev[b,t,k]  (plain MFMA
-        # GEMM in the compute dtype; fp32 accumulation inside rocBLAS).
-        # h_prev is the stored out shifted one step in each direction's
-        # processing order; the t=0 step's h_prev is zero, so the shifted
-        # slices drop it instead of materializing a zero-padded copy.
+        # dW_hh[n, k] = sum_{b,t} dGh_shifted[b,t,n] * out[b,t,k]: the
+        # kernel stores dGh time-shifted so slot t pairs with out[t] — one
+        # contiguous MFMA GEMM (transpose view, zero copies); the off-
+        # diagonal direction blocks of the cross product are discarded.
+        cross = torch.matmul(dgh.reshape(-1, n_dir * threeHp).t(),
+                             out.reshape(-1, n_dir * Hp))
         dw = torch.empty_like(w)
         for d in range(n_dir):
-            dgh_d = dgh[:, :, d * threeHp:(d + 1) * threeHp]
-            out_d = out[:, :, d * Hp:(d + 1) * Hp]
-            if d == 0:
-                dw[d] = torch.einsum("btn,bth->nh", dgh_d[:, 1:],
-                                     out_d[:, :-1])
-            else:
-                dw[d] = torch.einsum("btn,bth->nh", dgh_d[:, :-1],
-                                     out_d[:, 1:])
+            dw[d] = cross[d * threeHp:(d + 1) * threeHp,
+                          d * Hp:(d + 1) * Hp]
         return dgi, dw, dbhh
 
 
