@@ -43,7 +43,10 @@ def _worker(rank, world, port, out_q):
     loss = nn.functional.binary_cross_entropy_with_logits(logits, y[shard])
     loss.backward()
     engine.finalize()
-    grads = {n: p.grad.clone() for n, p in model.named_parameters()}
+    # Send plain numpy copies: torch tensors over a Queue use fd-passing
+    # through a unix socket that can vanish if this process exits first.
+    grads = {n: p.grad.detach().numpy().copy()
+             for n, p in model.named_parameters()}
     if rank == 0:
         out_q.put(grads)
     dist.barrier()
@@ -73,4 +76,4 @@ def test_allreduced_grads_match_single_process():
         assert p.exitcode == 0
 
     for n in ref:
-        assert torch.allclose(ref[n], got[n], atol=1e-6), n
+        assert torch.allclose(ref[n], torch.from_numpy(got[n]), atol=1e-6), n
