@@ -649,19 +649,23 @@ __global__ __launch_bounds__(NT) void gru_bwd_kernel(
                     const float hprev = to_f32<T>(hb_s[b * WP + j]);
                     const float dht = dh_s[b * HFP + j];
                     const bool live = (b < rows_valid);
+                    // Every dead-row product must be gated on `live`, not
+                    // just zeroed upstream: uninitialized LDS can hold inf/
+                    // NaN and 0 * inf = NaN would leak into dbacc below.
                     const float dz_pre =
                         live ? dht * (hprev - n) * z * (1.0f - z) : 0.0f;
                     const float dn_pre =
                         live ? dht * (1.0f - z) * (1.0f - n * n) : 0.0f;
-                    const float dr_pre = dn_pre * hn * r * (1.0f - r);
-                    const float dhn = dn_pre * r;
+                    const float dr_pre =
+                        live ? dn_pre * hn * r * (1.0f - r) : 0.0f;
+                    const float dhn = live ? dn_pre * r : 0.0f;
                     gi_s[b * GP3 + j] = from_f32<T>(dr_pre);
                     gi_s[b * GP3 + Hp + j] = from_f32<T>(dz_pre);
                     gi_s[b * GP3 + 2 * Hp + j] = from_f32<T>(dn_pre);
                     dgh_s[b * GP3 + j] = from_f32<T>(dr_pre);
                     dgh_s[b * GP3 + Hp + j] = from_f32<T>(dz_pre);
                     dgh_s[b * GP3 + 2 * Hp + j] = from_f32<T>(dhn);
-                    dh_s[b * HFP + j] = dht * z;
+                    dh_s[b * HFP + j] = live ? dht * z : 0.0f;
                     dbacc[i][0] += dr_pre;
                     dbacc[i][1] += dz_pre;
                     dbacc[i][2] += dhn;
