@@ -10,6 +10,8 @@ import time
 
 import torch
 
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from fmda_amd.ops import load_extension
 
 
