@@ -255,7 +255,7 @@ FMDA_DEV bf16x8_t load_wfragB(const __hip_bfloat16* __restrict__ w, long pitch,
 // grid:  (ceil(B/BT), n_dir); block NT threads; direction 1 runs reversed.
 // ===========================================================================
 template <typename T, int BT, int Hp, bool WLDS, int NT, bool HOIST>
-__global__ __launch_bounds__(NT) void gru_fwd_kernel(
+__global__ __launch_bounds__(NT, NT / 256) void gru_fwd_kernel(
     const T* __restrict__ gi, const T* __restrict__ w,
     const float* __restrict__ bhh, T* __restrict__ out,
     float* __restrict__ hlast, int B, int Tseq, int n_dir) {
@@ -434,7 +434,11 @@ __global__ __launch_bounds__(NT) void gru_fwd_kernel(
         }
         __syncthreads();  // h_t complete; gi_s free
 
-        // ---- phase C: write out[t]; commit prefetched gi
+        // ---- phase C: commit prefetched gi, then write out[t]. Commit
+        // goes FIRST: its s_waitcnt vmcnt(0) then waits only for the gi
+        // loads issued back in phase A (long returned), not for the out[t]
+        // stores (which would add a store-completion stall every step).
+        if (u + 1 < Tseq) pf.commit(gi_s, 3 * Hp, GP3, tid);
         if constexpr (IS_BF16) {
             store_tile<__hip_bfloat16, BT, NT>(
                 hb_s, (__hip_bfloat16*)(out_b + (long)tt * n_dir * Hp), Hp, WP,
@@ -444,7 +448,6 @@ __global__ __launch_bounds__(NT) void gru_fwd_kernel(
                 hf_s, (float*)(out_b + (long)tt * n_dir * Hp), Hp, HFP,
                 out_row, rows_valid, tid);
         }
-        if (u + 1 < Tseq) pf.commit(gi_s, 3 * Hp, GP3, tid);
         __syncthreads();
     }
 
@@ -462,7 +465,7 @@ __global__ __launch_bounds__(NT) void gru_fwd_kernel(
 // the in-kernel db_hh reduction.
 // ===========================================================================
 template <typename T, int BT, int Hp, bool WLDS, int NT, bool HOIST>
-__global__ __launch_bounds__(NT) void gru_bwd_kernel(
+__global__ __launch_bounds__(NT, NT / 256) void gru_bwd_kernel(
     const T* __restrict__ gi, const T* __restrict__ w,
     const float* __restrict__ bhh, const T* __restrict__ out,
     const T* __restrict__ dout, const float* __restrict__ dhT,
@@ -747,7 +750,20 @@ __global__ __launch_bounds__(NT) void gru_bwd_kernel(
         }
         __syncthreads();
 
-        // ---- phase C: write dGi and time-shifted dGh; commit prefetches
+        // ---- phase C: commit prefetches, then write dGi / dGh. The first
+        // commit's s_waitcnt vmcnt(0) waits only for phase-A loads, not the
+        // stores below (which would stall every step on store completion).
+        // gi_s is read (dGi values) and rewritten (next gi) by the SAME
+        // thread's chunk assignment, so store-then-commit within one thread
+        // needs no barrier; commits of hb_s/dh_s touch buffers no store
+        // reads.
+        if (have_next) {
+            if (un > 0)
+                pf_hb.commit(hb_s, Hp, WP, tid);
+            else
+                zero_tile<T, BT, NT>(hb_s, WP, tid);
+            pf_do.commit_accum_f32(dh_s, Hp, HFP, rows_valid, tid);
+        }
         store_tile<T, BT, NT>(gi_s, dgi_b + (long)tt * n_dir * 3 * Hp, 3 * Hp,
                               GP3, gi_row, rows_valid, tid);
         {
@@ -761,14 +777,8 @@ __global__ __launch_bounds__(NT) void gru_bwd_kernel(
                     dgh_b + (long)(rev ? 0 : (Tseq - 1)) * n_dir * 3 * Hp,
                     3 * Hp, gi_row, rows_valid, tid);
         }
-        if (have_next) {
+        if (have_next)
             pf_gi.commit(gi_s, 3 * Hp, GP3, tid);
-            if (un > 0)
-                pf_hb.commit(hb_s, Hp, WP, tid);
-            else
-                zero_tile<T, BT, NT>(hb_s, WP, tid);
-            pf_do.commit_accum_f32(dh_s, Hp, HFP, rows_valid, tid);
-        }
         __syncthreads();
     }
 
