@@ -9,7 +9,8 @@ extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
                                    float* hlast, int B, int Tseq, int n_dir,
                                    hipStream_t stream);
 extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
-                                   const void* w, const float* bhh,
+                                   const void* w, const void* wt,
+                                   const float* bhh,
                                    const void* out, const void* dout,
                                    const float* dhT, void* dgi, void* dgh,
                                    float* dh0, float* dbhh, int B, int Tseq,
@@ -75,9 +76,17 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
                             gi.options().dtype(torch::kFloat32));
     auto dbhh = torch::zeros({n_dir, 3 * Hp},
                              gi.options().dtype(torch::kFloat32));
+    // W^T for the carry GEMM's B fragments (v3 kernel streams them from
+    // L2 as contiguous bf16x8 rows instead of hoisting 48 VGPRs).
+    torch::Tensor wt;
+    const void* wt_ptr = nullptr;
+    if (is_bf16 && Hp == 128) {
+        wt = w.transpose(1, 2).contiguous();
+        wt_ptr = wt.data_ptr();
+    }
     auto stream = at::hip::getCurrentHIPStream();
     int rc = fmda_gru_bwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
-                                 w.data_ptr(), bhh.data_ptr<float>(),
+                                 w.data_ptr(), wt_ptr, bhh.data_ptr<float>(),
                                  out.data_ptr(), dout.data_ptr(),
                                  dhT.data_ptr<float>(), dgi.data_ptr(),
                                  dgh.data_ptr(), dh0.data_ptr<float>(),

@@ -255,7 +255,8 @@ FMDA_DEV bf16x8_t load_wfragB(const __hip_bfloat16* __restrict__ w, long pitch,
 // grid:  (ceil(B/BT), n_dir); block NT threads; direction 1 runs reversed.
 // ===========================================================================
 template <typename T, int BT, int Hp, bool WLDS, int NT, bool HOIST>
-__global__ __launch_bounds__(NT, NT / 256) void gru_fwd_kernel(
+__global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
+                          amdgpu_waves_per_eu(NT / 256, NT / 256))) void gru_fwd_kernel(
     const T* __restrict__ gi, const T* __restrict__ w,
     const float* __restrict__ bhh, T* __restrict__ out,
     float* __restrict__ hlast, int B, int Tseq, int n_dir) {
@@ -465,7 +466,8 @@ __global__ __launch_bounds__(NT, NT / 256) void gru_fwd_kernel(
 // the in-kernel db_hh reduction.
 // ===========================================================================
 template <typename T, int BT, int Hp, bool WLDS, int NT, bool HOIST>
-__global__ __launch_bounds__(NT, NT / 256) void gru_bwd_kernel(
+__global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
+                          amdgpu_waves_per_eu(NT / 256, NT / 256))) void gru_bwd_kernel(
     const T* __restrict__ gi, const T* __restrict__ w,
     const float* __restrict__ bhh, const T* __restrict__ out,
     const T* __restrict__ dout, const float* __restrict__ dhT,
@@ -809,6 +811,612 @@ __global__ __launch_bounds__(NT, NT / 256) void gru_bwd_kernel(
 }
 
 // ===========================================================================
+// v3 kernels (bf16, Hp = 128 hot path).
+//
+// What changed vs the v2 phase structure above, and why (measured on
+// MI355X, profiles/r01*):
+// - v2 spent 72% of wave time parked: 3 __syncthreads per step, a register
+//   prefetch the allocator spilled to scratch (global!) with an immediate
+//   vmcnt(0) stall, and gi staged through chunk registers.
+// - v3 stages gi by LDS-DMA (`global_load_lds`, 16 B/lane, lane-linear
+//   image) into a ring, so staging costs zero registers and no commit
+//   pass; raw `s_barrier` + counted `s_waitcnt vmcnt(N)` keep the DMA in
+//   flight across barriers (cdna_hip_programming.md glds rules).
+// - fp32 h lives in REGISTERS (each lane owns its MFMA C elements (b, j)
+//   across all timesteps); LDS holds only the bf16 copy the MFMA
+//   A-fragments need, double-buffered so the step loop has exactly TWO raw
+//   barriers (GEMM reads hb[p] while gates write hb[1-p]).
+// - forward runs BT=16/NT=256 with amdgpu_waves_per_eu(2,2): two BLOCKS
+//   per CU, whose independent serial chains hide each other's latencies.
+// - out[t-1] is stored one step late (from the buffer the GEMM is reading
+//   anyway), decoupling store completion from the chain.
+// ===========================================================================
+
+FMDA_DEV void glds16(const void* gsrc, void* ldst) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)gsrc,
+        (__attribute__((address_space(3))) unsigned int*)ldst, 16, 0, 0);
+}
+
+template <int BT, int Hp, int NT, int WPE>
+__global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
+                          amdgpu_waves_per_eu(WPE, WPE)))
+void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
+                       const __hip_bfloat16* __restrict__ w,
+                       const float* __restrict__ bhh,
+                       __hip_bfloat16* __restrict__ out,
+                       float* __restrict__ hlast, int B, int Tseq,
+                       int n_dir) {
+    constexpr int NW = NT / 64;
+    constexpr int MT = BT / 16;
+    constexpr int NCT = Hp / 16;
+    constexpr int CPW = NCT / NW;
+    constexpr int KK = Hp / 32;
+    constexpr int GP = 3 * Hp;           // unpadded gi pitch (lane-linear DMA)
+    constexpr int WP = Hp + 8;           // padded bf16 h pitch
+    constexpr int PIECES = (BT * GP * 2) / 1024;
+    constexpr int PW = PIECES / NW;      // glds instructions per wave per step
+    static_assert(PIECES % NW == 0 && NCT % NW == 0, "tiling mismatch");
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int b0 = blockIdx.x * BT;
+    const int dir = blockIdx.y;
+    const bool rev = (dir == 1);
+    const int rows_valid = min(BT, B - b0);
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    char* p = smem;
+    __hip_bfloat16* gi_s[2];
+    gi_s[0] = (__hip_bfloat16*)p; p += 2 * BT * GP;
+    gi_s[1] = (__hip_bfloat16*)p; p += 2 * BT * GP;
+    __hip_bfloat16* hb_s[2];
+    hb_s[0] = (__hip_bfloat16*)p; p += 2 * BT * WP;
+    hb_s[1] = (__hip_bfloat16*)p; p += 2 * BT * WP;
+    float* bhh_s = (float*)p;
+
+    const long gi_row = (long)Tseq * n_dir * 3 * Hp;
+    const long out_row = (long)Tseq * n_dir * Hp;
+    const __hip_bfloat16* gi_b = gi + (long)b0 * gi_row + (long)dir * 3 * Hp;
+    __hip_bfloat16* out_b = out + (long)b0 * out_row + (long)dir * Hp;
+
+    for (int c = tid; c < 3 * Hp; c += NT)
+        bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
+    zero_tile<__hip_bfloat16, BT, NT>(hb_s[0], WP, tid);
+
+    // W_hh B-fragments, register-resident for the whole sequence.
+    const __hip_bfloat16* wdir = w + (long)dir * 3 * Hp * Hp;
+    bf16x8_t wA[CPW][3][KK];
+#pragma unroll
+    for (int i = 0; i < CPW; ++i) {
+        const int ct = wave + NW * i;
+#pragma unroll
+        for (int g = 0; g < 3; ++g)
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk)
+                wA[i][g][kk] = load_wfragA<Hp>(wdir, Hp, ct, g, kk, lane);
+    }
+
+    float hreg[CPW][MT][4];
+#pragma unroll
+    for (int i = 0; i < CPW; ++i)
+#pragma unroll
+        for (int m = 0; m < MT; ++m)
+#pragma unroll
+            for (int e = 0; e < 4; ++e) hreg[i][m][e] = 0.0f;
+
+    // Lane-linear LDS-DMA of one timestep's gi tile (per-lane global
+    // addresses handle the strided batch rows; dead rows clamp to row 0).
+    const int wavu = __builtin_amdgcn_readfirstlane(wave);
+    auto glds_tile = [&](int tt, int buf) {
+        const __hip_bfloat16* src_t = gi_b + (long)tt * n_dir * 3 * Hp;
+#pragma unroll
+        for (int k = 0; k < PW; ++k) {
+            const int off = (wavu * PW + k) * 1024 + lane * 16;  // bytes
+            const int e = off >> 1;
+            int r = e / GP;
+            const int c = e % GP;
+            if (r >= rows_valid) r = 0;
+            glds16(src_t + (long)r * gi_row + c,
+                   (char*)gi_s[buf] + (long)(wavu * PW + k) * 1024);
+        }
+    };
+
+    {   // prologue: stage step 0 synchronously
+        glds_tile(rev ? (Tseq - 1) : 0, 0);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+    }
+
+    for (int u = 0; u < Tseq; ++u) {
+        const int pb = u & 1;
+        const int tt = rev ? (Tseq - 1 - u) : u;
+
+        // ---- phase A: store out[u-1], DMA gi[u+1], recurrent GEMM ----
+        if (u > 0) {
+            const int ttp = rev ? (Tseq - u) : (u - 1);
+            store_tile<__hip_bfloat16, BT, NT>(
+                hb_s[pb], out_b + (long)ttp * n_dir * Hp, Hp, WP, out_row,
+                rows_valid, tid);
+        }
+        if (u + 1 < Tseq)
+            glds_tile(rev ? (Tseq - 2 - u) : (u + 1), 1 - pb);
+
+        f32x4_t acc[CPW][3][MT];
+#pragma unroll
+        for (int i = 0; i < CPW; ++i)
+#pragma unroll
+            for (int g = 0; g < 3; ++g)
+#pragma unroll
+                for (int m = 0; m < MT; ++m) acc[i][g][m] = f32x4_t{0.f};
+#pragma unroll
+        for (int i = 0; i < CPW; ++i) {
+            const int arow = lane & 15;
+            const int koff = 8 * (lane >> 4);
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                const int kbase = 32 * kk + koff;
+                bf16x8_t a[MT];
+#pragma unroll
+                for (int m = 0; m < MT; ++m)
+                    a[m] = *(const bf16x8_t*)&(
+                        (const __bf16*)hb_s[pb])[(16 * m + arow) * WP + kbase];
+#pragma unroll
+                for (int g = 0; g < 3; ++g)
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+                        acc[i][g][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[m], wA[i][g][kk], acc[i][g][m], 0, 0, 0);
+            }
+        }
+        // Drain everything older than this step's issues (FIFO): the
+        // previous step's glds must have landed before phase B reads
+        // gi_s[pb]. Counted waits keep this step's DMA in flight.
+        if (u + 1 < Tseq) {
+            if (u > 0)
+                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(PW + 1) : "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(PW) : "memory");
+        } else {
+            if (u > 0)
+                asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        // ---- phase B: fused gates; h stays in registers ----
+#pragma unroll
+        for (int i = 0; i < CPW; ++i) {
+            const int ct = wave + NW * i;
+            const int j = ct * 16 + (lane & 15);
+            const __bf16* gprow = (const __bf16*)gi_s[pb];
+            __bf16* hrow = (__bf16*)hb_s[1 - pb];
+#pragma unroll
+            for (int m = 0; m < MT; ++m) {
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int b = 16 * m + 4 * (lane >> 4) + e;
+                    const float gr = acc[i][0][m][e] + bhh_s[j];
+                    const float gz = acc[i][1][m][e] + bhh_s[Hp + j];
+                    const float hn = acc[i][2][m][e] + bhh_s[2 * Hp + j];
+                    const float ir = (float)gprow[b * GP + j];
+                    const float iz = (float)gprow[b * GP + Hp + j];
+                    const float in_ = (float)gprow[b * GP + 2 * Hp + j];
+                    const float r = sigmoidf(ir + gr);
+                    const float z = sigmoidf(iz + gz);
+                    const float n = fast_tanh(in_ + r * hn);
+                    const float hprev = hreg[i][m][e];
+                    const float hnew = (1.0f - z) * n + z * hprev;
+                    hreg[i][m][e] = hnew;
+                    hrow[b * WP + j] = (__bf16)__float2bfloat16(hnew);
+                }
+            }
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+    }
+
+    {   // epilogue: out[T-1] + hlast
+        const int pl = Tseq & 1;
+        const int ttl = rev ? 0 : (Tseq - 1);
+        store_tile<__hip_bfloat16, BT, NT>(
+            hb_s[pl], out_b + (long)ttl * n_dir * Hp, Hp, WP, out_row,
+            rows_valid, tid);
+        float* hl = hlast + ((long)dir * B + b0) * Hp;
+#pragma unroll
+        for (int i = 0; i < CPW; ++i) {
+            const int ct = wave + NW * i;
+            const int j = ct * 16 + (lane & 15);
+#pragma unroll
+            for (int m = 0; m < MT; ++m)
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int b = 16 * m + 4 * (lane >> 4) + e;
+                    if (b < rows_valid) hl[(long)b * Hp + j] = hreg[i][m][e];
+                }
+        }
+    }
+}
+
+// v3 backward: glds-staged, three raw barriers per step with counted
+// vmcnt waits so every input stream stays one step ahead in flight.
+//   A: one-step-late dGi/dGh stores, prefetch/DMA issues for step u-1,
+//      recompute GEMM gh = h_{u-1} W^T (acc alive into B).
+//   B: fused gate gradients — writes dGi in place into the gi ring slot
+//      (thread-local), dGh into the padded LDS tile, db_hh partials and
+//      the dh carry into registers.
+//   C: carry GEMM dh += dGh W (accumulator lifetime disjoint from A's) and
+//      the h_{u-2} prefetch commit.
+// gi rides a 3-deep ring because its slot is rewritten with dGi and only
+// stored one step later. Stores are issued LAST in A so the counted waits
+// (vmcnt leaves exactly this step's issues) never stall on store
+// completion.
+template <int BT, int Hp, int NT, int WPE>
+__global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
+                          amdgpu_waves_per_eu(WPE, WPE)))
+void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
+                       const __hip_bfloat16* __restrict__ w,
+                       const __hip_bfloat16* __restrict__ wt,
+                       const float* __restrict__ bhh,
+                       const __hip_bfloat16* __restrict__ out,
+                       const __hip_bfloat16* __restrict__ dout,
+                       const float* __restrict__ dhT,
+                       __hip_bfloat16* __restrict__ dgi,
+                       __hip_bfloat16* __restrict__ dgh, float* __restrict__ dh0,
+                       float* __restrict__ dbhh, int B, int Tseq, int n_dir) {
+    constexpr int NW = NT / 64;
+    constexpr int MT = BT / 16;
+    constexpr int NCT = Hp / 16;
+    constexpr int CPW = NCT / NW;
+    constexpr int KK = Hp / 32;
+    constexpr int KK2 = (3 * Hp) / 32;
+    constexpr int GP = 3 * Hp;
+    constexpr int WP = Hp + 8;
+    constexpr int GP3 = 3 * Hp + 8;      // padded dgh_s pitch (MFMA A reads)
+    constexpr int PIECES = (BT * GP * 2) / 1024;
+    constexpr int PW = PIECES / NW;
+    constexpr int DOP = (BT * Hp * 2) / 1024 / NW;  // dout glds per wave
+    constexpr int NST = 2 * ((BT * GP * 2) / 16) / NT;  // store instrs (dgi+dgh)
+    static_assert(PIECES % NW == 0 && NCT % NW == 0, "tiling mismatch");
+    static_assert((BT * Hp * 2) % (1024 * NW) == 0, "dout tiling mismatch");
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int b0 = blockIdx.x * BT;
+    const int dir = blockIdx.y;
+    const bool rev = (dir == 1);
+    const int rows_valid = min(BT, B - b0);
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    char* p = smem;
+    __hip_bfloat16* gi_s[3];
+    gi_s[0] = (__hip_bfloat16*)p; p += 2 * BT * GP;
+    gi_s[1] = (__hip_bfloat16*)p; p += 2 * BT * GP;
+    gi_s[2] = (__hip_bfloat16*)p; p += 2 * BT * GP;
+    __hip_bfloat16* hb_s[2];
+    hb_s[0] = (__hip_bfloat16*)p; p += 2 * BT * WP;
+    hb_s[1] = (__hip_bfloat16*)p; p += 2 * BT * WP;
+    __hip_bfloat16* do_s[2];
+    do_s[0] = (__hip_bfloat16*)p; p += 2 * BT * Hp;
+    do_s[1] = (__hip_bfloat16*)p; p += 2 * BT * Hp;
+    __hip_bfloat16* dgh_s = (__hip_bfloat16*)p; p += 2 * BT * GP3;
+    float* bhh_s = (float*)p;
+
+    const long gi_row = (long)Tseq * n_dir * 3 * Hp;
+    const long out_row = (long)Tseq * n_dir * Hp;
+    const __hip_bfloat16* gi_b = gi + (long)b0 * gi_row + (long)dir * 3 * Hp;
+    __hip_bfloat16* dgi_b = dgi + (long)b0 * gi_row + (long)dir * 3 * Hp;
+    __hip_bfloat16* dgh_b = dgh + (long)b0 * gi_row + (long)dir * 3 * Hp;
+    const __hip_bfloat16* out_b = out + (long)b0 * out_row + (long)dir * Hp;
+    const __hip_bfloat16* dout_b = dout + (long)b0 * out_row + (long)dir * Hp;
+
+    for (int c = tid; c < 3 * Hp; c += NT)
+        bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
+
+    const __hip_bfloat16* wdir = w + (long)dir * 3 * Hp * Hp;
+    // wt_dir rows are W^T[j][n]: the carry GEMM's B fragments are then
+    // contiguous bf16x8 loads, streamed from L2 each step (hoisting them
+    // as well pushed the allocator into loop-carried scratch spills).
+    const __hip_bfloat16* wt_dir = wt + (long)dir * Hp * 3 * Hp;
+    bf16x8_t wA[CPW][3][KK];
+#pragma unroll
+    for (int i = 0; i < CPW; ++i) {
+        const int ct = wave + NW * i;
+#pragma unroll
+        for (int g = 0; g < 3; ++g)
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk)
+                wA[i][g][kk] = load_wfragA<Hp>(wdir, Hp, ct, g, kk, lane);
+    }
+
+    float dhreg[CPW][MT][4];   // dh carry (fp32, lane-owned)
+    float dbacc[CPW][3];
+#pragma unroll
+    for (int i = 0; i < CPW; ++i) {
+#pragma unroll
+        for (int g = 0; g < 3; ++g) dbacc[i][g] = 0.0f;
+        const int ct = wave + NW * i;
+        const int j = ct * 16 + (lane & 15);
+        const float* hT = dhT + ((long)dir * B + b0) * Hp;
+#pragma unroll
+        for (int m = 0; m < MT; ++m)
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+                const int b = 16 * m + 4 * (lane >> 4) + e;
+                dhreg[i][m][e] =
+                    (b < rows_valid) ? hT[(long)b * Hp + j] : 0.0f;
+            }
+    }
+
+    const int wavu = __builtin_amdgcn_readfirstlane(wave);
+    auto glds_gi = [&](int tt, int slot) {
+        const __hip_bfloat16* src_t = gi_b + (long)tt * n_dir * 3 * Hp;
+#pragma unroll
+        for (int k = 0; k < PW; ++k) {
+            const int off = (wavu * PW + k) * 1024 + lane * 16;
+            const int e = off >> 1;
+            int r = e / GP;
+            const int c = e % GP;
+            if (r >= rows_valid) r = 0;
+            glds16(src_t + (long)r * gi_row + c,
+                   (char*)gi_s[slot] + (long)(wavu * PW + k) * 1024);
+        }
+    };
+    auto glds_do = [&](int tt, int buf) {
+        const __hip_bfloat16* src_t = dout_b + (long)tt * n_dir * Hp;
+#pragma unroll
+        for (int k = 0; k < DOP; ++k) {
+            const int off = (wavu * DOP + k) * 1024 + lane * 16;
+            const int e = off >> 1;
+            int r = e / Hp;
+            const int c = e % Hp;
+            if (r >= rows_valid) r = 0;
+            glds16(src_t + (long)r * out_row + c,
+                   (char*)do_s[buf] + (long)(wavu * DOP + k) * 1024);
+        }
+    };
+
+    TilePrefetch<__hip_bfloat16, BT, NT,
+                 ((BT * Hp * 2) / 16 + NT - 1) / NT> pf_hb;
+
+    {   // prologue: stage step Tseq-1 inputs synchronously
+        const int u = Tseq - 1;
+        const int tt = rev ? 0 : u;
+        glds_gi(tt, u % 3);
+        glds_do(tt, u & 1);
+        if (u > 0) {
+            const int ttp = rev ? 1 : (u - 1);
+            stage_tile<__hip_bfloat16, BT, NT>(
+                hb_s[u & 1], out_b + (long)ttp * n_dir * Hp, Hp, WP, out_row,
+                rows_valid, tid);
+        } else {
+            zero_tile<__hip_bfloat16, BT, NT>(hb_s[u & 1], WP, tid);
+        }
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+    }
+
+    for (int u = Tseq - 1; u >= 0; --u) {
+        const int q = u & 1;
+        const int slot = u % 3;
+        const int tt = rev ? (Tseq - 1 - u) : u;
+        const bool have_next = (u > 0);
+        const bool have_prev = (u + 1 < Tseq);  // a step was processed before
+
+        // ---- phase A: issues first, stores last, recompute GEMM ----
+        if (have_next) {
+            const int ttn = rev ? (Tseq - u) : (u - 1);
+            pf_hb.issue(  // h_{u-2} for the next step's recompute GEMM
+                (u >= 2) ? (out_b + (long)(rev ? (Tseq + 1 - u) : (u - 2)) *
+                                        n_dir * Hp)
+                         : nullptr,
+                Hp, out_row, (u >= 2) ? rows_valid : 0, tid);
+            glds_gi(ttn, (u - 1) % 3);
+            glds_do(ttn, 1 - q);
+        }
+
+        // one-step-late stores of dGi / time-shifted dGh of step u+1
+        if (have_prev) {
+            const int ttp = rev ? (Tseq - 2 - u) : (u + 1);
+            store_tile<__hip_bfloat16, BT, NT>(
+                gi_s[(u + 1) % 3], dgi_b + (long)ttp * n_dir * 3 * Hp, GP, GP,
+                gi_row, rows_valid, tid);
+            const int sh = rev ? (ttp + 1) : (ttp - 1);  // always in range
+            store_tile<__hip_bfloat16, BT, NT>(
+                dgh_s, dgh_b + (long)sh * n_dir * 3 * Hp, GP, GP3, gi_row,
+                rows_valid, tid);
+        }
+
+        // Drain everything older than this step's issues: the previous
+        // step's DMA must have landed before phase B reads its tiles.
+        if (have_next) {
+            if (have_prev)
+                asm volatile("s_waitcnt vmcnt(%0)"
+                             ::"i"(1 + PW + DOP + NST) : "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(%0)"
+                             ::"i"(1 + PW + DOP) : "memory");
+        } else {
+            asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NST) : "memory");
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        // ---- phase B: recompute GEMM, then fused gate gradients ----
+        // (GEMM here, not in A: the accumulator then never lives across a
+        // barrier, which was pushing the allocator past 256 VGPRs.)
+        f32x4_t acc[CPW][3][MT];
+#pragma unroll
+        for (int i = 0; i < CPW; ++i)
+#pragma unroll
+            for (int g = 0; g < 3; ++g)
+#pragma unroll
+                for (int m = 0; m < MT; ++m) acc[i][g][m] = f32x4_t{0.f};
+#pragma unroll
+        for (int i = 0; i < CPW; ++i) {
+            const int arow = lane & 15;
+            const int koff = 8 * (lane >> 4);
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                const int kbase = 32 * kk + koff;
+                bf16x8_t a[MT];
+#pragma unroll
+                for (int m = 0; m < MT; ++m)
+                    a[m] = *(const bf16x8_t*)&(
+                        (const __bf16*)hb_s[q])[(16 * m + arow) * WP + kbase];
+#pragma unroll
+                for (int g = 0; g < 3; ++g)
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+                        acc[i][g][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[m], wA[i][g][kk], acc[i][g][m], 0, 0, 0);
+            }
+        }
+#pragma unroll
+        for (int i = 0; i < CPW; ++i) {
+            const int ct = wave + NW * i;
+            const int j = ct * 16 + (lane & 15);
+            __bf16* grow = (__bf16*)gi_s[slot];
+            __bf16* dgrow = (__bf16*)dgh_s;
+            const __bf16* hrow = (const __bf16*)hb_s[q];
+            const __bf16* dorow = (const __bf16*)do_s[q];
+#pragma unroll
+            for (int m = 0; m < MT; ++m) {
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int b = 16 * m + 4 * (lane >> 4) + e;
+                    const bool live = (b < rows_valid);
+                    const float gr = acc[i][0][m][e] + bhh_s[j];
+                    const float gz = acc[i][1][m][e] + bhh_s[Hp + j];
+                    const float hn = acc[i][2][m][e] + bhh_s[2 * Hp + j];
+                    const float ir = (float)grow[b * GP + j];
+                    const float iz = (float)grow[b * GP + Hp + j];
+                    const float in_ = (float)grow[b * GP + 2 * Hp + j];
+                    const float r = sigmoidf(ir + gr);
+                    const float z = sigmoidf(iz + gz);
+                    const float n = fast_tanh(in_ + r * hn);
+                    const float hprev = (float)hrow[b * WP + j];
+                    const float dht =
+                        dhreg[i][m][e] + (live ? (float)dorow[b * Hp + j]
+                                               : 0.0f);
+                    // live-gating every product: 0 * inf = NaN would leak
+                    // into dbacc from uninitialized dead-row LDS.
+                    const float dz_pre =
+                        live ? dht * (hprev - n) * z * (1.0f - z) : 0.0f;
+                    const float dn_pre =
+                        live ? dht * (1.0f - z) * (1.0f - n * n) : 0.0f;
+                    const float dr_pre =
+                        live ? dn_pre * hn * r * (1.0f - r) : 0.0f;
+                    const float dhn = live ? dn_pre * r : 0.0f;
+                    grow[b * GP + j] = (__bf16)__float2bfloat16(dr_pre);
+                    grow[b * GP + Hp + j] = (__bf16)__float2bfloat16(dz_pre);
+                    grow[b * GP + 2 * Hp + j] =
+                        (__bf16)__float2bfloat16(dn_pre);
+                    dgrow[b * GP3 + j] = (__bf16)__float2bfloat16(dr_pre);
+                    dgrow[b * GP3 + Hp + j] = (__bf16)__float2bfloat16(dz_pre);
+                    dgrow[b * GP3 + 2 * Hp + j] =
+                        (__bf16)__float2bfloat16(dhn);
+                    dhreg[i][m][e] = live ? dht * z : 0.0f;
+                    dbacc[i][0] += dr_pre;
+                    dbacc[i][1] += dz_pre;
+                    dbacc[i][2] += dhn;
+                }
+            }
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        // ---- phase C: carry GEMM dh += dGh W; commit h_{u-2} ----
+        if (have_next) {
+            // prefetch load drained: leave only this step's glds in flight
+            asm volatile("s_waitcnt vmcnt(%0)" ::"i"(PW + DOP) : "memory");
+            if (u >= 2)
+                pf_hb.commit(hb_s[1 - q], Hp, WP, tid);
+            else
+                zero_tile<__hip_bfloat16, BT, NT>(hb_s[1 - q], WP, tid);
+        }
+        {
+            f32x4_t acc2[CPW][MT];
+#pragma unroll
+            for (int i = 0; i < CPW; ++i)
+#pragma unroll
+                for (int m = 0; m < MT; ++m) acc2[i][m] = f32x4_t{0.f};
+#pragma unroll
+            for (int i = 0; i < CPW; ++i) {
+                const int arow = lane & 15;
+                const int koff = 8 * (lane >> 4);
+                const int jcol = (wave + NW * i) * 16 + (lane & 15);
+                const __bf16* wtrow =
+                    (const __bf16*)wt_dir + (long)jcol * 3 * Hp;
+#pragma unroll
+                for (int kk = 0; kk < KK2; ++kk) {
+                    const int kbase = 32 * kk + koff;
+                    bf16x8_t a[MT];
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+                        a[m] = *(const bf16x8_t*)&((const __bf16*)
+                                   dgh_s)[(16 * m + arow) * GP3 + kbase];
+                    const bf16x8_t bfr = *(const bf16x8_t*)&wtrow[kbase];
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+                        acc2[i][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[m], bfr, acc2[i][m], 0, 0, 0);
+                }
+            }
+#pragma unroll
+            for (int i = 0; i < CPW; ++i)
+#pragma unroll
+                for (int m = 0; m < MT; ++m)
+#pragma unroll
+                    for (int e = 0; e < 4; ++e)
+                        dhreg[i][m][e] += acc2[i][m][e];
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+    }
+
+    {   // epilogue: last dGi store, boundary zero dGh slot, dh0, db_hh
+        const int ttl = rev ? (Tseq - 1) : 0;
+        store_tile<__hip_bfloat16, BT, NT>(
+            gi_s[0], dgi_b + (long)ttl * n_dir * 3 * Hp, GP, GP, gi_row,
+            rows_valid, tid);
+        // step u=0's dGh pairs with h_{-1}=0: discarded; its global slot
+        // (rev ? 0 : Tseq-1) gets zeros instead.
+        store_zero_tile<__hip_bfloat16, BT, NT>(
+            dgh_b + (long)(rev ? 0 : (Tseq - 1)) * n_dir * 3 * Hp, GP, gi_row,
+            rows_valid, tid);
+        float* d0 = dh0 + ((long)dir * B + b0) * Hp;
+#pragma unroll
+        for (int i = 0; i < CPW; ++i) {
+            const int ct = wave + NW * i;
+            const int j = ct * 16 + (lane & 15);
+#pragma unroll
+            for (int m = 0; m < MT; ++m)
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int b = 16 * m + 4 * (lane >> 4) + e;
+                    if (b < rows_valid) d0[(long)b * Hp + j] = dhreg[i][m][e];
+                }
+#pragma unroll
+            for (int g = 0; g < 3; ++g) {
+                float v = dbacc[i][g];
+                v += __shfl_xor(v, 16);
+                v += __shfl_xor(v, 32);
+                if ((lane >> 4) == 0)
+                    atomicAdd(&dbhh[(long)dir * 3 * Hp + g * Hp + ct * 16 +
+                                    (lane & 15)], v);
+            }
+        }
+    }
+}
+
+
+// ===========================================================================
 // MFMA layout self-test.
 // ===========================================================================
 __global__ void mfma_selftest_kernel(const __hip_bfloat16* __restrict__ A,
@@ -914,6 +1522,46 @@ static int launch_bwd(const void* gi, const void* w, const float* bhh,
 
 using bf16_t = __hip_bfloat16;
 
+
+// v3 launch helpers (bf16 Hp=128).
+static int launch_fwd_v3_128(const void* gi, const void* w, const float* bhh,
+                             void* out, float* hlast, int B, int Tseq,
+                             int n_dir, hipStream_t stream) {
+    constexpr int BT = 16, Hp = 128, NT = 256;
+    const size_t lds = 2 * 2 * BT * 3 * Hp + 2 * 2 * BT * (Hp + 8) +
+                       4 * 3 * Hp;
+    auto k = gru_fwd_v3_kernel<BT, Hp, NT, 2>;
+    (void)hipFuncSetAttribute((const void*)k,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    const dim3 grid((B + BT - 1) / BT, n_dir);
+    k<<<grid, NT, lds, stream>>>((const __hip_bfloat16*)gi,
+                                 (const __hip_bfloat16*)w, bhh,
+                                 (__hip_bfloat16*)out, hlast, B, Tseq, n_dir);
+    return 0;
+}
+
+static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
+                             const float* bhh,
+                             const void* out, const void* dout,
+                             const float* dhT, void* dgi, void* dgh,
+                             float* dh0, float* dbhh, int B, int Tseq,
+                             int n_dir, hipStream_t stream) {
+    constexpr int BT = 32, Hp = 128, NT = 512;
+    const size_t lds = 3 * 2 * BT * 3 * Hp + 2 * 2 * BT * (Hp + 8) +
+                       2 * 2 * BT * Hp + 2 * BT * (3 * Hp + 8) + 4 * 3 * Hp;
+    auto k = gru_bwd_v3_kernel<BT, Hp, NT, 2>;
+    (void)hipFuncSetAttribute((const void*)k,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    const dim3 grid((B + BT - 1) / BT, n_dir);
+    k<<<grid, NT, lds, stream>>>(
+        (const __hip_bfloat16*)gi, (const __hip_bfloat16*)w,
+        (const __hip_bfloat16*)wt, bhh,
+        (const __hip_bfloat16*)out, (const __hip_bfloat16*)dout, dhT,
+        (__hip_bfloat16*)dgi, (__hip_bfloat16*)dgh, dh0, dbhh, B, Tseq,
+        n_dir);
+    return 0;
+}
+
 extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const float* bhh, void* out,
                                    float* hlast, int B, int Tseq, int n_dir,
@@ -929,7 +1577,10 @@ extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
             case 16: F(bf16_t, 32, 16, true, 256, false); break;
             case 32: F(bf16_t, 32, 32, true, 256, false); break;
             case 64: F(bf16_t, 32, 64, true, 256, false); break;
-            case 128: F(bf16_t, 32, 128, false, 512, true); break;
+            case 128:
+                launch_fwd_v3_128(gi, w, bhh, out, hlast, B, Tseq, n_dir,
+                                  stream);
+                break;
             case 256: F(bf16_t, 32, 256, false, 512, false); break;
             case 512: F(bf16_t, 16, 512, false, 512, false); break;
             default: return -1;
@@ -950,7 +1601,8 @@ extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
 }
 
 extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
-                                   const void* w, const float* bhh,
+                                   const void* w, const void* wt,
+                                   const float* bhh,
                                    const void* out, const void* dout,
                                    const float* dhT, void* dgi, void* dgh,
                                    float* dh0, float* dbhh, int B, int Tseq,
@@ -968,7 +1620,10 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
             case 16: G(bf16_t, 32, 16, true, 256, false); break;
             case 32: G(bf16_t, 32, 32, true, 256, false); break;
             case 64: G(bf16_t, 32, 64, true, 256, false); break;
-            case 128: G(bf16_t, 32, 128, false, 512, true); break;
+            case 128:
+                launch_bwd_v3_128(gi, w, wt, bhh, out, dout, dhT, dgi, dgh,
+                                  dh0, dbhh, B, Tseq, n_dir, stream);
+                break;
             case 256: G(bf16_t, 32, 256, false, 512, false); break;
             case 512: G(bf16_t, 16, 512, false, 512, false); break;
             default: return -1;
