@@ -1333,8 +1333,16 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 
         // ---- phase C: carry GEMM dh += dGh W; commit h_{u-2} ----
         if (have_next) {
-            // prefetch load drained: leave only this step's glds in flight
-            asm volatile("s_waitcnt vmcnt(%0)" ::"i"(PW + DOP) : "memory");
+            // Drain the h prefetch (oldest of this step's issues) while
+            // leaving the glds tiles AND the trailing dGi/dGh stores in
+            // flight — waiting on fresh store completion here cost ~1 us
+            // per step.
+            if (have_prev)
+                asm volatile("s_waitcnt vmcnt(%0)"
+                             ::"i"(PW + DOP + NST) : "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(%0)"
+                             ::"i"(PW + DOP) : "memory");
             if (u >= 2)
                 pf_hb.commit(hb_s[1 - q], Hp, WP, tid);
             else
@@ -1376,9 +1384,12 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                     for (int e = 0; e < 4; ++e)
                         dhreg[i][m][e] += acc2[i][m][e];
         }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
+        // no barrier here: phase A of the next step touches no LDS that C
+        // wrote, and its end-of-phase rendezvous publishes the commit
+        // before the next gate phase reads hb.
     }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
 
     {   // epilogue: last dGi store, boundary zero dGh slot, dh0, db_hh
         const int ttl = rev ? (Tseq - 1) : 0;
