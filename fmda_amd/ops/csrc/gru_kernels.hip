@@ -1041,19 +1041,26 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
     }
 }
 
-// v3 backward: glds-staged, three raw barriers per step with counted
-// vmcnt waits so every input stream stays one step ahead in flight.
-//   A: one-step-late dGi/dGh stores, prefetch/DMA issues for step u-1,
-//      recompute GEMM gh = h_{u-1} W^T (acc alive into B).
-//   B: fused gate gradients — writes dGi in place into the gi ring slot
-//      (thread-local), dGh into the padded LDS tile, db_hh partials and
-//      the dh carry into registers.
-//   C: carry GEMM dh += dGh W (accumulator lifetime disjoint from A's) and
-//      the h_{u-2} prefetch commit.
-// gi rides a 3-deep ring because its slot is rewritten with dGi and only
-// stored one step later. Stores are issued LAST in A so the counted waits
-// (vmcnt leaves exactly this step's issues) never stall on store
-// completion.
+// v3 backward (third structure — the one that measures fastest; the
+// previous two are documented here so they are not rediscovered):
+//   (1) glds tiles + per-step W^T fragments as ordinary L2 loads: ROCm 7.2
+//       rewrites every use of an ordinary-load result into s_waitcnt
+//       vmcnt(0) while ANY glds is in flight, so each step drained the
+//       whole DMA pipeline and parked 67% of wave time.
+//   (2) register staging everywhere (no glds): the chunk registers pushed
+//       allocation past 256 VGPRs and the spill RELOADS are themselves
+//       ordinary loads — 2x slower again.
+//   => ALL in-loop VMEM must be glds or stores; both W fragment sets are
+//      hoisted ONCE in the prologue (their one-time loads complete before
+//      the loop); the kernel must allocate with ZERO spills.
+// h tiles (read as MFMA A-fragments) are DMA'd with an XOR-swizzled
+// SOURCE address (lane-linear LDS image, cdna_hip_programming.md rule 21):
+// LDS[r][k16] holds out[r][k16 ^ (r & 15)], so fragment reads at
+// row*256 + ((k16 ^ (r & 15)) * 16) touch 16 distinct 16-byte slots per
+// 16-row group instead of one (16-way conflict at the unswizzled pitch).
+// Phases: A[glds issues + one-step-late dGi/dGh stores] -> rendezvous ->
+// B[recompute GEMM + fused gate grads] -> barrier -> C[carry GEMM from
+// dgh_s, register-only] -> (no barrier).
 template <int BT, int Hp, int NT, int WPE>
 __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
                           amdgpu_waves_per_eu(WPE, WPE)))
@@ -1074,14 +1081,14 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
     constexpr int KK = Hp / 32;
     constexpr int KK2 = (3 * Hp) / 32;
     constexpr int GP = 3 * Hp;
-    constexpr int WP = Hp + 8;
+    constexpr int HROW = Hp * 2;         // h/dout tile row bytes (unpadded)
     constexpr int GP3 = 3 * Hp + 8;      // padded dgh_s pitch (MFMA A reads)
-    constexpr int PIECES = (BT * GP * 2) / 1024;
-    constexpr int PW = PIECES / NW;
-    constexpr int DOP = (BT * Hp * 2) / 1024 / NW;  // dout glds per wave
-    constexpr int NST = 2 * ((BT * GP * 2) / 16) / NT;  // store instrs (dgi+dgh)
-    static_assert(PIECES % NW == 0 && NCT % NW == 0, "tiling mismatch");
-    static_assert((BT * Hp * 2) % (1024 * NW) == 0, "dout tiling mismatch");
+    constexpr int PW = (BT * GP * 2) / 1024 / NW;   // gi glds per wave
+    constexpr int DOP = (BT * HROW) / 1024 / NW;    // dout glds per wave
+    constexpr int HBP = DOP;                        // h glds per wave
+    constexpr int NST = 2 * ((BT * GP * 2) / 16) / NT;  // dGi+dGh store instrs
+    static_assert(NCT % NW == 0 && (BT * GP * 2) % (1024 * NW) == 0 &&
+                  (BT * HROW) % (1024 * NW) == 0, "tiling mismatch");
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -1093,18 +1100,24 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     char* p = smem;
-    __hip_bfloat16* gi_s[3];
-    gi_s[0] = (__hip_bfloat16*)p; p += 2 * BT * GP;
-    gi_s[1] = (__hip_bfloat16*)p; p += 2 * BT * GP;
-    gi_s[2] = (__hip_bfloat16*)p; p += 2 * BT * GP;
-    __hip_bfloat16* hb_s[2];
-    hb_s[0] = (__hip_bfloat16*)p; p += 2 * BT * WP;
-    hb_s[1] = (__hip_bfloat16*)p; p += 2 * BT * WP;
-    __hip_bfloat16* do_s[2];
-    do_s[0] = (__hip_bfloat16*)p; p += 2 * BT * Hp;
-    do_s[1] = (__hip_bfloat16*)p; p += 2 * BT * Hp;
+    // Plain pointers + select helpers, NOT runtime-indexed local arrays:
+    // a dynamically indexed private pointer array lands in scratch, and
+    // every scratch reload is an ordinary VMEM load that triggers the
+    // glds vmcnt(0) drain (this alone cost 3x on this kernel).
+    __hip_bfloat16* gi_s0 = (__hip_bfloat16*)p; p += 2 * BT * GP;
+    __hip_bfloat16* gi_s1 = (__hip_bfloat16*)p; p += 2 * BT * GP;
+    __hip_bfloat16* gi_s2 = (__hip_bfloat16*)p; p += 2 * BT * GP;
+    __hip_bfloat16* hb_s0 = (__hip_bfloat16*)p; p += BT * HROW;
+    __hip_bfloat16* hb_s1 = (__hip_bfloat16*)p; p += BT * HROW;
+    __hip_bfloat16* do_s0 = (__hip_bfloat16*)p; p += BT * HROW;
+    __hip_bfloat16* do_s1 = (__hip_bfloat16*)p; p += BT * HROW;
     __hip_bfloat16* dgh_s = (__hip_bfloat16*)p; p += 2 * BT * GP3;
     float* bhh_s = (float*)p;
+    auto gi_slot = [&](int sl) {
+        return sl == 0 ? gi_s0 : (sl == 1 ? gi_s1 : gi_s2);
+    };
+    auto hb_buf = [&](int b) { return b ? hb_s1 : hb_s0; };
+    auto do_buf = [&](int b) { return b ? do_s1 : do_s0; };
 
     const long gi_row = (long)Tseq * n_dir * 3 * Hp;
     const long out_row = (long)Tseq * n_dir * Hp;
@@ -1117,20 +1130,25 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
     for (int c = tid; c < 3 * Hp; c += NT)
         bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
 
+    // Both W fragment sets, register-resident; loaded ONCE here (ordinary
+    // loads are legal before any glds is issued).
     const __hip_bfloat16* wdir = w + (long)dir * 3 * Hp * Hp;
-    // wt_dir rows are W^T[j][n]: the carry GEMM's B fragments are then
-    // contiguous bf16x8 loads, streamed from L2 each step (hoisting them
-    // as well pushed the allocator into loop-carried scratch spills).
     const __hip_bfloat16* wt_dir = wt + (long)dir * Hp * 3 * Hp;
     bf16x8_t wA[CPW][3][KK];
+    bf16x8_t wB[CPW][KK2];
 #pragma unroll
     for (int i = 0; i < CPW; ++i) {
         const int ct = wave + NW * i;
+        const int jcol = ct * 16 + (lane & 15);
+        const __bf16* wtrow = (const __bf16*)wt_dir + (long)jcol * 3 * Hp;
 #pragma unroll
         for (int g = 0; g < 3; ++g)
 #pragma unroll
             for (int kk = 0; kk < KK; ++kk)
                 wA[i][g][kk] = load_wfragA<Hp>(wdir, Hp, ct, g, kk, lane);
+#pragma unroll
+        for (int kk = 0; kk < KK2; ++kk)
+            wB[i][kk] = *(const bf16x8_t*)&wtrow[32 * kk + 8 * (lane >> 4)];
     }
 
     float dhreg[CPW][MT][4];   // dh carry (fp32, lane-owned)
@@ -1153,8 +1171,9 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
     }
 
     const int wavu = __builtin_amdgcn_readfirstlane(wave);
-    auto glds_gi = [&](int tt, int slot) {
+    auto glds_gi = [&](int tt, int sl) {
         const __hip_bfloat16* src_t = gi_b + (long)tt * n_dir * 3 * Hp;
+        __hip_bfloat16* dst = gi_slot(sl);
 #pragma unroll
         for (int k = 0; k < PW; ++k) {
             const int off = (wavu * PW + k) * 1024 + lane * 16;
@@ -1163,7 +1182,7 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
             const int c = e % GP;
             if (r >= rows_valid) r = 0;
             glds16(src_t + (long)r * gi_row + c,
-                   (char*)gi_s[slot] + (long)(wavu * PW + k) * 1024);
+                   (char*)dst + (long)(wavu * PW + k) * 1024);
         }
     };
     auto glds_do = [&](int tt, int buf) {
@@ -1176,25 +1195,47 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
             const int c = e % Hp;
             if (r >= rows_valid) r = 0;
             glds16(src_t + (long)r * out_row + c,
-                   (char*)do_s[buf] + (long)(wavu * DOP + k) * 1024);
+                   (char*)do_buf(buf) + (long)(wavu * DOP + k) * 1024);
         }
     };
+    // XOR-swizzled source: LDS[r][k16] = out[r][k16 ^ (r & 15)] so the
+    // MFMA A-fragment reads are bank-conflict-free at the unpadded pitch.
+    auto glds_hb = [&](int tt, int buf) {
+        const __hip_bfloat16* src_t = out_b + (long)tt * n_dir * Hp;
+#pragma unroll
+        for (int k = 0; k < HBP; ++k) {
+            const int off = (wavu * HBP + k) * 1024 + lane * 16;
+            int r = off / HROW;
+            const int k16 = (off % HROW) >> 4;
+            const int csrc = (k16 ^ (r & 15)) << 4;  // byte col in row
+            if (r >= rows_valid) r = 0;
+            glds16((const char*)(src_t + (long)r * out_row) + csrc,
+                   (char*)hb_buf(buf) + (long)(wavu * HBP + k) * 1024);
+        }
+    };
+    auto zero_hb = [&](int buf) {
+        for (int c = tid; c < BT * Hp; c += NT)
+            ((__bf16*)hb_buf(buf))[c] = (__bf16)0.0f;
+    };
+    // swizzled scalar read of h_prev[b][j]
+    auto hb_read = [&](const __hip_bfloat16* hb, int b, int j) -> float {
+        const int k16 = (j * 2) >> 4;           // 16B slot of column j
+        const int ksw = k16 ^ (b & 15);
+        const int within = (j * 2) & 15;
+        return (float)*(const __bf16*)((const char*)hb + b * HROW +
+                                       ksw * 16 + within);
+    };
 
-    TilePrefetch<__hip_bfloat16, BT, NT,
-                 ((BT * Hp * 2) / 16 + NT - 1) / NT> pf_hb;
-
-    {   // prologue: stage step Tseq-1 inputs synchronously
+    {   // prologue: stage step Tseq-1 inputs
         const int u = Tseq - 1;
         const int tt = rev ? 0 : u;
         glds_gi(tt, u % 3);
         glds_do(tt, u & 1);
         if (u > 0) {
             const int ttp = rev ? 1 : (u - 1);
-            stage_tile<__hip_bfloat16, BT, NT>(
-                hb_s[u & 1], out_b + (long)ttp * n_dir * Hp, Hp, WP, out_row,
-                rows_valid, tid);
+            glds_hb(ttp, u & 1);
         } else {
-            zero_tile<__hip_bfloat16, BT, NT>(hb_s[u & 1], WP, tid);
+            zero_hb(u & 1);
         }
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -1208,23 +1249,20 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
         const bool have_next = (u > 0);
         const bool have_prev = (u + 1 < Tseq);  // a step was processed before
 
-        // ---- phase A: issues first, stores last, recompute GEMM ----
+        // ---- phase A: DMA issues, then one-step-late stores ----
         if (have_next) {
             const int ttn = rev ? (Tseq - u) : (u - 1);
-            pf_hb.issue(  // h_{u-2} for the next step's recompute GEMM
-                (u >= 2) ? (out_b + (long)(rev ? (Tseq + 1 - u) : (u - 2)) *
-                                        n_dir * Hp)
-                         : nullptr,
-                Hp, out_row, (u >= 2) ? rows_valid : 0, tid);
             glds_gi(ttn, (u - 1) % 3);
             glds_do(ttn, 1 - q);
+            if (u >= 2)
+                glds_hb(rev ? (Tseq + 1 - u) : (u - 2), 1 - q);
+            else
+                zero_hb(1 - q);
         }
-
-        // one-step-late stores of dGi / time-shifted dGh of step u+1
         if (have_prev) {
             const int ttp = rev ? (Tseq - 2 - u) : (u + 1);
             store_tile<__hip_bfloat16, BT, NT>(
-                gi_s[(u + 1) % 3], dgi_b + (long)ttp * n_dir * 3 * Hp, GP, GP,
+                gi_slot((u + 1) % 3), dgi_b + (long)ttp * n_dir * 3 * Hp, GP, GP,
                 gi_row, rows_valid, tid);
             const int sh = rev ? (ttp + 1) : (ttp - 1);  // always in range
             store_tile<__hip_bfloat16, BT, NT>(
@@ -1232,24 +1270,27 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                 rows_valid, tid);
         }
 
-        // Drain everything older than this step's issues: the previous
-        // step's DMA must have landed before phase B reads its tiles.
+        // Rendezvous: drain everything older than this step's issues so
+        // the PREVIOUS step's DMA is visible to phase B; keep this step's
+        // glds (and trailing stores) in flight.
         if (have_next) {
             if (have_prev)
                 asm volatile("s_waitcnt vmcnt(%0)"
-                             ::"i"(1 + PW + DOP + NST) : "memory");
+                             ::"i"(PW + DOP + HBP + NST) : "memory");
             else
                 asm volatile("s_waitcnt vmcnt(%0)"
-                             ::"i"(1 + PW + DOP) : "memory");
+                             ::"i"(PW + DOP + HBP) : "memory");
         } else {
-            asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NST) : "memory");
+            if (have_prev)
+                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NST) : "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
 
         // ---- phase B: recompute GEMM, then fused gate gradients ----
-        // (GEMM here, not in A: the accumulator then never lives across a
-        // barrier, which was pushing the allocator past 256 VGPRs.)
+        const __hip_bfloat16* hbq = hb_buf(q);
         f32x4_t acc[CPW][3][MT];
 #pragma unroll
         for (int i = 0; i < CPW; ++i)
@@ -1260,15 +1301,16 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 #pragma unroll
         for (int i = 0; i < CPW; ++i) {
             const int arow = lane & 15;
-            const int koff = 8 * (lane >> 4);
 #pragma unroll
             for (int kk = 0; kk < KK; ++kk) {
-                const int kbase = 32 * kk + koff;
                 bf16x8_t a[MT];
 #pragma unroll
-                for (int m = 0; m < MT; ++m)
-                    a[m] = *(const bf16x8_t*)&(
-                        (const __bf16*)hb_s[q])[(16 * m + arow) * WP + kbase];
+                for (int m = 0; m < MT; ++m) {
+                    const int row = 16 * m + arow;
+                    const int k16 = (4 * kk + (lane >> 4)) ^ (row & 15);
+                    a[m] = *(const bf16x8_t*)((const char*)hbq +
+                                              row * HROW + k16 * 16);
+                }
 #pragma unroll
                 for (int g = 0; g < 3; ++g)
 #pragma unroll
@@ -1281,10 +1323,9 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
         for (int i = 0; i < CPW; ++i) {
             const int ct = wave + NW * i;
             const int j = ct * 16 + (lane & 15);
-            __bf16* grow = (__bf16*)gi_s[slot];
+            __bf16* grow = (__bf16*)gi_slot(slot);
             __bf16* dgrow = (__bf16*)dgh_s;
-            const __bf16* hrow = (const __bf16*)hb_s[q];
-            const __bf16* dorow = (const __bf16*)do_s[q];
+            const __bf16* dorow = (const __bf16*)do_buf(q);
 #pragma unroll
             for (int m = 0; m < MT; ++m) {
 #pragma unroll
@@ -1300,7 +1341,7 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                     const float r = sigmoidf(ir + gr);
                     const float z = sigmoidf(iz + gz);
                     const float n = fast_tanh(in_ + r * hn);
-                    const float hprev = (float)hrow[b * WP + j];
+                    const float hprev = hb_read(hbq, b, j);
                     const float dht =
                         dhreg[i][m][e] + (live ? (float)dorow[b * Hp + j]
                                                : 0.0f);
@@ -1331,23 +1372,7 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
 
-        // ---- phase C: carry GEMM dh += dGh W; commit h_{u-2} ----
-        if (have_next) {
-            // Drain the h prefetch (oldest of this step's issues) while
-            // leaving the glds tiles AND the trailing dGi/dGh stores in
-            // flight — waiting on fresh store completion here cost ~1 us
-            // per step.
-            if (have_prev)
-                asm volatile("s_waitcnt vmcnt(%0)"
-                             ::"i"(PW + DOP + NST) : "memory");
-            else
-                asm volatile("s_waitcnt vmcnt(%0)"
-                             ::"i"(PW + DOP) : "memory");
-            if (u >= 2)
-                pf_hb.commit(hb_s[1 - q], Hp, WP, tid);
-            else
-                zero_tile<__hip_bfloat16, BT, NT>(hb_s[1 - q], WP, tid);
-        }
+        // ---- phase C: carry GEMM dh += dGh W — registers + LDS only ----
         {
             f32x4_t acc2[CPW][MT];
 #pragma unroll
@@ -1358,9 +1383,6 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
             for (int i = 0; i < CPW; ++i) {
                 const int arow = lane & 15;
                 const int koff = 8 * (lane >> 4);
-                const int jcol = (wave + NW * i) * 16 + (lane & 15);
-                const __bf16* wtrow =
-                    (const __bf16*)wt_dir + (long)jcol * 3 * Hp;
 #pragma unroll
                 for (int kk = 0; kk < KK2; ++kk) {
                     const int kbase = 32 * kk + koff;
@@ -1369,11 +1391,10 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                     for (int m = 0; m < MT; ++m)
                         a[m] = *(const bf16x8_t*)&((const __bf16*)
                                    dgh_s)[(16 * m + arow) * GP3 + kbase];
-                    const bf16x8_t bfr = *(const bf16x8_t*)&wtrow[kbase];
 #pragma unroll
                     for (int m = 0; m < MT; ++m)
                         acc2[i][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            a[m], bfr, acc2[i][m], 0, 0, 0);
+                            a[m], wB[i][kk], acc2[i][m], 0, 0, 0);
                 }
             }
 #pragma unroll
@@ -1384,9 +1405,7 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                     for (int e = 0; e < 4; ++e)
                         dhreg[i][m][e] += acc2[i][m][e];
         }
-        // no barrier here: phase A of the next step touches no LDS that C
-        // wrote, and its end-of-phase rendezvous publishes the commit
-        // before the next gate phase reads hb.
+        // no barrier: the next step's A-end rendezvous orders everything.
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
@@ -1394,7 +1413,7 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
     {   // epilogue: last dGi store, boundary zero dGh slot, dh0, db_hh
         const int ttl = rev ? (Tseq - 1) : 0;
         store_tile<__hip_bfloat16, BT, NT>(
-            gi_s[0], dgi_b + (long)ttl * n_dir * 3 * Hp, GP, GP, gi_row,
+            gi_s0, dgi_b + (long)ttl * n_dir * 3 * Hp, GP, GP, gi_row,
             rows_valid, tid);
         // step u=0's dGh pairs with h_{-1}=0: discarded; its global slot
         // (rev ? 0 : Tseq-1) gets zeros instead.
@@ -1427,7 +1446,6 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 }
 
 
-// ===========================================================================
 // MFMA layout self-test.
 // ===========================================================================
 __global__ void mfma_selftest_kernel(const __hip_bfloat16* __restrict__ A,
@@ -1557,10 +1575,13 @@ static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
                              const float* dhT, void* dgi, void* dgh,
                              float* dh0, float* dbhh, int B, int Tseq,
                              int n_dir, hipStream_t stream) {
-    constexpr int BT = 32, Hp = 128, NT = 512;
-    const size_t lds = 3 * 2 * BT * 3 * Hp + 2 * 2 * BT * (Hp + 8) +
+    constexpr int BT = 32, Hp = 128, NT = 256;
+    const size_t lds = 3 * 2 * BT * 3 * Hp + 2 * 2 * BT * Hp +
                        2 * 2 * BT * Hp + 2 * BT * (3 * Hp + 8) + 4 * 3 * Hp;
-    auto k = gru_bwd_v3_kernel<BT, Hp, NT, 2>;
+    // 4 waves at 1 wave/SIMD: this shape is spill-free (NT=512 at 2
+    // waves/SIMD spills 3 VGPRs, and every spill reload re-triggers the
+    // glds vmcnt(0) drain).
+    auto k = gru_bwd_v3_kernel<BT, Hp, NT, 1>;
     (void)hipFuncSetAttribute((const void*)k,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     const dim3 grid((B + BT - 1) / BT, n_dir);
