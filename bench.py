@@ -145,7 +145,10 @@ def main():
     weight = 1.0 / rates
     pos_weight = (1.0 - rates) / rates
     loss_fn = nn.BCEWithLogitsLoss(weight=weight, pos_weight=pos_weight)
-    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    # Fused multi-tensor grad-clip + Adam (HIP kernels on GPU, equivalent
+    # eager pair on CPU) — reference semantics: clip_grad_norm_(50) + Adam.
+    from fmda_amd.optim import FusedClipAdam
+    opt = FusedClipAdam(model.parameters(), lr=1e-3, clip=50.0)
 
     # Pre-generate a small rotating pool of synthetic batches per rank.
     pool = []
@@ -167,8 +170,7 @@ def main():
         loss.backward()
         if engine is not None:
             engine.finalize()
-        nn.utils.clip_grad_norm_(model.parameters(), 50.0)
-        opt.step()
+        opt.step()  # fused clip(50) + Adam
         pred = torch.sigmoid(logits.detach().float()) > 0.5
         acc_sum.add_(subset_accuracy(y, pred))  # stays on device, no sync
         if measure_quality:

@@ -96,12 +96,16 @@ class BiGRU(nn.Module):
                             self.hidden_size)
         last_hidden = hidden_v[-1].sum(dim=0)
 
-        if self.bidirectional:
-            gru_out = (gru_out[:, :, :self.hidden_size]
-                       + gru_out[:, :, self.hidden_size:])
-
-        max_pool = gru_out.max(dim=1).values
-        avg_pool = gru_out.sum(dim=1) / float(input_length)
+        if input_seq.is_cuda:
+            # fused direction-sum + max/avg pooling HIP kernel
+            from ..ops.interface import dirsum_pool
+            max_pool, avg_pool = dirsum_pool(gru_out, self.n_directions)
+        else:
+            if self.bidirectional:
+                gru_out = (gru_out[:, :, :self.hidden_size]
+                           + gru_out[:, :, self.hidden_size:])
+            max_pool = gru_out.max(dim=1).values
+            avg_pool = gru_out.sum(dim=1) / float(input_length)
 
         concat_out = torch.cat([last_hidden, max_pool, avg_pool], dim=1)
         if concat_out.dtype != self.linear.weight.dtype:

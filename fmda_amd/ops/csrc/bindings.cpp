@@ -17,6 +17,19 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    int n_dir, hipStream_t stream);
 extern "C" int fmda_mfma_selftest_launch(const void* A, const void* Bm,
                                          float* C, hipStream_t stream);
+extern "C" int fmda_pool_fwd_launch(int is_bf16, const void* out, float* maxv,
+                                    float* avgv, int* amax, int B, int Tseq,
+                                    int H, int n_dir, hipStream_t stream);
+extern "C" int fmda_pool_bwd_launch(int is_bf16, const float* dmax,
+                                    const float* davg, const int* amax,
+                                    void* dout, int B, int Tseq, int H,
+                                    int n_dir, hipStream_t stream);
+extern "C" int fmda_opt_norm2_launch(const void* chunks, int n_chunks,
+                                     float* out, hipStream_t stream);
+extern "C" int fmda_opt_adam_launch(const void* chunks, int n_chunks,
+                                    const float* norm2, float clip, float lr,
+                                    float beta1, float beta2, float eps,
+                                    float bc1, float bc2, hipStream_t stream);
 
 namespace {
 
@@ -109,8 +122,71 @@ torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor Bm) {
     return C;
 }
 
+std::vector<torch::Tensor> pool_fwd(torch::Tensor out, int64_t n_dir) {
+    TORCH_CHECK(out.is_cuda() && out.is_contiguous() && out.dim() == 3);
+    const bool bf16 = out.scalar_type() == torch::kBFloat16;
+    TORCH_CHECK(bf16 || out.scalar_type() == torch::kFloat32);
+    const int B = out.size(0), T = out.size(1);
+    const int H = out.size(2) / n_dir;
+    auto f32 = out.options().dtype(torch::kFloat32);
+    auto maxv = torch::empty({B, H}, f32);
+    auto avgv = torch::empty({B, H}, f32);
+    auto amax = torch::empty({B, H}, out.options().dtype(torch::kInt32));
+    auto stream = at::hip::getCurrentHIPStream();
+    int rc = fmda_pool_fwd_launch(bf16 ? 1 : 0, out.data_ptr(),
+                                  maxv.data_ptr<float>(), avgv.data_ptr<float>(),
+                                  amax.data_ptr<int>(), B, T, H, (int)n_dir,
+                                  stream.stream());
+    TORCH_CHECK(rc == 0, "pool_fwd launch failed");
+    return {maxv, avgv, amax};
+}
+
+torch::Tensor pool_bwd(torch::Tensor dmax, torch::Tensor davg,
+                       torch::Tensor amax, int64_t T, int64_t n_dir,
+                       torch::ScalarType dtype) {
+    TORCH_CHECK(dmax.is_cuda() && dmax.is_contiguous() && davg.is_contiguous());
+    const int B = dmax.size(0), H = dmax.size(1);
+    const bool bf16 = dtype == torch::kBFloat16;
+    auto dout = torch::empty({B, T, n_dir * H}, dmax.options().dtype(dtype));
+    auto stream = at::hip::getCurrentHIPStream();
+    int rc = fmda_pool_bwd_launch(bf16 ? 1 : 0, dmax.data_ptr<float>(),
+                                  davg.data_ptr<float>(), amax.data_ptr<int>(),
+                                  dout.data_ptr(), B, (int)T, H, (int)n_dir,
+                                  stream.stream());
+    TORCH_CHECK(rc == 0, "pool_bwd launch failed");
+    return dout;
+}
+
+// chunks: int64 tensor (n_chunks, 5) holding {p, g, m, v, n} built by
+// fmda_amd.optim.FusedClipAdam (device buffer; pointers as int64).
+torch::Tensor fused_clip_adam(torch::Tensor chunks, int64_t n_chunks,
+                              double clip, double lr, double beta1,
+                              double beta2, double eps, int64_t step) {
+    TORCH_CHECK(chunks.is_cuda() && chunks.scalar_type() == torch::kInt64);
+    auto norm2 = torch::zeros({1}, chunks.options().dtype(torch::kFloat32));
+    auto stream = at::hip::getCurrentHIPStream();
+    if (clip > 0) {
+        int rc = fmda_opt_norm2_launch(chunks.data_ptr(), (int)n_chunks,
+                                       norm2.data_ptr<float>(),
+                                       stream.stream());
+        TORCH_CHECK(rc == 0, "opt norm2 launch failed");
+    }
+    const float bc1 = 1.0f - powf((float)beta1, (float)step);
+    const float bc2 = 1.0f - powf((float)beta2, (float)step);
+    int rc = fmda_opt_adam_launch(chunks.data_ptr(), (int)n_chunks,
+                                  norm2.data_ptr<float>(), (float)clip,
+                                  (float)lr, (float)beta1, (float)beta2,
+                                  (float)eps, bc1, bc2, stream.stream());
+    TORCH_CHECK(rc == 0, "opt adam launch failed");
+    return norm2;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gru_fwd", &gru_fwd, "fused biGRU recurrence forward (HIP/CDNA4)");
     m.def("gru_bwd", &gru_bwd, "fused biGRU recurrence backward (HIP/CDNA4)");
     m.def("mfma_selftest", &mfma_selftest, "mfma fragment layout self-test");
+    m.def("pool_fwd", &pool_fwd, "fused dirsum+max/avg pooling forward");
+    m.def("pool_bwd", &pool_bwd, "fused pooling backward (d_out assembly)");
+    m.def("fused_clip_adam", &fused_clip_adam,
+          "fused multi-tensor grad-clip + Adam step");
 }

@@ -1446,6 +1446,89 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 }
 
 
+// ===========================================================================
+// Fused direction-sum + temporal max/avg pooling (biGRU_model.py:108-133
+// semantics: gru_out = fwd_dir + bwd_dir; max over T with argmax; sum/T).
+// One thread per (b, h): replaces four eager kernels (direction add, max
+// reduce, sum reduce, and the backward scatter) with one read / one write
+// pass over the (B, T, n_dir*H) activations.
+// ===========================================================================
+template <typename T>
+__global__ void pool_fwd_kernel(const T* __restrict__ out,
+                                float* __restrict__ maxv,
+                                float* __restrict__ avgv,
+                                int* __restrict__ amax, int B, int Tseq,
+                                int H, int n_dir) {
+    const int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= B * H) return;
+    const int b = idx / H;
+    const int h = idx % H;
+    const int HD = n_dir * H;
+    const T* p = out + (long)b * Tseq * HD + h;
+    float mx = -3.4e38f, s = 0.0f;
+    int im = 0;
+    for (int t = 0; t < Tseq; ++t) {
+        float v = to_f32<T>(p[(long)t * HD]);
+        if (n_dir == 2) v += to_f32<T>(p[(long)t * HD + H]);
+        s += v;
+        if (v > mx) { mx = v; im = t; }
+    }
+    maxv[idx] = mx;
+    avgv[idx] = s / (float)Tseq;
+    amax[idx] = im;
+}
+
+template <typename T>
+__global__ void pool_bwd_kernel(const float* __restrict__ dmax,
+                                const float* __restrict__ davg,
+                                const int* __restrict__ amax,
+                                T* __restrict__ dout, int B, int Tseq, int H,
+                                int n_dir) {
+    const int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= B * H) return;
+    const int b = idx / H;
+    const int h = idx % H;
+    const int HD = n_dir * H;
+    const float ga = davg[idx] / (float)Tseq;
+    const float gm = dmax[idx];
+    const int im = amax[idx];
+    T* p = dout + (long)b * Tseq * HD + h;
+    for (int t = 0; t < Tseq; ++t) {
+        const float g = ga + (t == im ? gm : 0.0f);
+        p[(long)t * HD] = from_f32<T>(g);
+        if (n_dir == 2) p[(long)t * HD + H] = from_f32<T>(g);
+    }
+}
+
+extern "C" int fmda_pool_fwd_launch(int is_bf16, const void* out, float* maxv,
+                                    float* avgv, int* amax, int B, int Tseq,
+                                    int H, int n_dir, hipStream_t stream) {
+    const int n = B * H;
+    const dim3 grid((n + 255) / 256);
+    if (is_bf16)
+        pool_fwd_kernel<__hip_bfloat16><<<grid, 256, 0, stream>>>(
+            (const __hip_bfloat16*)out, maxv, avgv, amax, B, Tseq, H, n_dir);
+    else
+        pool_fwd_kernel<float><<<grid, 256, 0, stream>>>(
+            (const float*)out, maxv, avgv, amax, B, Tseq, H, n_dir);
+    return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
+extern "C" int fmda_pool_bwd_launch(int is_bf16, const float* dmax,
+                                    const float* davg, const int* amax,
+                                    void* dout, int B, int Tseq, int H,
+                                    int n_dir, hipStream_t stream) {
+    const int n = B * H;
+    const dim3 grid((n + 255) / 256);
+    if (is_bf16)
+        pool_bwd_kernel<__hip_bfloat16><<<grid, 256, 0, stream>>>(
+            dmax, davg, amax, (__hip_bfloat16*)dout, B, Tseq, H, n_dir);
+    else
+        pool_bwd_kernel<float><<<grid, 256, 0, stream>>>(
+            dmax, davg, amax, (float*)dout, B, Tseq, H, n_dir);
+    return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
 // MFMA layout self-test.
 // ===========================================================================
 __global__ void mfma_selftest_kernel(const __hip_bfloat16* __restrict__ A,

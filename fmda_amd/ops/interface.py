@@ -16,7 +16,7 @@ from typing import List, Optional, Tuple
 import torch
 
 from . import load_extension
-from .blas import chunked_outer, enable_tunableop
+from .blas import chunked_colsum, chunked_outer, enable_tunableop
 
 _ALLOWED_HP = [16, 32, 64, 128, 256, 512]
 
@@ -121,8 +121,37 @@ class _IHProjection(torch.autograd.Function):
         if ctx.needs_input_grad[1]:
             dw = chunked_outer(dgi, x2d)
         if ctx.needs_input_grad[2]:
-            db = dgi.sum(dim=0)
+            db = chunked_colsum(dgi).to(dgi.dtype)
         return dx, dw, db
+
+
+class _DirSumPool(torch.autograd.Function):
+    """Fused direction-sum + max/avg temporal pooling on the HIP engine
+    (biGRU_model.py:108-133 semantics). Returns (max (B,H), avg (B,H)) in
+    the activation dtype; backward assembles the full d_out tensor in one
+    kernel instead of an eager add/scatter chain."""
+
+    @staticmethod
+    def forward(ctx, out, n_dir):
+        ext = load_extension()
+        out = out.contiguous()
+        maxv, avgv, amax = ext.pool_fwd(out, n_dir)
+        ctx.save_for_backward(amax)
+        ctx.meta = (out.shape[1], n_dir, out.dtype)
+        return maxv.to(out.dtype), avgv.to(out.dtype)
+
+    @staticmethod
+    def backward(ctx, dmax, davg):
+        ext = load_extension()
+        (amax,) = ctx.saved_tensors
+        T, n_dir, dtype = ctx.meta
+        dout = ext.pool_bwd(dmax.float().contiguous(),
+                            davg.float().contiguous(), amax, T, n_dir, dtype)
+        return dout, None
+
+
+def dirsum_pool(out: torch.Tensor, n_dir: int):
+    return _DirSumPool.apply(out, n_dir)
 
 
 def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,

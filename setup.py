@@ -23,6 +23,7 @@ setup(
             sources=[
                 "fmda_amd/ops/csrc/bindings.cpp",
                 "fmda_amd/ops/csrc/gru_kernels.hip",
+                "fmda_amd/ops/csrc/optim_kernels.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],

@@ -172,3 +172,27 @@ def test_model_train_step_gpu_bf16():
     opt.step()
     assert not torch.equal(before, m.linear.weight.detach())
     assert torch.isfinite(loss).item()
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_dirsum_pool_matches_eager(dtype):
+    """Fused pooling kernel vs the plain torch ops (fwd + bwd)."""
+    from fmda_amd.ops.interface import dirsum_pool
+    torch.manual_seed(7)
+    B, T, H = 33, 19, 24
+    out = (torch.randn(B, T, 2 * H) * 2).to(dtype).cuda().requires_grad_(True)
+    mx, av = dirsum_pool(out, 2)
+    gm = torch.randn(B, H).to(dtype).cuda()
+    ga = torch.randn(B, H).to(dtype).cuda()
+    ((mx * gm).sum() + (av * ga).sum()).backward()
+    g1 = out.grad.clone()
+    out.grad = None
+
+    s = (out[:, :, :H] + out[:, :, H:]).float()
+    mx2 = s.max(dim=1).values
+    av2 = s.sum(dim=1) / T
+    ((mx2 * gm.float()).sum() + (av2 * ga.float()).sum()).backward()
+    tol = 1e-5 if dtype == torch.float32 else 5e-2
+    assert torch.allclose(mx.float(), mx2.to(mx.dtype).float(), atol=tol)
+    assert torch.allclose(av.float(), av2.to(av.dtype).float(), atol=tol)
+    assert torch.allclose(g1.float(), out.grad.float(), atol=tol)
