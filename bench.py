@@ -164,7 +164,7 @@ def main():
     def step(i, measure_quality=False):
         nonlocal acc3_sum, n_acc
         x, y = pool[i % len(pool)]
-        opt.zero_grad(set_to_none=True)
+        opt.zero_grad(set_to_none=False)
         logits = model(x)
         loss = loss_fn(logits.float(), y)
         loss.backward()

@@ -51,8 +51,10 @@ def chunked_colsum(dg: torch.Tensor, chunks: int = 48) -> torch.Tensor:
     while c > 1 and M % c != 0:
         c //= 2
     if c <= 1:
-        return dg.float().sum(dim=0)
-    return dg.view(c, M // c, N).float().sum(dim=1).sum(dim=0)
+        return dg.sum(dim=0, dtype=torch.float32)
+    # sum(dtype=fp32) accumulates in fp32 WITHOUT materializing an fp32
+    # copy of the (M, N) input (a .float() here costs 1.5 GB of traffic).
+    return dg.view(c, M // c, N).sum(dim=1, dtype=torch.float32).sum(dim=0)
 
 
 def chunked_outer(dg: torch.Tensor, x: torch.Tensor,

@@ -17,7 +17,7 @@ import torch
 
 from .ops import load_extension
 
-_CHUNK = 1 << 16
+_CHUNK = 1 << 12
 
 
 class FusedClipAdam:
@@ -37,6 +37,7 @@ class FusedClipAdam:
         self.step_count = 0
         self.state = {}
         self._chunks: Optional[torch.Tensor] = None
+        self._ptrs = None
         self.last_norm2: Optional[torch.Tensor] = None
         for p in self.params:
             assert p.dtype == torch.float32, "fp32 masters required"
@@ -76,9 +77,13 @@ class FusedClipAdam:
         ext = load_extension()
         for p in self.params:
             assert p.grad is not None and p.grad.is_contiguous()
-        # grad storage can move between steps (set_to_none): rebuild the
-        # table each step (tiny; one H2D copy overlapped on the stream).
-        self._chunks = self._build_chunks()
+        ptrs = tuple(p.grad.data_ptr() for p in self.params)
+        if self._chunks is None or ptrs != self._ptrs:
+            # grad storage moved (e.g. zero_grad(set_to_none=True));
+            # rebuild the table. Use zero_grad(set_to_none=False) in hot
+            # loops so this happens once.
+            self._chunks = self._build_chunks()
+            self._ptrs = ptrs
         self.last_norm2 = ext.fused_clip_adam(
             self._chunks, self._chunks.shape[0], float(self.clip),
             float(self.lr), float(self.beta1), float(self.beta2),

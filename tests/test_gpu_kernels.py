@@ -188,7 +188,7 @@ def test_dirsum_pool_matches_eager(dtype):
     g1 = out.grad.clone()
     out.grad = None
 
-    s = (out[:, :, :H] + out[:, :, H:]).float()
+    s = out[:, :, :H].float() + out[:, :, H:].float()
     mx2 = s.max(dim=1).values
     av2 = s.sum(dim=1) / T
     ((mx2 * gm.float()).sum() + (av2 * ga.float()).sum()).backward()
