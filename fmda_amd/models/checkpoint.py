@@ -15,7 +15,17 @@ from .bigru import BiGRU
 
 
 def save_checkpoint(model: BiGRU, path: str) -> None:
-    torch.save(model.state_dict(), path)
+    """Write the state_dict in the reference `model_params.pt` format.
+    Uses the native C++ zip/pickle writer (ops/csrc/checkpoint.cpp) when
+    the extension is importable; the output is torch.load-compatible
+    either way."""
+    sd = model.state_dict()
+    try:
+        from ..ops import _fmda_hip
+        _fmda_hip.save_state_dict_native(path, list(sd.keys()),
+                                         [t.cpu() for t in sd.values()])
+    except ImportError:
+        torch.save(sd, path)
 
 
 def load_checkpoint(path: str, model: Optional[BiGRU] = None,

@@ -26,6 +26,14 @@ extern "C" int fmda_pool_bwd_launch(int is_bf16, const float* dmax,
                                     int n_dir, hipStream_t stream);
 extern "C" int fmda_opt_norm2_launch(const void* chunks, int n_chunks,
                                      float* out, hipStream_t stream);
+namespace fmda_ckpt {
+void save_state_dict(const std::string& path,
+                     const std::vector<std::string>& keys,
+                     const std::vector<torch::Tensor>& tensors);
+std::vector<std::pair<std::string, torch::Tensor>> load_state_dict(
+        const std::string& path);
+}  // namespace fmda_ckpt
+
 extern "C" int fmda_opt_adam_launch(const void* chunks, int n_chunks,
                                     const float* norm2, float clip, float lr,
                                     float beta1, float beta2, float eps,
@@ -189,4 +197,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pool_bwd", &pool_bwd, "fused pooling backward (d_out assembly)");
     m.def("fused_clip_adam", &fused_clip_adam,
           "fused multi-tensor grad-clip + Adam step");
+    m.def("save_state_dict_native", &fmda_ckpt::save_state_dict,
+          "write a torch.load-compatible zip state_dict (native C++)");
+    m.def("load_state_dict_native", &fmda_ckpt::load_state_dict,
+          "read a torch.save zip state_dict (native C++)");
 }

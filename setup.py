@@ -24,6 +24,7 @@ setup(
                 "fmda_amd/ops/csrc/bindings.cpp",
                 "fmda_amd/ops/csrc/gru_kernels.hip",
                 "fmda_amd/ops/csrc/optim_kernels.hip",
+                "fmda_amd/ops/csrc/checkpoint.cpp",
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],

@@ -1,0 +1,68 @@
+"""Native C++ checkpoint I/O vs torch serialization (both directions).
+
+The reference deployment depends on `model_params.pt` (torch zip format,
+predict.py:104 / notebook cell 39); the native reader/writer
+(fmda_amd/ops/csrc/checkpoint.cpp) must interoperate byte-level with
+torch.save/torch.load including the real BiGRU state_dict layout."""
+import collections
+
+import pytest
+import torch
+
+from fmda_amd.models import BiGRU
+from fmda_amd.ops import load_extension
+
+ext = pytest.importorskip("fmda_amd.ops._fmda_hip")
+
+
+def _bigru_sd():
+    torch.manual_seed(0)
+    # the shipped checkpoint architecture: H=8, F=108, C=4, bidirectional
+    return BiGRU(8, 108, 4, n_layers=1, spatial_dropout=False).state_dict()
+
+
+def test_native_save_torch_load(tmp_path):
+    sd = _bigru_sd()
+    p = str(tmp_path / "m.pt")
+    ext.save_state_dict_native(p, list(sd.keys()), list(sd.values()))
+    sd2 = torch.load(p)  # weights_only=True default must accept it
+    assert list(sd2.keys()) == list(sd.keys())
+    for k in sd:
+        assert torch.equal(sd[k], sd2[k]), k
+
+
+def test_torch_save_native_load(tmp_path):
+    sd = _bigru_sd()
+    p = str(tmp_path / "m.pt")
+    torch.save(sd, p)
+    got = dict(ext.load_state_dict_native(p))
+    assert set(got.keys()) == set(sd.keys())
+    for k in sd:
+        assert torch.equal(sd[k], got[k]), k
+
+
+def test_model_loads_native_checkpoint(tmp_path):
+    sd = _bigru_sd()
+    p = str(tmp_path / "m.pt")
+    ext.save_state_dict_native(p, list(sd.keys()), list(sd.values()))
+    m = BiGRU(8, 108, 4, n_layers=1, spatial_dropout=False)
+    m.load_state_dict(torch.load(p))
+    x = torch.randn(2, 5, 108)
+    m.eval()
+    assert m(x).shape == (2, 4)
+
+
+def test_mixed_dtypes_roundtrip(tmp_path):
+    sd = collections.OrderedDict()
+    torch.manual_seed(1)
+    sd["a"] = torch.randn(7, 3)
+    sd["b"] = torch.randn(11).to(torch.bfloat16)
+    sd["c"] = torch.arange(5, dtype=torch.int64)
+    sd["d"] = torch.randn(2, 2, 2, dtype=torch.float64)
+    p = str(tmp_path / "m.pt")
+    ext.save_state_dict_native(p, list(sd.keys()), list(sd.values()))
+    got = dict(ext.load_state_dict_native(p))
+    sd2 = torch.load(p)
+    for k in sd:
+        assert torch.equal(sd[k], got[k]), k
+        assert torch.equal(sd[k], sd2[k]), k
