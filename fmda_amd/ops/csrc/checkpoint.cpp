@@ -424,13 +424,22 @@ std::vector<std::pair<std::string, torch::Tensor>> load_state_dict(
                             "SETITEMS not on dict");
                 break;
             }
+            case 'b': {                                    // BUILD
+                // attaches state (e.g. OrderedDict._metadata) to the
+                // object below; the state never holds tensors we need.
+                stack.pop_back();
+                break;
+            }
             case '.':
                 return result;
             case '2': break;                               // DUP (unused)
             case 0x95: { need(8); i += 8; break; }         // FRAME
             default:
-                TORCH_CHECK(false, "unsupported pickle opcode 0x",
-                            std::to_string((int)op));
+                {
+                    char hx[8];
+                    snprintf(hx, sizeof hx, "%02x", (int)op);
+                    TORCH_CHECK(false, "unsupported pickle opcode 0x", hx);
+                }
         }
     }
 }
