@@ -112,9 +112,11 @@ BENCH_CONFIGS = {
     # configs[1]/[2]: repo config, 1 and 8 GPU
     "repo": BenchConfig(name="repo", hidden_size=128, n_layers=2, seq_len=120,
                         n_features=96, batch_per_gpu=4096, dtype="bf16"),
-    # configs[3]: stress
+    # configs[3]: stress. batch 2048 fills the chip: the recurrence
+    # kernels parallelize over (batch-tile, direction) blocks, and H=512
+    # at batch 256 dispatches only 32 workgroups on 256 CUs.
     "stress": BenchConfig(name="stress", hidden_size=512, n_layers=4,
-                          seq_len=512, n_features=96, batch_per_gpu=256,
+                          seq_len=512, n_features=96, batch_per_gpu=2048,
                           dtype="bf16"),
     # configs[4]: streaming inference
     "predict": BenchConfig(name="predict", hidden_size=128, n_layers=2,
