@@ -165,8 +165,16 @@ def main():
         nonlocal acc3_sum, n_acc
         x, y = pool[i % len(pool)]
         opt.zero_grad(set_to_none=False)
-        logits = model(x)
-        loss = loss_fn(logits.float(), y)
+        if use_cuda:
+            # fused head GEMM + BCEWithLogitsLoss(weight, pos_weight) kernel
+            from fmda_amd.ops.interface import fused_head_loss
+            feats = model.forward_features(x)
+            loss, logits = fused_head_loss(feats, model.linear.weight,
+                                           model.linear.bias, y, weight,
+                                           pos_weight)
+        else:
+            logits = model(x)
+            loss = loss_fn(logits.float(), y)
         loss.backward()
         if engine is not None:
             engine.finalize()

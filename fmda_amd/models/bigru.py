@@ -75,6 +75,18 @@ class BiGRU(nn.Module):
                 hidden: Optional[torch.Tensor] = None) -> torch.Tensor:
         """Forward pass; returns (B, output_size) logits
         (semantics of biGRU_model.py:63-138)."""
+        concat_out = self.forward_features(input_seq, hidden)
+        if concat_out.dtype != self.linear.weight.dtype:
+            # bf16 compute path with fp32 master weights: cast the head.
+            return F.linear(concat_out, self.linear.weight.to(concat_out.dtype),
+                            self.linear.bias.to(concat_out.dtype))
+        return self.linear(concat_out)
+
+    def forward_features(self, input_seq: torch.Tensor,
+                         hidden: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Everything before the linear head: returns the pooled concat
+        (B, 3H) feature vector — the input of the reference's linear head
+        (biGRU_model.py:133-137). Used by the fused head+loss kernel."""
         batch_size = input_seq.size(0)
         input_length = input_seq.size(1)
 
@@ -107,12 +119,7 @@ class BiGRU(nn.Module):
             max_pool = gru_out.max(dim=1).values
             avg_pool = gru_out.sum(dim=1) / float(input_length)
 
-        concat_out = torch.cat([last_hidden, max_pool, avg_pool], dim=1)
-        if concat_out.dtype != self.linear.weight.dtype:
-            # bf16 compute path with fp32 master weights: cast the head.
-            return F.linear(concat_out, self.linear.weight.to(concat_out.dtype),
-                            self.linear.bias.to(concat_out.dtype))
-        return self.linear(concat_out)
+        return torch.cat([last_hidden, max_pool, avg_pool], dim=1)
 
     def _gru_hip(self, x: torch.Tensor, hidden: Optional[torch.Tensor]):
         """CUDA path: MFMA input projections + persistent HIP recurrence."""

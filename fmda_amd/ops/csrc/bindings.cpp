@@ -24,6 +24,17 @@ extern "C" int fmda_pool_bwd_launch(int is_bf16, const float* dmax,
                                     const float* davg, const int* amax,
                                     void* dout, int B, int Tseq, int H,
                                     int n_dir, hipStream_t stream);
+extern "C" int fmda_head_fwd_launch(int is_bf16, const void* x, const void* W,
+                                    const void* bias, const float* y,
+                                    const float* wgt, const float* pw,
+                                    float* logits, float* sig,
+                                    float* loss_sum, int B, int K, int C,
+                                    hipStream_t stream);
+extern "C" int fmda_head_bwd_launch(int is_bf16, const float* sig,
+                                    const float* y, const float* wgt,
+                                    const float* pw, const float* gscale,
+                                    const void* W, float* dlogits, void* dx,
+                                    int B, int K, int C, hipStream_t stream);
 extern "C" int fmda_opt_norm2_launch(const void* chunks, int n_chunks,
                                      float* out, hipStream_t stream);
 namespace fmda_ckpt {
@@ -189,12 +200,53 @@ torch::Tensor fused_clip_adam(torch::Tensor chunks, int64_t n_chunks,
     return norm2;
 }
 
+std::vector<torch::Tensor> head_loss_fwd(torch::Tensor x, torch::Tensor W,
+                                         torch::Tensor b, torch::Tensor y,
+                                         torch::Tensor wgt, torch::Tensor pw) {
+    TORCH_CHECK(x.is_cuda() && x.is_contiguous() && W.is_contiguous());
+    const bool bf16 = x.scalar_type() == torch::kBFloat16;
+    const int B = x.size(0), K = x.size(1), C = W.size(0);
+    auto f32 = x.options().dtype(torch::kFloat32);
+    auto logits = torch::empty({B, C}, f32);
+    auto sig = torch::empty({B, C}, f32);
+    auto loss = torch::zeros({1}, f32);
+    auto stream = at::hip::getCurrentHIPStream();
+    int rc = fmda_head_fwd_launch(
+        bf16 ? 1 : 0, x.data_ptr(), W.data_ptr(), b.data_ptr(),
+        y.data_ptr<float>(), wgt.data_ptr<float>(), pw.data_ptr<float>(),
+        logits.data_ptr<float>(), sig.data_ptr<float>(),
+        loss.data_ptr<float>(), B, K, C, stream.stream());
+    TORCH_CHECK(rc == 0, "head fwd launch failed");
+    return {logits, sig, loss};
+}
+
+std::vector<torch::Tensor> head_loss_bwd(torch::Tensor sig, torch::Tensor y,
+                                         torch::Tensor wgt, torch::Tensor pw,
+                                         torch::Tensor gscale,
+                                         torch::Tensor W,
+                                         torch::ScalarType xdtype) {
+    const int B = sig.size(0), C = sig.size(1), K = W.size(1);
+    const bool bf16 = xdtype == torch::kBFloat16;
+    auto dlogits = torch::empty_like(sig);
+    auto dx = torch::empty({B, K}, sig.options().dtype(xdtype));
+    auto stream = at::hip::getCurrentHIPStream();
+    int rc = fmda_head_bwd_launch(
+        bf16 ? 1 : 0, sig.data_ptr<float>(), y.data_ptr<float>(),
+        wgt.data_ptr<float>(), pw.data_ptr<float>(),
+        gscale.data_ptr<float>(), W.data_ptr(), dlogits.data_ptr<float>(),
+        dx.data_ptr(), B, K, C, stream.stream());
+    TORCH_CHECK(rc == 0, "head bwd launch failed");
+    return {dlogits, dx};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gru_fwd", &gru_fwd, "fused biGRU recurrence forward (HIP/CDNA4)");
     m.def("gru_bwd", &gru_bwd, "fused biGRU recurrence backward (HIP/CDNA4)");
     m.def("mfma_selftest", &mfma_selftest, "mfma fragment layout self-test");
     m.def("pool_fwd", &pool_fwd, "fused dirsum+max/avg pooling forward");
     m.def("pool_bwd", &pool_bwd, "fused pooling backward (d_out assembly)");
+    m.def("head_loss_fwd", &head_loss_fwd, "fused head GEMM + BCE loss");
+    m.def("head_loss_bwd", &head_loss_bwd, "fused head/loss backward");
     m.def("fused_clip_adam", &fused_clip_adam,
           "fused multi-tensor grad-clip + Adam step");
     m.def("save_state_dict_native", &fmda_ckpt::save_state_dict,

@@ -1529,6 +1529,128 @@ extern "C" int fmda_pool_bwd_launch(int is_bf16, const float* dmax,
     return hipGetLastError() == hipSuccess ? 0 : -1;
 }
 
+// ===========================================================================
+// Fused classifier head + BCEWithLogitsLoss (weight, pos_weight).
+// Replaces nn.Linear (biGRU_model.py:137) + the notebook's
+// BCEWithLogitsLoss(weight, pos_weight) (cell 29) forward/backward chain
+// (~12 eager launches) with:
+//   fwd: logits = x @ W^T + b; s = sigmoid(logits); per-element loss
+//        l = -w[ pw*y*log(s) + (1-y)*log(1-s) ], mean-reduced to a scalar
+//        by block partials + one atomicAdd. C = #classes (4), 3H small.
+//   bwd: dlogits = gscale * w[ s(1 - y + pw*y) - pw*y ];
+//        dx = dlogits @ W (tiny K) in one elementwise-ish kernel.
+//   (dW/db reduce over B on the host via the split-K helpers.)
+// ===========================================================================
+template <typename T>
+__global__ void head_loss_fwd_kernel(const T* __restrict__ x,
+                                     const T* __restrict__ W,
+                                     const T* __restrict__ bias,
+                                     const float* __restrict__ y,
+                                     const float* __restrict__ wgt,
+                                     const float* __restrict__ pw,
+                                     float* __restrict__ logits,
+                                     float* __restrict__ sig,
+                                     float* __restrict__ loss_sum, int B,
+                                     int K, int C) {
+    const int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    float l = 0.0f;
+    if (idx < B * C) {
+        const int r = idx / C, c = idx % C;
+        const T* xr = x + (long)r * K;
+        const T* wr = W + (long)c * K;
+        float acc = to_f32<T>(bias[c]);
+        for (int k = 0; k < K; ++k)
+            acc += to_f32<T>(xr[k]) * to_f32<T>(wr[k]);
+        logits[idx] = acc;
+        const float s = sigmoidf(acc);
+        sig[idx] = s;
+        const float yy = y[idx];
+        // numerically-stable log-sigmoid forms
+        const float m = acc > 0.0f ? acc : 0.0f;
+        const float lse = m + __builtin_logf(fast_exp2(-m * FMDA_LOG2E) +
+                                             fast_exp2((acc - m) * FMDA_LOG2E));
+        const float log_s = acc - lse;      // log sigmoid(acc)
+        const float log_1ms = -lse;         // log(1 - sigmoid(acc))
+        l = -wgt[c] * (pw[c] * yy * log_s + (1.0f - yy) * log_1ms);
+    }
+    // block-reduce partials, one atomic per block
+    __shared__ float red[4];
+    float v = l;
+    v += __shfl_xor(v, 1); v += __shfl_xor(v, 2); v += __shfl_xor(v, 4);
+    v += __shfl_xor(v, 8); v += __shfl_xor(v, 16); v += __shfl_xor(v, 32);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = v;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t = 0.0f;
+        for (int w = 0; w < (int)(blockDim.x >> 6); ++w) t += red[w];
+        atomicAdd(loss_sum, t);
+    }
+}
+
+// dx[r][k] = sum_c dlogit(r,c) * W[c][k];  dlogits written for host dW/db.
+template <typename T>
+__global__ void head_loss_bwd_kernel(const float* __restrict__ sig,
+                                     const float* __restrict__ y,
+                                     const float* __restrict__ wgt,
+                                     const float* __restrict__ pw,
+                                     const float* __restrict__ gscale,
+                                     const T* __restrict__ W,
+                                     float* __restrict__ dlogits,
+                                     T* __restrict__ dx, int B, int K, int C) {
+    const int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= B * K) return;
+    const int r = idx / K, k = idx % K;
+    const float g = *gscale / (float)(B * C);
+    float acc = 0.0f;
+    for (int c = 0; c < C; ++c) {
+        const float s = sig[r * C + c];
+        const float yy = y[r * C + c];
+        const float dl = g * wgt[c] * (s * (1.0f - yy + pw[c] * yy) -
+                                       pw[c] * yy);
+        if (k == 0) dlogits[r * C + c] = dl;
+        acc += dl * to_f32<T>(W[(long)c * K + k]);
+    }
+    dx[idx] = from_f32<T>(acc);
+}
+
+extern "C" int fmda_head_fwd_launch(int is_bf16, const void* x, const void* W,
+                                    const void* bias, const float* y,
+                                    const float* wgt, const float* pw,
+                                    float* logits, float* sig,
+                                    float* loss_sum, int B, int K, int C,
+                                    hipStream_t stream) {
+    const int n = B * C;
+    const dim3 grid((n + 255) / 256);
+    if (is_bf16)
+        head_loss_fwd_kernel<__hip_bfloat16><<<grid, 256, 0, stream>>>(
+            (const __hip_bfloat16*)x, (const __hip_bfloat16*)W,
+            (const __hip_bfloat16*)bias, y, wgt, pw, logits, sig, loss_sum,
+            B, K, C);
+    else
+        head_loss_fwd_kernel<float><<<grid, 256, 0, stream>>>(
+            (const float*)x, (const float*)W, (const float*)bias, y, wgt, pw,
+            logits, sig, loss_sum, B, K, C);
+    return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
+extern "C" int fmda_head_bwd_launch(int is_bf16, const float* sig,
+                                    const float* y, const float* wgt,
+                                    const float* pw, const float* gscale,
+                                    const void* W, float* dlogits, void* dx,
+                                    int B, int K, int C, hipStream_t stream) {
+    const int n = B * K;
+    const dim3 grid((n + 255) / 256);
+    if (is_bf16)
+        head_loss_bwd_kernel<__hip_bfloat16><<<grid, 256, 0, stream>>>(
+            sig, y, wgt, pw, gscale, (const __hip_bfloat16*)W, dlogits,
+            (__hip_bfloat16*)dx, B, K, C);
+    else
+        head_loss_bwd_kernel<float><<<grid, 256, 0, stream>>>(
+            sig, y, wgt, pw, gscale, (const float*)W, dlogits, (float*)dx,
+            B, K, C);
+    return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
 // MFMA layout self-test.
 // ===========================================================================
 __global__ void mfma_selftest_kernel(const __hip_bfloat16* __restrict__ A,

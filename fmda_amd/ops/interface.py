@@ -154,6 +154,46 @@ def dirsum_pool(out: torch.Tensor, n_dir: int):
     return _DirSumPool.apply(out, n_dir)
 
 
+class _HeadLoss(torch.autograd.Function):
+    """Fused classifier head + BCEWithLogitsLoss(weight, pos_weight):
+    one kernel computes logits, sigmoid and the mean loss; backward is one
+    kernel (dlogits + dx) plus split-K dW/db reductions. Returns
+    (loss, logits); logits are non-differentiable (metrics only)."""
+
+    @staticmethod
+    def forward(ctx, x2d, W, b, y, wgt, pw):
+        ext = load_extension()
+        x2d = x2d.contiguous()
+        logits, sig, loss_sum = ext.head_loss_fwd(
+            x2d, W.contiguous(), b.contiguous(), y.contiguous(),
+            wgt.contiguous(), pw.contiguous())
+        B, C = logits.shape
+        loss = (loss_sum / float(B * C)).reshape(())
+        ctx.save_for_backward(x2d, W, y, wgt, pw, sig)
+        ctx.mark_non_differentiable(logits)
+        return loss, logits
+
+    @staticmethod
+    def backward(ctx, dloss, _dlogits):
+        ext = load_extension()
+        x2d, W, y, wgt, pw, sig = ctx.saved_tensors
+        dlogits, dx = ext.head_loss_bwd(sig, y, wgt, pw,
+                                        dloss.reshape(1).float().contiguous(),
+                                        W, x2d.dtype)
+        dl = dlogits.to(x2d.dtype)
+        dW = chunked_outer(dl, x2d)
+        db = dlogits.sum(dim=0).to(W.dtype)
+        return dx, dW, db, None, None, None
+
+
+def fused_head_loss(x2d, weight, bias, y, wgt, pw):
+    """(loss, logits) for the reference head + class-weighted BCE
+    (biGRU_model.py:137 + notebook cell 29 loss)."""
+    Wc = weight.to(x2d.dtype)
+    bc = bias.to(x2d.dtype)
+    return _HeadLoss.apply(x2d, Wc, bc, y.float(), wgt.float(), pw.float())
+
+
 def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
                 bidirectional: bool, dropout_p: float, training: bool,
                 hidden: Optional[torch.Tensor] = None

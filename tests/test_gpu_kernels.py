@@ -196,3 +196,39 @@ def test_dirsum_pool_matches_eager(dtype):
     assert torch.allclose(mx.float(), mx2.to(mx.dtype).float(), atol=tol)
     assert torch.allclose(av.float(), av2.to(av.dtype).float(), atol=tol)
     assert torch.allclose(g1.float(), out.grad.float(), atol=tol)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_head_loss_matches_eager(dtype):
+    """Fused head GEMM + BCEWithLogitsLoss(weight, pos_weight) vs the plain
+    torch ops: loss value and x/W/b gradients."""
+    from fmda_amd.ops.interface import fused_head_loss
+    torch.manual_seed(11)
+    B, K, C = 37, 96, 4
+    x = (torch.randn(B, K) * 0.5).to(dtype).cuda().requires_grad_(True)
+    W = torch.randn(C, K, device="cuda") * 0.2
+    bparm = torch.randn(C, device="cuda") * 0.1
+    W.requires_grad_(True); bparm.requires_grad_(True)
+    y = (torch.rand(B, C, device="cuda") < 0.3).float()
+    wgt = torch.rand(C, device="cuda") * 3 + 0.5
+    pw = torch.rand(C, device="cuda") * 4 + 0.5
+
+    loss1, logits1 = fused_head_loss(x, W, bparm, y, wgt, pw)
+    loss1.backward()
+    g1 = (x.grad.clone(), W.grad.clone(), bparm.grad.clone())
+    x.grad = W.grad = bparm.grad = None
+
+    lf = nn.BCEWithLogitsLoss(weight=wgt, pos_weight=pw)
+    logits2 = torch.nn.functional.linear(x.float(), W.to(dtype).float(),
+                                         bparm.to(dtype).float())
+    loss2 = lf(logits2, y)
+    loss2.backward()
+
+    tol = 1e-5 if dtype == torch.float32 else 3e-2
+    assert torch.allclose(loss1.float(), loss2, atol=tol, rtol=tol)
+    assert torch.allclose(logits1, logits2, atol=5e-2)
+    rel = lambda a, b: ((a.float() - b.float()).abs().max() /
+                        b.float().abs().max().clamp(min=1e-3))
+    assert rel(g1[0], x.grad) < (1e-4 if dtype == torch.float32 else 5e-2)
+    assert rel(g1[1], W.grad) < (1e-4 if dtype == torch.float32 else 5e-2)
+    assert rel(g1[2], bparm.grad) < (1e-4 if dtype == torch.float32 else 5e-2)
