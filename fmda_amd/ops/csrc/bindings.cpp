@@ -106,8 +106,10 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
     auto dgh = torch::empty_like(gi);
     auto dh0 = torch::empty({n_dir, B, Hp},
                             gi.options().dtype(torch::kFloat32));
-    auto dbhh = torch::zeros({n_dir, 3 * Hp},
-                             gi.options().dtype(torch::kFloat32));
+    // packed bias-grad sums: [dr, dz, dhn | dn] per direction; db_hh and
+    // db_ih are assembled from slices below (dr/dz shared).
+    auto dbsum = torch::zeros({n_dir, 4 * Hp},
+                              gi.options().dtype(torch::kFloat32));
     // W^T for the carry GEMM's B fragments (v3 kernel streams them from
     // L2 as contiguous bf16x8 rows instead of hoisting 48 VGPRs).
     torch::Tensor wt;
@@ -122,10 +124,14 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
                                  out.data_ptr(), dout.data_ptr(),
                                  dhT.data_ptr<float>(), dgi.data_ptr(),
                                  dgh.data_ptr(), dh0.data_ptr<float>(),
-                                 dbhh.data_ptr<float>(), B, T,
+                                 dbsum.data_ptr<float>(), B, T,
                                  n_dir, stream.stream());
     TORCH_CHECK(rc == 0, "fmda gru_bwd launch failed rc=", rc, " Hp=", Hp);
-    return {dgi, dgh, dh0, dbhh};
+    using torch::indexing::Slice;
+    auto dbhh = dbsum.index({Slice(), Slice(0, 3 * Hp)});
+    auto dbih = torch::cat({dbsum.index({Slice(), Slice(0, 2 * Hp)}),
+                            dbsum.index({Slice(), Slice(3 * Hp, 4 * Hp)})}, 1);
+    return {dgi, dgh, dh0, dbhh.contiguous(), dbih};
 }
 
 torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor Bm) {

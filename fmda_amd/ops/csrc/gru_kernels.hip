@@ -570,11 +570,11 @@ __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
     }
     __syncthreads();
 
-    float dbacc[CPW][3];
+    float dbacc[CPW][4];   // dr, dz, dhn (db_hh) + dn (db_ih n-slot)
 #pragma unroll
     for (int i = 0; i < CPW; ++i)
 #pragma unroll
-        for (int g = 0; g < 3; ++g) dbacc[i][g] = 0.0f;
+        for (int g = 0; g < 4; ++g) dbacc[i][g] = 0.0f;
 
     TilePrefetch<T, BT, NT, PCG> pf_gi;
     TilePrefetch<T, BT, NT, PCH> pf_hb;
@@ -693,6 +693,7 @@ __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
                     dbacc[i][0] += dr_pre;
                     dbacc[i][1] += dz_pre;
                     dbacc[i][2] += dhn;
+                    dbacc[i][3] += dn_pre;
                 }
             }
         }
@@ -792,19 +793,20 @@ __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
             d0[(long)r * Hp + (c % Hp)] = dh_s[r * HFP + (c % Hp)];
     }
 
-    // db_hh: reduce the 4 row-quarter lanes (same j = lane&15), then one
-    // atomicAdd per (gate, column) per block.
+    // bias grads: reduce the 4 row-quarter lanes (same j = lane&15), then
+    // one atomicAdd per (slot, column) per block. dbhh layout (n_dir, 4Hp):
+    // [dr, dz, dhn | dn] — db_hh = slots 0..2, db_ih = slots 0,1,3.
 #pragma unroll
     for (int i = 0; i < CPW; ++i) {
         const int ct = wave + NW * i;
         if (ct >= NCT) continue;
 #pragma unroll
-        for (int g = 0; g < 3; ++g) {
+        for (int g = 0; g < 4; ++g) {
             float v = dbacc[i][g];
             v += __shfl_xor(v, 16);
             v += __shfl_xor(v, 32);
             if ((lane >> 4) == 0)
-                atomicAdd(&dbhh[(long)dir * 3 * Hp + g * Hp + ct * 16 +
+                atomicAdd(&dbhh[(long)dir * 4 * Hp + g * Hp + ct * 16 +
                                 (lane & 15)], v);
         }
     }
@@ -1152,11 +1154,13 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
     }
 
     float dhreg[CPW][MT][4];   // dh carry (fp32, lane-owned)
-    float dbacc[CPW][3];
+    // 4 running sums: dr, dz, dhn (-> db_hh) and dn (-> db_ih; dr/dz are
+    // shared between the two bias gradients).
+    float dbacc[CPW][4];
 #pragma unroll
     for (int i = 0; i < CPW; ++i) {
 #pragma unroll
-        for (int g = 0; g < 3; ++g) dbacc[i][g] = 0.0f;
+        for (int g = 0; g < 4; ++g) dbacc[i][g] = 0.0f;
         const int ct = wave + NW * i;
         const int j = ct * 16 + (lane & 15);
         const float* hT = dhT + ((long)dir * B + b0) * Hp;
