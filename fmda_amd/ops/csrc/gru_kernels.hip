@@ -935,15 +935,17 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
         const int pb = u & 1;
         const int tt = rev ? (Tseq - 1 - u) : u;
 
-        // ---- phase A: store out[u-1], DMA gi[u+1], recurrent GEMM ----
+        // ---- phase A: DMA gi[u+1] first, then store out[u-1] (issue
+        // order matters: the B-end counted wait leaves only the store in
+        // flight), then the recurrent GEMM ----
+        if (u + 1 < Tseq)
+            glds_tile(rev ? (Tseq - 2 - u) : (u + 1), 1 - pb);
         if (u > 0) {
             const int ttp = rev ? (Tseq - u) : (u - 1);
             store_tile<__hip_bfloat16, BT, NT>(
                 hb_s[pb], out_b + (long)ttp * n_dir * Hp, Hp, WP, out_row,
                 rows_valid, tid);
         }
-        if (u + 1 < Tseq)
-            glds_tile(rev ? (Tseq - 2 - u) : (u + 1), 1 - pb);
 
         f32x4_t acc[CPW][3][MT];
 #pragma unroll
@@ -972,22 +974,11 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                             a[m], wA[i][g][kk], acc[i][g][m], 0, 0, 0);
             }
         }
-        // Drain everything older than this step's issues (FIFO): the
-        // previous step's glds must have landed before phase B reads
-        // gi_s[pb]. Counted waits keep this step's DMA in flight.
-        if (u + 1 < Tseq) {
-            if (u > 0)
-                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(PW + 1) : "memory");
-            else
-                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(PW) : "memory");
-        } else {
-            if (u > 0)
-                asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
-            else
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
+        // No barrier between A and B: the GEMM reads hb[pb] and glds
+        // writes gi_s[1-pb] while the gates (B) read gi_s[pb] and write
+        // hb[1-pb] — all buffer-disjoint. The single per-step barrier
+        // lives at B-end and doubles as the DMA rendezvous for the tile
+        // issued THIS step (read next step).
 
         // ---- phase B: fused gates; h stays in registers ----
 #pragma unroll
@@ -1017,6 +1008,13 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                 }
             }
         }
+        // rendezvous: drain this step's glds (tile u+1) across all waves,
+        // leaving the out[u-1] store in flight (it was issued after the
+        // glds, so the counted wait never stalls on store completion).
+        if (u > 0)
+            asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
     }
@@ -1113,8 +1111,10 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
     __hip_bfloat16* hb_s1 = (__hip_bfloat16*)p; p += BT * HROW;
     __hip_bfloat16* do_s0 = (__hip_bfloat16*)p; p += BT * HROW;
     __hip_bfloat16* do_s1 = (__hip_bfloat16*)p; p += BT * HROW;
-    __hip_bfloat16* dgh_s = (__hip_bfloat16*)p; p += 2 * BT * GP3;
+    __hip_bfloat16* dgh_s0 = (__hip_bfloat16*)p; p += 2 * BT * GP3;
+    __hip_bfloat16* dgh_s1 = (__hip_bfloat16*)p; p += 2 * BT * GP3;
     float* bhh_s = (float*)p;
+    auto dgh_buf = [&](int b) { return b ? dgh_s1 : dgh_s0; };
     auto gi_slot = [&](int sl) {
         return sl == 0 ? gi_s0 : (sl == 1 ? gi_s1 : gi_s2);
     };
@@ -1264,34 +1264,19 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                 zero_hb(1 - q);
         }
         if (have_prev) {
+            // one-step-late stores; dgh_s is double-buffered so these
+            // reads never race phase B's writes and no A-end barrier is
+            // needed (the single per-step barrier is at B-end, where it
+            // also rendezvouses this step's DMA).
             const int ttp = rev ? (Tseq - 2 - u) : (u + 1);
             store_tile<__hip_bfloat16, BT, NT>(
                 gi_slot((u + 1) % 3), dgi_b + (long)ttp * n_dir * 3 * Hp, GP, GP,
                 gi_row, rows_valid, tid);
             const int sh = rev ? (ttp + 1) : (ttp - 1);  // always in range
             store_tile<__hip_bfloat16, BT, NT>(
-                dgh_s, dgh_b + (long)sh * n_dir * 3 * Hp, GP, GP3, gi_row,
-                rows_valid, tid);
+                dgh_buf(1 - q), dgh_b + (long)sh * n_dir * 3 * Hp, GP, GP3,
+                gi_row, rows_valid, tid);
         }
-
-        // Rendezvous: drain everything older than this step's issues so
-        // the PREVIOUS step's DMA is visible to phase B; keep this step's
-        // glds (and trailing stores) in flight.
-        if (have_next) {
-            if (have_prev)
-                asm volatile("s_waitcnt vmcnt(%0)"
-                             ::"i"(PW + DOP + HBP + NST) : "memory");
-            else
-                asm volatile("s_waitcnt vmcnt(%0)"
-                             ::"i"(PW + DOP + HBP) : "memory");
-        } else {
-            if (have_prev)
-                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NST) : "memory");
-            else
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
 
         // ---- phase B: recompute GEMM, then fused gate gradients ----
         const __hip_bfloat16* hbq = hb_buf(q);
@@ -1328,7 +1313,7 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
             const int ct = wave + NW * i;
             const int j = ct * 16 + (lane & 15);
             __bf16* grow = (__bf16*)gi_slot(slot);
-            __bf16* dgrow = (__bf16*)dgh_s;
+            __bf16* dgrow = (__bf16*)dgh_buf(q);
             const __bf16* dorow = (const __bf16*)do_buf(q);
 #pragma unroll
             for (int m = 0; m < MT; ++m) {
@@ -1370,8 +1355,20 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                     dbacc[i][0] += dr_pre;
                     dbacc[i][1] += dz_pre;
                     dbacc[i][2] += dhn;
+                    dbacc[i][3] += dn_pre;
                 }
             }
+        }
+        // single per-step barrier: publishes dgh_s[q] for phase C AND
+        // rendezvouses this step's DMA (tiles for step u-1, read at
+        // B(u-1)), leaving the trailing dGi/dGh stores in flight.
+        if (have_next) {
+            if (have_prev)
+                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NST) : "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        } else {
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
@@ -1394,7 +1391,7 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 #pragma unroll
                     for (int m = 0; m < MT; ++m)
                         a[m] = *(const bf16x8_t*)&((const __bf16*)
-                                   dgh_s)[(16 * m + arow) * GP3 + kbase];
+                                   dgh_buf(q))[(16 * m + arow) * GP3 + kbase];
 #pragma unroll
                     for (int m = 0; m < MT; ++m)
                         acc2[i][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -1437,12 +1434,14 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                     if (b < rows_valid) d0[(long)b * Hp + j] = dhreg[i][m][e];
                 }
 #pragma unroll
-            for (int g = 0; g < 3; ++g) {
+            for (int g = 0; g < 4; ++g) {
+                // dbhh layout (n_dir, 4Hp): [dr, dz, dhn | dn] — db_hh =
+                // slots 0..2, db_ih = slots 0,1,3 (assembled host-side).
                 float v = dbacc[i][g];
                 v += __shfl_xor(v, 16);
                 v += __shfl_xor(v, 32);
                 if ((lane >> 4) == 0)
-                    atomicAdd(&dbhh[(long)dir * 3 * Hp + g * Hp + ct * 16 +
+                    atomicAdd(&dbhh[(long)dir * 4 * Hp + g * Hp + ct * 16 +
                                     (lane & 15)], v);
             }
         }
@@ -1786,7 +1785,8 @@ static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
                              int n_dir, hipStream_t stream) {
     constexpr int BT = 32, Hp = 128, NT = 256;
     const size_t lds = 3 * 2 * BT * 3 * Hp + 2 * 2 * BT * Hp +
-                       2 * 2 * BT * Hp + 2 * BT * (3 * Hp + 8) + 4 * 3 * Hp;
+                       2 * 2 * BT * Hp + 2 * 2 * BT * (3 * Hp + 8) +
+                       4 * 3 * Hp;
     // 4 waves at 1 wave/SIMD: this shape is spill-free (NT=512 at 2
     // waves/SIMD spills 3 VGPRs, and every spill reload re-triggers the
     // glds vmcnt(0) drain).
