@@ -8,6 +8,11 @@ extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const float* bhh, void* out,
                                    float* hlast, int B, int Tseq, int n_dir,
                                    hipStream_t stream);
+extern "C" int fmda_gru_fwd_cs_launch(const void* gi, const void* w,
+                                      const float* bhh, void* out,
+                                      float* hlast, void* hpub,
+                                      unsigned int* cnt, int B, int Tseq,
+                                      int n_dir, hipStream_t stream);
 extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const void* wt,
                                    const float* bhh,
@@ -85,10 +90,25 @@ std::vector<torch::Tensor> gru_fwd(torch::Tensor gi, torch::Tensor w,
     auto hlast = torch::empty({n_dir, B, Hp},
                               gi.options().dtype(torch::kFloat32));
     auto stream = at::hip::getCurrentHIPStream();
-    int rc = fmda_gru_fwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
+    int rc;
+    if (is_bf16 && Hp == 512) {
+        // column-split persistent kernel: publication ring + group counters
+        const int BR = 256;
+        const int GB = (B + BR - 1) / BR;
+        const int G = GB * n_dir;
+        auto hpub = torch::empty({2, G, BR, (int64_t)Hp}, gi.options());
+        auto cnt = torch::zeros({G}, gi.options().dtype(torch::kUInt32));
+        rc = fmda_gru_fwd_cs_launch(gi.data_ptr(), w.data_ptr(),
+                                    bhh.data_ptr<float>(), out.data_ptr(),
+                                    hlast.data_ptr<float>(), hpub.data_ptr(),
+                                    (unsigned int*)cnt.data_ptr(), B, T,
+                                    n_dir, stream.stream());
+    } else {
+        rc = fmda_gru_fwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
                                  w.data_ptr(), bhh.data_ptr<float>(),
                                  out.data_ptr(), hlast.data_ptr<float>(), B, T,
                                  n_dir, stream.stream());
+    }
     TORCH_CHECK(rc == 0, "fmda gru_fwd launch failed rc=", rc, " Hp=", Hp);
     return {out, hlast};
 }

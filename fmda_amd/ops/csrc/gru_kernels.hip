@@ -1450,6 +1450,323 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 
 
 // ===========================================================================
+// Column-split persistent forward for LARGE hidden sizes (Hp = 512).
+//
+// At H = 512 the batch-parallel kernels above starve the chip (W_hh is
+// 1.5 MB/direction — far past LDS and the register file — so every block
+// re-streams it from L2 each step, and B/BT blocks can't fill 256 CUs).
+// This kernel splits the HIDDEN dimension instead (SURVEY.md section 7
+// "hard part (a)"): a GROUP of CT=16 column-blocks owns one (batch-rows,
+// direction) pair; each block keeps its 96-gate-row slice of W_hh
+// row-major in LDS (~98 KB, read as contiguous MFMA B-fragments) and owns
+// 32 h columns. Per timestep every block:
+//   1. polls the group's monotonic arrival counter (relaxed agent load)
+//      until all CT slices of h_{t-1} are published,
+//   2. MFMA-multiplies the FULL h_{t-1} (asm sc1 b128 loads from the
+//      publication ring — sc1 both sides is the placement-independent
+//      visibility recipe; block->XCD placement is speed-only),
+//   3. fuses the gates for its 32 columns (own h carried in fp32 regs),
+//   4. publishes h_t: LDS stage -> cooperative 16-B sc1 stores ->
+//      s_waitcnt vmcnt(0) -> one relaxed agent atomicAdd per block.
+// The 2-deep publication ring is safe because the counter bounds block
+// skew to one step: a block enters step t only after every peer published
+// t-1, so nobody can overwrite a slot a peer still reads.
+// gi tiles stage through LDS one step ahead (ordinary loads; this kernel
+// has NO LDS-DMA, so the ROCm 7.2 glds/ordinary-load drain hazard does
+// not apply).
+// ===========================================================================
+
+FMDA_DEV bf16x8_t load16_sc1(const void* p) {
+    bf16x8_t v;
+    asm volatile("global_load_dwordx4 %0, %1, off sc0 sc1"
+                 : "=v"(v) : "v"(p));
+    return v;
+}
+typedef unsigned int u32x4_t __attribute__((ext_vector_type(4)));
+
+FMDA_DEV void store16_sc1(void* p, u32x4_t v) {
+    asm volatile("global_store_dwordx4 %0, %1, off sc0 sc1\ns_nop 1"
+                 :: "v"(p), "v"(v) : "memory");
+}
+
+template <int BR, int Hp, int CS, int NT>
+__global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
+                          amdgpu_waves_per_eu(2, 2)))
+void gru_fwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
+                       const __hip_bfloat16* __restrict__ w,
+                       const float* __restrict__ bhh,
+                       __hip_bfloat16* __restrict__ out,
+                       float* __restrict__ hlast,
+                       __hip_bfloat16* __restrict__ hpub,  // (2,G,BR,Hp)
+                       unsigned int* __restrict__ cnt,     // (G)
+                       int B, int Tseq, int n_dir, int GB) {
+    constexpr int CT = Hp / CS;          // column-blocks per group
+    constexpr int NW = NT / 64;          // waves
+    constexpr int MT = BR / 16 / NW;     // m-tiles per wave
+    constexpr int NCT = 3 * CS / 16;     // gate-column tiles (all per wave)
+    constexpr int KK = Hp / 32;
+    constexpr int WPITCH = Hp + 8;       // W slice LDS pitch (elements)
+    constexpr int GP = 3 * CS;           // gi tile pitch (elements)
+    static_assert(BR % (16 * NW) == 0, "rows must tile");
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    // id -> (group, member): member-major within an XCD (id % 8 observed
+    // = XCD) so one group's 16 members share an L2. Speed-only.
+    const int id = blockIdx.x;
+    const int nG = GB * n_dir;
+    const int xcd = id % 8;
+    const int slot_in_xcd = id / 8;
+    int g, ct;
+    if (nG >= 8 && (nG % 8) == 0) {
+        ct = slot_in_xcd % CT;
+        g = xcd + 8 * (slot_in_xcd / CT);
+    } else {                              // small grids: plain mapping
+        g = id / CT;
+        ct = id % CT;
+    }
+    const int dir = g / GB;
+    const int gb = g % GB;
+    const bool rev = (dir == 1);
+    const int b0 = gb * BR;
+    const int rows_valid = min(BR, B - b0);
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    char* p = smem;
+    __hip_bfloat16* w_s = (__hip_bfloat16*)p;   // (3CS, WPITCH)
+    p += 2 * 3 * CS * WPITCH;
+    __hip_bfloat16* gi_s = (__hip_bfloat16*)p;  // (BR, GP), also h stage
+    p += 2 * BR * GP;
+    float* bhh_s = (float*)p;
+
+    const long gi_row = (long)Tseq * n_dir * 3 * Hp;
+    const long out_row = (long)Tseq * n_dir * Hp;
+    // this block's gate rows: dir block, gate gslices at ct*CS
+    const __hip_bfloat16* wdir = w + (long)dir * 3 * Hp * Hp;
+    for (int g3 = 0; g3 < 3; ++g3) {
+        stage_tile<__hip_bfloat16, CS, NT>(
+            w_s + g3 * CS * WPITCH, wdir + (long)(g3 * Hp + ct * CS) * Hp,
+            Hp, WPITCH, Hp, CS, tid);
+        for (int c = tid; c < CS; c += NT)
+            bhh_s[g3 * CS + c] = bhh[(long)dir * 3 * Hp + g3 * Hp + ct * CS + c];
+    }
+
+    const __hip_bfloat16* gi_b =
+        gi + (long)b0 * gi_row + (long)dir * 3 * Hp;
+    __hip_bfloat16* out_b = out + (long)b0 * out_row + (long)dir * Hp;
+    __hip_bfloat16* pub_base = hpub + (long)g * BR * Hp;   // + ring*G*BR*Hp
+    const long ring_stride = (long)nG * BR * Hp;
+
+    // stage gi tile for step tt into gi_s (this block's 3*CS gate columns,
+    // in r|z|n blocks of CS)
+    auto stage_gi = [&](int tt) {
+        const __hip_bfloat16* src =
+            gi_b + (long)tt * n_dir * 3 * Hp;
+        for (int g3 = 0; g3 < 3; ++g3)
+            stage_tile<__hip_bfloat16, BR, NT>(
+                gi_s + g3 * CS, src + g3 * Hp + ct * CS, CS, GP,
+                gi_row, rows_valid, tid);
+    };
+
+    // own h (fp32) carried in registers: lane owns (b, jj) for its C tiles
+    float hreg[NCT / 3][MT][4];   // jj tiles per gate = CS/16 = NCT/3
+    constexpr int JT = CS / 16;
+#pragma unroll
+    for (int i = 0; i < JT; ++i)
+#pragma unroll
+        for (int m = 0; m < MT; ++m)
+#pragma unroll
+            for (int e = 0; e < 4; ++e) hreg[i][m][e] = 0.0f;
+
+    {   // prologue: publish h_{-1} = 0 into ring slot 0; stage gi(step 0)
+        stage_gi(rev ? (Tseq - 1) : 0);
+        __hip_bfloat16* dst = pub_base;            // ring slot 0
+        const u32x4_t z = {0, 0, 0, 0};
+        for (int c = tid; c < BR * CS * 2 / 16; c += NT) {
+            const int r = c / (CS * 2 / 16);
+            const int jc = c % (CS * 2 / 16);
+            store16_sc1((char*)(dst + (long)r * Hp + ct * CS) + jc * 16, z);
+        }
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __syncthreads();
+        if (tid == 0)
+            __hip_atomic_fetch_add(&cnt[g], 1u, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+        __syncthreads();
+    }
+
+    for (int u = 0; u < Tseq; ++u) {
+        const int tt = rev ? (Tseq - 1 - u) : u;
+        const unsigned want = (unsigned)CT * (u + 1);
+
+        // ---- wait for all of h_{u-1} ----
+        // The poll is BOUNDED: if a peer block were never scheduled the
+        // kernel must terminate (garbage output) rather than wedge the
+        // GPU. The grid is sized to exactly fit co-resident (1 block/CU),
+        // so the cap should never trip in practice.
+        if (tid == 0) {
+            long spins = 0;
+            while (__hip_atomic_load(&cnt[g], __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT) < want) {
+                __builtin_amdgcn_s_sleep(8);
+                if (++spins > (1L << 24)) break;
+            }
+        }
+        __syncthreads();
+
+        // ---- load h_{u-1} fragments (sc1) and run the recurrent GEMM ----
+        const __hip_bfloat16* hsrc =
+            pub_base + (long)(u & 1) * ring_stride;   // slot of h_{u-1}
+        f32x4_t acc[NCT][MT];
+#pragma unroll
+        for (int i = 0; i < NCT; ++i)
+#pragma unroll
+            for (int m = 0; m < MT; ++m) acc[i][m] = f32x4_t{0.f};
+        {
+            const int arow = lane & 15;
+            const int koff = 8 * (lane >> 4);
+            // issue ALL h fragment loads, then one drain: a wait per
+            // k-slice would expose L2 latency 16x per step
+            bf16x8_t a[KK][MT];
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk)
+#pragma unroll
+                for (int m = 0; m < MT; ++m) {
+                    const int row = 16 * (wave + NW * m) + arow;
+                    a[kk][m] = load16_sc1(hsrc + (long)row * Hp + 32 * kk +
+                                          koff);
+                }
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                const int kbase = 32 * kk + koff;
+#pragma unroll
+                for (int i = 0; i < NCT; ++i) {
+                    const bf16x8_t b = *(const bf16x8_t*)&(
+                        (const __bf16*)w_s)[(i * 16 + (lane & 15)) * WPITCH +
+                                            kbase];
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+                        acc[i][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[kk][m], b, acc[i][m], 0, 0, 0);
+                }
+            }
+        }
+
+        // ---- fused gates for the block's CS columns ----
+        float hnew_keep[JT][MT][4];
+#pragma unroll
+        for (int j = 0; j < JT; ++j) {
+            const int jj = j * 16 + (lane & 15);       // 0..CS
+#pragma unroll
+            for (int m = 0; m < MT; ++m) {
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int b = 16 * (wave + NW * m) + 4 * (lane >> 4) + e;
+                    const float gr = acc[j][m][e] + bhh_s[jj];
+                    const float gz = acc[JT + j][m][e] + bhh_s[CS + jj];
+                    const float hn = acc[2 * JT + j][m][e] + bhh_s[2 * CS + jj];
+                    const float ir = (float)((const __bf16*)gi_s)[b * GP + jj];
+                    const float iz =
+                        (float)((const __bf16*)gi_s)[b * GP + CS + jj];
+                    const float in_ =
+                        (float)((const __bf16*)gi_s)[b * GP + 2 * CS + jj];
+                    const float r = sigmoidf(ir + gr);
+                    const float z = sigmoidf(iz + gz);
+                    const float n = fast_tanh(in_ + r * hn);
+                    const float hnew = (1.0f - z) * n + z * hreg[j][m][e];
+                    hreg[j][m][e] = hnew;
+                    hnew_keep[j][m][e] = hnew;
+                }
+            }
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __syncthreads();   // gi_s reads done; reuse its space as h stage
+
+        // ---- stage h_t slice in LDS (BR x CS bf16), write out + publish --
+        __hip_bfloat16* hst = gi_s;   // (BR, CS) in the gi buffer's space
+#pragma unroll
+        for (int j = 0; j < JT; ++j) {
+            const int jj = j * 16 + (lane & 15);
+#pragma unroll
+            for (int m = 0; m < MT; ++m)
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int b = 16 * (wave + NW * m) + 4 * (lane >> 4) + e;
+                    ((__bf16*)hst)[b * CS + jj] =
+                        (__bf16)__float2bfloat16(hnew_keep[j][m][e]);
+                }
+        }
+        __syncthreads();
+        // out[t] slice (plain stores) + publication (sc1 stores)
+        store_tile<__hip_bfloat16, BR, NT>(
+            hst, out_b + (long)tt * n_dir * Hp + ct * CS, CS, CS, out_row,
+            rows_valid, tid);
+        {
+            __hip_bfloat16* dst = pub_base + (long)((u + 1) & 1) * ring_stride;
+            constexpr int CPR = CS * 2 / 16;
+            for (int c = tid; c < BR * CPR; c += NT) {
+                const int r = c / CPR;
+                const int jc = c % CPR;
+                u32x4_t v = *(const u32x4_t*)((const char*)hst +
+                                              (long)r * CS * 2 + jc * 16);
+                store16_sc1((char*)(dst + (long)r * Hp + ct * CS) + jc * 16,
+                            v);
+            }
+        }
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __syncthreads();
+        if (tid == 0)
+            __hip_atomic_fetch_add(&cnt[g], 1u, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+        // stage gi for the next step (overwrites the h stage; safe: the
+        // publish reads above finished before this barrier)
+        if (u + 1 < Tseq)
+            stage_gi(rev ? (Tseq - 2 - u) : (u + 1));
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __syncthreads();
+    }
+
+    {   // epilogue: hlast (fp32) for the block's columns
+        float* hl = hlast + ((long)dir * B + b0) * Hp + ct * CS;
+#pragma unroll
+        for (int j = 0; j < JT; ++j) {
+            const int jj = j * 16 + (lane & 15);
+#pragma unroll
+            for (int m = 0; m < MT; ++m)
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int b = 16 * (wave + NW * m) + 4 * (lane >> 4) + e;
+                    if (b < rows_valid) hl[(long)b * Hp + jj] = hreg[j][m][e];
+                }
+        }
+    }
+}
+
+extern "C" int fmda_gru_fwd_cs_launch(const void* gi, const void* w,
+                                      const float* bhh, void* out,
+                                      float* hlast, void* hpub,
+                                      unsigned int* cnt, int B, int Tseq,
+                                      int n_dir, hipStream_t stream) {
+    constexpr int BR = 256, Hp = 512, CS = 32, NT = 512;
+    const int GB = (B + BR - 1) / BR;
+    const size_t lds = 2 * 3 * CS * (Hp + 8) + 2 * (size_t)BR * 3 * CS +
+                       4 * 3 * CS;
+    auto k = gru_fwd_cs_kernel<BR, Hp, CS, NT>;
+    (void)hipFuncSetAttribute((const void*)k,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    const dim3 grid(GB * (Hp / CS) * n_dir);
+    k<<<grid, NT, lds, stream>>>((const __hip_bfloat16*)gi,
+                                 (const __hip_bfloat16*)w, bhh,
+                                 (__hip_bfloat16*)out, hlast,
+                                 (__hip_bfloat16*)hpub, cnt, B, Tseq, n_dir,
+                                 GB);
+    return 0;
+}
+
+// ===========================================================================
 // Fused direction-sum + temporal max/avg pooling (biGRU_model.py:108-133
 // semantics: gru_out = fwd_dir + bwd_dir; max over T with argmax; sum/T).
 // One thread per (b, h): replaces four eager kernels (direction add, max
