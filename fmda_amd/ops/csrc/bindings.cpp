@@ -13,6 +13,13 @@ extern "C" int fmda_gru_fwd_cs_launch(const void* gi, const void* w,
                                       float* hlast, void* hpub,
                                       unsigned int* cnt, int B, int Tseq,
                                       int n_dir, hipStream_t stream);
+extern "C" int fmda_gru_bwd_cs_launch(const void* gi, const void* w,
+                                      const void* wt, const float* bhh,
+                                      const void* out, const void* dout,
+                                      const float* dhT, void* dgi, void* dgh,
+                                      float* dh0, float* dbhh, void* gpub,
+                                      unsigned int* cnt, int B, int Tseq,
+                                      int n_dir, hipStream_t stream);
 extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const void* wt,
                                    const float* bhh,
@@ -134,18 +141,33 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
     // L2 as contiguous bf16x8 rows instead of hoisting 48 VGPRs).
     torch::Tensor wt;
     const void* wt_ptr = nullptr;
-    if (is_bf16 && Hp == 128) {
+    if (is_bf16 && (Hp == 128 || Hp == 512)) {
         wt = w.transpose(1, 2).contiguous();
         wt_ptr = wt.data_ptr();
     }
     auto stream = at::hip::getCurrentHIPStream();
-    int rc = fmda_gru_bwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
+    int rc;
+    if (is_bf16 && Hp == 512) {
+        const int BR = 256;
+        const int GB = (B + BR - 1) / BR;
+        const int G = GB * n_dir;
+        auto gpub = torch::empty({2, G, BR, (int64_t)3 * Hp}, gi.options());
+        auto cnt = torch::zeros({G}, gi.options().dtype(torch::kUInt32));
+        rc = fmda_gru_bwd_cs_launch(
+            gi.data_ptr(), w.data_ptr(), wt_ptr, bhh.data_ptr<float>(),
+            out.data_ptr(), dout.data_ptr(), dhT.data_ptr<float>(),
+            dgi.data_ptr(), dgh.data_ptr(), dh0.data_ptr<float>(),
+            dbsum.data_ptr<float>(), gpub.data_ptr(),
+            (unsigned int*)cnt.data_ptr(), B, T, n_dir, stream.stream());
+    } else {
+        rc = fmda_gru_bwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
                                  w.data_ptr(), wt_ptr, bhh.data_ptr<float>(),
                                  out.data_ptr(), dout.data_ptr(),
                                  dhT.data_ptr<float>(), dgi.data_ptr(),
                                  dgh.data_ptr(), dh0.data_ptr<float>(),
                                  dbsum.data_ptr<float>(), B, T,
                                  n_dir, stream.stream());
+    }
     TORCH_CHECK(rc == 0, "fmda gru_bwd launch failed rc=", rc, " Hp=", Hp);
     using torch::indexing::Slice;
     auto dbhh = dbsum.index({Slice(), Slice(0, 3 * Hp)});

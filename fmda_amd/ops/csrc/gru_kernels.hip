@@ -1484,6 +1484,14 @@ FMDA_DEV bf16x8_t load16_sc1(const void* p) {
 }
 typedef unsigned int u32x4_t __attribute__((ext_vector_type(4)));
 
+FMDA_DEV bf16x8_t load16_asm(const void* p) {
+    // plain 16-B load via asm: invisible to the compiler's waitcnt
+    // bookkeeping, so it can share a hand-counted batch with the sc1 loads
+    bf16x8_t v;
+    asm volatile("global_load_dwordx4 %0, %1, off" : "=v"(v) : "v"(p));
+    return v;
+}
+
 FMDA_DEV void store16_sc1(void* p, u32x4_t v) {
     asm volatile("global_store_dwordx4 %0, %1, off sc0 sc1\ns_nop 1"
                  :: "v"(p), "v"(v) : "memory");
@@ -1764,6 +1772,419 @@ extern "C" int fmda_gru_fwd_cs_launch(const void* gi, const void* w,
                                  (__hip_bfloat16*)hpub, cnt, B, Tseq, n_dir,
                                  GB);
     return 0;
+}
+
+// ===========================================================================
+// Column-split persistent BPTT for Hp = 512 (design notes on the forward
+// kernel above apply). The exchanged quantity here is dGh: the carry GEMM
+// dh_j += sum_n dGh[n] W[n][j] needs ALL 3Hp gate rows, so each block
+// publishes its three 32-column dGh strips into a (BR, 3Hp) ring image
+// that readers consume as contiguous MFMA A-fragments. Per step:
+//   1. poll for the full dGh(u+1) set; carry GEMM (A = sc1 asm loads from
+//      the ring, software-pipelined in 8-k-slice chunks; B = W^T fragments
+//      streamed from the XCD L2 as ordinary loads — legal, no glds here),
+//   2. recompute GEMM gh = out(u-1) @ W_A^T (W_A slice LDS-resident,
+//      out read with plain loads - it is a static kernel input),
+//   3. fused gate grads for the block's 32 columns (dout/h_prev scalars
+//      register-prefetched one step ahead),
+//   4. dGi strip stores, then the n-slot swap (dn -> dhn), dGh strip
+//      stores + sc1 publication + arrival atomic.
+// dir = group & 1 so each XCD hosts one direction (W + W^T fit its L2).
+// ===========================================================================
+template <int BR, int Hp, int CS, int NT>
+__global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
+                          amdgpu_waves_per_eu(2, 2)))
+void gru_bwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
+                       const __hip_bfloat16* __restrict__ w,
+                       const __hip_bfloat16* __restrict__ wt,
+                       const float* __restrict__ bhh,
+                       const __hip_bfloat16* __restrict__ out,
+                       const __hip_bfloat16* __restrict__ dout,
+                       const float* __restrict__ dhT,
+                       __hip_bfloat16* __restrict__ dgi,
+                       __hip_bfloat16* __restrict__ dgh,
+                       float* __restrict__ dh0, float* __restrict__ dbhh,
+                       __hip_bfloat16* __restrict__ gpub,  // (2,G,BR,3Hp)
+                       unsigned int* __restrict__ cnt, int B, int Tseq,
+                       int n_dir, int GB) {
+    constexpr int CT = Hp / CS;
+    constexpr int NW = NT / 64;
+    constexpr int MT = BR / 16 / NW;
+    constexpr int NCT = 3 * CS / 16;     // 6 gate-column tiles
+    constexpr int JT = CS / 16;          // 2 h-column tiles
+    constexpr int KK = Hp / 32;          // 16 (recompute GEMM K)
+    constexpr int KK2 = 3 * Hp / 32;     // 48 (carry GEMM K)
+    constexpr int KCH = 2;               // carry-GEMM k-slices per chunk
+    constexpr int WPITCH = Hp + 8;
+    constexpr int GP = 3 * CS;           // LDS gi/stage pitch
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int id = blockIdx.x;
+    const int nG = GB * n_dir;
+    int g, ct;
+    if (nG >= 8 && (nG % 8) == 0) {
+        ct = (id / 8) % CT;
+        g = (id % 8) + 8 * ((id / 8) / CT);
+    } else {
+        g = id / CT;
+        ct = id % CT;
+    }
+    const int dir = (n_dir == 2) ? (g & 1) : 0;   // one direction per XCD
+    const int gb = (n_dir == 2) ? (g >> 1) : g;
+    const bool rev = (dir == 1);
+    const int b0 = gb * BR;
+    const int rows_valid = min(BR, B - b0);
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    char* p = smem;
+    __hip_bfloat16* w_s = (__hip_bfloat16*)p;   // W_A slice (3CS, WPITCH)
+    p += 2 * 3 * CS * WPITCH;
+    __hip_bfloat16* gi_s = (__hip_bfloat16*)p;  // (BR, GP); also dGi/dGh stage
+    p += 2 * BR * GP;
+    float* bhh_s = (float*)p;
+
+    const long gi_row = (long)Tseq * n_dir * 3 * Hp;
+    const long out_row = (long)Tseq * n_dir * Hp;
+    const __hip_bfloat16* wdir = w + (long)dir * 3 * Hp * Hp;
+    const __hip_bfloat16* wt_dir = wt + (long)dir * Hp * 3 * Hp;
+    for (int g3 = 0; g3 < 3; ++g3) {
+        stage_tile<__hip_bfloat16, CS, NT>(
+            w_s + g3 * CS * WPITCH, wdir + (long)(g3 * Hp + ct * CS) * Hp,
+            Hp, WPITCH, Hp, CS, tid);
+        for (int c = tid; c < CS; c += NT)
+            bhh_s[g3 * CS + c] =
+                bhh[(long)dir * 3 * Hp + g3 * Hp + ct * CS + c];
+    }
+
+    const __hip_bfloat16* gi_b = gi + (long)b0 * gi_row + (long)dir * 3 * Hp;
+    __hip_bfloat16* dgi_b = dgi + (long)b0 * gi_row + (long)dir * 3 * Hp;
+    __hip_bfloat16* dgh_b = dgh + (long)b0 * gi_row + (long)dir * 3 * Hp;
+    const __hip_bfloat16* out_b = out + (long)b0 * out_row + (long)dir * Hp;
+    const __hip_bfloat16* dout_b = dout + (long)b0 * out_row + (long)dir * Hp;
+    __hip_bfloat16* pub_base = gpub + (long)g * BR * 3 * Hp;
+    const long ring_stride = (long)nG * BR * 3 * Hp;
+
+    auto stage_gi = [&](int tt) {
+        const __hip_bfloat16* src = gi_b + (long)tt * n_dir * 3 * Hp;
+        for (int g3 = 0; g3 < 3; ++g3)
+            stage_tile<__hip_bfloat16, BR, NT>(
+                gi_s + g3 * CS, src + g3 * Hp + ct * CS, CS, GP, gi_row,
+                rows_valid, tid);
+    };
+
+    // carry + per-lane scalars for the block's own columns
+    float dhcar[JT][MT][4];
+    float dbacc[JT][4];
+#pragma unroll
+    for (int j = 0; j < JT; ++j)
+#pragma unroll
+        for (int q4 = 0; q4 < 4; ++q4) dbacc[j][q4] = 0.0f;
+    {
+        const float* hT = dhT + ((long)dir * B + b0) * Hp + ct * CS;
+#pragma unroll
+        for (int j = 0; j < JT; ++j) {
+            const int jj = j * 16 + (lane & 15);
+#pragma unroll
+            for (int m = 0; m < MT; ++m)
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int b = 16 * (wave + NW * m) + 4 * (lane >> 4) + e;
+                    dhcar[j][m][e] =
+                        (b < rows_valid) ? hT[(long)b * Hp + jj] : 0.0f;
+                }
+        }
+    }
+
+    {   // prologue: stage gi(T-1)
+        stage_gi(rev ? 0 : (Tseq - 1));
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __syncthreads();
+    }
+
+    for (int u = Tseq - 1; u >= 0; --u) {
+        const int tt = rev ? (Tseq - 1 - u) : u;
+        const bool have_prev = (u + 1 < Tseq);
+
+        // ---- carry GEMM: dh += dGh(u+1) @ W^T (own columns) ----
+        if (have_prev) {
+            const unsigned want = (unsigned)CT * (Tseq - 1 - u);
+            if (tid == 0) {
+                long spins = 0;
+                while (__hip_atomic_load(&cnt[g], __ATOMIC_RELAXED,
+                                         __HIP_MEMORY_SCOPE_AGENT) < want) {
+                    __builtin_amdgcn_s_sleep(8);
+                    if (++spins > (1L << 24)) break;
+                }
+            }
+            __syncthreads();
+            const __hip_bfloat16* asrc =
+                pub_base + (long)((u + 1) & 1) * ring_stride;
+            f32x4_t acc2[JT][MT];
+#pragma unroll
+            for (int j = 0; j < JT; ++j)
+#pragma unroll
+                for (int m = 0; m < MT; ++m) acc2[j][m] = f32x4_t{0.f};
+            const int arow = lane & 15;
+            const int koff = 8 * (lane >> 4);
+            // ALL carry-GEMM operands load through hand-counted asm
+            // batches (A = sc1 ring reads, B = plain W^T reads): a
+            // compiler-tracked load in this loop would over-wait the
+            // whole in-flight batch. Two-chunk software pipeline with
+            // COMPILE-TIME buffer selection (a runtime-indexed register
+            // array would spill to scratch — the pointer-array lesson).
+            bf16x8_t a0[KCH][MT], a1[KCH][MT];
+            bf16x8_t b0[KCH][JT], b1[KCH][JT];
+            auto issue = [&](int c, bf16x8_t (&aa)[KCH][MT],
+                             bf16x8_t (&bb)[KCH][JT]) {
+#pragma unroll
+                for (int k = 0; k < KCH; ++k) {
+#pragma unroll
+                    for (int m = 0; m < MT; ++m) {
+                        const int row = 16 * (wave + NW * m) + arow;
+                        aa[k][m] = load16_sc1(
+                            asrc + (long)row * 3 * Hp +
+                            32 * (c * KCH + k) + koff);
+                    }
+#pragma unroll
+                    for (int j = 0; j < JT; ++j)
+                        bb[k][j] = load16_asm(
+                            (const __bf16*)wt_dir +
+                            (long)(ct * CS + j * 16 + (lane & 15)) * 3 * Hp +
+                            32 * (c * KCH + k) + koff);
+                }
+            };
+            auto mul = [&](int c, bf16x8_t (&aa)[KCH][MT],
+                           bf16x8_t (&bb)[KCH][JT]) {
+                (void)c;
+#pragma unroll
+                for (int k = 0; k < KCH; ++k)
+#pragma unroll
+                    for (int j = 0; j < JT; ++j)
+#pragma unroll
+                        for (int m = 0; m < MT; ++m)
+                            acc2[j][m] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    aa[k][m], bb[k][j], acc2[j][m], 0, 0, 0);
+            };
+            constexpr int NCH = KK2 / KCH;     // 12 chunks, even
+            issue(0, a0, b0);
+            for (int c = 0; c < NCH; c += 2) {
+                if (c + 1 < NCH) {
+                    issue(c + 1, a1, b1);
+                    asm volatile("s_waitcnt vmcnt(%0)"
+                                 ::"i"(KCH * (MT + JT)) : "memory");
+                } else {
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                }
+                mul(c, a0, b0);
+                if (c + 2 < NCH) {
+                    issue(c + 2, a0, b0);
+                    asm volatile("s_waitcnt vmcnt(%0)"
+                                 ::"i"(KCH * (MT + JT)) : "memory");
+                } else {
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                }
+                if (c + 1 < NCH) mul(c + 1, a1, b1);
+            }
+#pragma unroll
+            for (int j = 0; j < JT; ++j)
+#pragma unroll
+                for (int m = 0; m < MT; ++m)
+#pragma unroll
+                    for (int e = 0; e < 4; ++e)
+                        dhcar[j][m][e] += acc2[j][m][e];
+        }
+
+        // ---- recompute GEMM: gh = out(u-1) @ W_A^T ----
+        const bool haveh = (u > 0);
+        const __hip_bfloat16* hsrc =
+            out_b + (long)(rev ? (Tseq - u) : (u - 1)) * n_dir * Hp;
+        f32x4_t acc[NCT][MT];
+#pragma unroll
+        for (int i = 0; i < NCT; ++i)
+#pragma unroll
+            for (int m = 0; m < MT; ++m) acc[i][m] = f32x4_t{0.f};
+        if (haveh) {
+            const int arow = lane & 15;
+            const int koff = 8 * (lane >> 4);
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                const int kbase = 32 * kk + koff;
+                bf16x8_t a[MT];
+#pragma unroll
+                for (int m = 0; m < MT; ++m) {
+                    const int row = 16 * (wave + NW * m) + arow;
+                    a[m] = *(const bf16x8_t*)(hsrc + (long)row * out_row +
+                                              kbase);
+                }
+#pragma unroll
+                for (int i = 0; i < NCT; ++i) {
+                    const bf16x8_t b = *(const bf16x8_t*)&(
+                        (const __bf16*)w_s)[(i * 16 + (lane & 15)) * WPITCH +
+                                            kbase];
+#pragma unroll
+                    for (int m = 0; m < MT; ++m)
+                        acc[i][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[m], b, acc[i][m], 0, 0, 0);
+                }
+            }
+        }
+
+        // ---- fused gate grads for the block's columns ----
+        const __hip_bfloat16* do_t = dout_b + (long)tt * n_dir * Hp + ct * CS;
+        float keep_dhn[JT][MT][4];
+#pragma unroll
+        for (int j = 0; j < JT; ++j) {
+            const int jj = j * 16 + (lane & 15);
+#pragma unroll
+            for (int m = 0; m < MT; ++m) {
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int b = 16 * (wave + NW * m) + 4 * (lane >> 4) + e;
+                    const bool live = (b < rows_valid);
+                    const float gr = acc[j][m][e] + bhh_s[jj];
+                    const float gz = acc[JT + j][m][e] + bhh_s[CS + jj];
+                    const float hn =
+                        acc[2 * JT + j][m][e] + bhh_s[2 * CS + jj];
+                    const float ir = (float)((const __bf16*)gi_s)[b * GP + jj];
+                    const float iz =
+                        (float)((const __bf16*)gi_s)[b * GP + CS + jj];
+                    const float in_ =
+                        (float)((const __bf16*)gi_s)[b * GP + 2 * CS + jj];
+                    const float r = sigmoidf(ir + gr);
+                    const float z = sigmoidf(iz + gz);
+                    const float n = fast_tanh(in_ + r * hn);
+                    const float hprev =
+                        (live && haveh)
+                            ? (float)hsrc[(long)b * out_row + ct * CS + jj]
+                            : 0.0f;
+                    const float dht =
+                        dhcar[j][m][e] +
+                        (live ? (float)do_t[(long)b * out_row + jj] : 0.0f);
+                    const float dz_pre =
+                        live ? dht * (hprev - n) * z * (1.0f - z) : 0.0f;
+                    const float dn_pre =
+                        live ? dht * (1.0f - z) * (1.0f - n * n) : 0.0f;
+                    const float dr_pre =
+                        live ? dn_pre * hn * r * (1.0f - r) : 0.0f;
+                    const float dhn = live ? dn_pre * r : 0.0f;
+                    // stage dGi (dr, dz, dn) into the gi buffer in place
+                    ((__bf16*)gi_s)[b * GP + jj] =
+                        (__bf16)__float2bfloat16(dr_pre);
+                    ((__bf16*)gi_s)[b * GP + CS + jj] =
+                        (__bf16)__float2bfloat16(dz_pre);
+                    ((__bf16*)gi_s)[b * GP + 2 * CS + jj] =
+                        (__bf16)__float2bfloat16(dn_pre);
+                    keep_dhn[j][m][e] = dhn;
+                    dhcar[j][m][e] = live ? dht * z : 0.0f;
+                    dbacc[j][0] += dr_pre;
+                    dbacc[j][1] += dz_pre;
+                    dbacc[j][2] += dhn;
+                    dbacc[j][3] += dn_pre;
+                }
+            }
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __syncthreads();
+
+        // ---- store dGi strips; swap n-slot to dhn; store + publish dGh --
+        {
+            __hip_bfloat16* dst = dgi_b + (long)tt * n_dir * 3 * Hp;
+            for (int g3 = 0; g3 < 3; ++g3)
+                store_tile<__hip_bfloat16, BR, NT>(
+                    gi_s + g3 * CS, dst + g3 * Hp + ct * CS, CS, GP, gi_row,
+                    rows_valid, tid);
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __syncthreads();
+#pragma unroll
+        for (int j = 0; j < JT; ++j) {
+            const int jj = j * 16 + (lane & 15);
+#pragma unroll
+            for (int m = 0; m < MT; ++m)
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int b = 16 * (wave + NW * m) + 4 * (lane >> 4) + e;
+                    ((__bf16*)gi_s)[b * GP + 2 * CS + jj] =
+                        (__bf16)__float2bfloat16(keep_dhn[j][m][e]);
+                }
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __syncthreads();
+        {
+            // time-shifted global dGh (slot pairs with out[t]; v2/v3
+            // semantics) + publication for the carry GEMM of step u-1
+            const int sh = rev ? (tt + 1) : (tt - 1);
+            if (sh >= 0 && sh < Tseq) {
+                __hip_bfloat16* dst = dgh_b + (long)sh * n_dir * 3 * Hp;
+                for (int g3 = 0; g3 < 3; ++g3)
+                    store_tile<__hip_bfloat16, BR, NT>(
+                        gi_s + g3 * CS, dst + g3 * Hp + ct * CS, CS, GP,
+                        gi_row, rows_valid, tid);
+            } else {
+                __hip_bfloat16* dst =
+                    dgh_b + (long)(rev ? 0 : (Tseq - 1)) * n_dir * 3 * Hp;
+                for (int g3 = 0; g3 < 3; ++g3)
+                    store_zero_tile<__hip_bfloat16, BR, NT>(
+                        dst + g3 * Hp + ct * CS, CS, gi_row, rows_valid, tid);
+            }
+            if (u > 0) {
+                __hip_bfloat16* dst =
+                    pub_base + (long)(u & 1) * ring_stride;
+                constexpr int CPR = CS * 2 / 16;
+                for (int g3 = 0; g3 < 3; ++g3)
+                    for (int c = tid; c < BR * CPR; c += NT) {
+                        const int r = c / CPR;
+                        const int jc = c % CPR;
+                        u32x4_t v = *(const u32x4_t*)((const char*)(gi_s +
+                                        g3 * CS) + (long)r * GP * 2 + jc * 16);
+                        store16_sc1((char*)(dst + (long)r * 3 * Hp +
+                                            g3 * Hp + ct * CS) + jc * 16, v);
+                    }
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                __syncthreads();
+                if (tid == 0)
+                    __hip_atomic_fetch_add(&cnt[g], 1u, __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_AGENT);
+            }
+        }
+        if (u > 0)
+            stage_gi(rev ? (Tseq - u) : (u - 1));
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __syncthreads();
+    }
+
+    {   // epilogue: dh0 + packed bias sums (layout (n_dir, 4Hp))
+        float* d0 = dh0 + ((long)dir * B + b0) * Hp + ct * CS;
+#pragma unroll
+        for (int j = 0; j < JT; ++j) {
+            const int jj = j * 16 + (lane & 15);
+#pragma unroll
+            for (int m = 0; m < MT; ++m)
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int b = 16 * (wave + NW * m) + 4 * (lane >> 4) + e;
+                    if (b < rows_valid) d0[(long)b * Hp + jj] = dhcar[j][m][e];
+                }
+        }
+        // packed bias sums: per j-tile column, reduce the row-quarter
+        // lanes and emit one atomic per (slot, column); dbhh layout is
+        // (n_dir, 4Hp) = [dr | dz | dhn | dn] (db_hh / db_ih assembled
+        // host-side exactly as for the Hp=128 kernels).
+#pragma unroll
+        for (int j = 0; j < JT; ++j)
+#pragma unroll
+            for (int q4 = 0; q4 < 4; ++q4) {
+                float v = dbacc[j][q4];
+                v += __shfl_xor(v, 16);
+                v += __shfl_xor(v, 32);
+                if ((lane >> 4) == 0)
+                    atomicAdd(&dbhh[(long)dir * 4 * Hp + q4 * Hp + ct * CS +
+                                    j * 16 + (lane & 15)], v);
+            }
+    }
 }
 
 // ===========================================================================
@@ -2156,6 +2577,30 @@ extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
     }
 #undef F
     return hipGetLastError() == hipSuccess ? 0 : -4;
+}
+
+extern "C" int fmda_gru_bwd_cs_launch(const void* gi, const void* w,
+                                      const void* wt, const float* bhh,
+                                      const void* out, const void* dout,
+                                      const float* dhT, void* dgi, void* dgh,
+                                      float* dh0, float* dbhh, void* gpub,
+                                      unsigned int* cnt, int B, int Tseq,
+                                      int n_dir, hipStream_t stream) {
+    constexpr int BR = 256, Hp = 512, CS = 32, NT = 512;
+    const int GB = (B + BR - 1) / BR;
+    const size_t lds = 2 * 3 * CS * (Hp + 8) + 2 * (size_t)BR * 3 * CS +
+                       4 * 3 * CS;
+    auto k = gru_bwd_cs_kernel<BR, Hp, CS, NT>;
+    (void)hipFuncSetAttribute((const void*)k,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    const dim3 grid(GB * (Hp / CS) * n_dir);
+    k<<<grid, NT, lds, stream>>>(
+        (const __hip_bfloat16*)gi, (const __hip_bfloat16*)w,
+        (const __hip_bfloat16*)wt, bhh, (const __hip_bfloat16*)out,
+        (const __hip_bfloat16*)dout, dhT, (__hip_bfloat16*)dgi,
+        (__hip_bfloat16*)dgh, dh0, dbhh, (__hip_bfloat16*)gpub, cnt, B,
+        Tseq, n_dir, GB);
+    return 0;
 }
 
 extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,

@@ -234,14 +234,13 @@ def test_fused_head_loss_matches_eager(dtype):
     assert rel(g1[2], bparm.grad) < (1e-4 if dtype == torch.float32 else 5e-2)
 
 
-@pytest.mark.parametrize("B", [64, 33])
-def test_bigru_layer_bf16_all_grads_vs_fp32_reference(B):
+@pytest.mark.parametrize("B,H", [(64, 128), (33, 128), (48, 512), (300, 512)])
+def test_bigru_layer_bf16_all_grads_vs_fp32_reference(B, H):
     """Full single-layer node (pack + projection + recurrence) in bf16 vs
     an fp32 autograd reference: checks out/h_last AND every parameter
     gradient (w_ih, w_hh, b_ih, b_hh, both directions) — the bias grads
     come from in-kernel register accumulators and are otherwise untested."""
     from fmda_amd.ops.interface import _BiGRULayer
-    H = 128
     T, F = 11, 96
     torch.manual_seed(5)
     params = []
