@@ -2015,7 +2015,11 @@ void gru_bwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
                 bf16x8_t a[MT];
 #pragma unroll
                 for (int m = 0; m < MT; ++m) {
-                    const int row = 16 * (wave + NW * m) + arow;
+                    // clamp: `out` is a real (B, ...) tensor, not the
+                    // always-BR-row publication ring — dead rows must not
+                    // read past the batch end (garbage rows are fine).
+                    int row = 16 * (wave + NW * m) + arow;
+                    if (row >= rows_valid) row = 0;
                     a[m] = *(const bf16x8_t*)(hsrc + (long)row * out_row +
                                               kbase);
                 }
