@@ -129,8 +129,14 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
     TORCH_CHECK(out.is_contiguous() && dout.is_contiguous() &&
                 dhT.is_contiguous(), "fmda gru_bwd: tensors must be contiguous");
     TORCH_CHECK(dhT.scalar_type() == torch::kFloat32, "dhT must be fp32");
-    auto dgi = torch::empty_like(gi);
-    auto dgh = torch::empty_like(gi);
+    torch::Tensor dgi, dgh;
+    if (getenv("FMDA_ZERO_GRADS")) {   // debug: expose unwritten positions
+        dgi = torch::zeros_like(gi);
+        dgh = torch::zeros_like(gi);
+    } else {
+        dgi = torch::empty_like(gi);
+        dgh = torch::empty_like(gi);
+    }
     auto dh0 = torch::empty({n_dir, B, Hp},
                             gi.options().dtype(torch::kFloat32));
     // packed bias-grad sums: [dr, dz, dhn | dn] per direction; db_hh and

@@ -1477,18 +1477,24 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 // ===========================================================================
 
 FMDA_DEV bf16x8_t load16_sc1(const void* p) {
+    // The "memory" clobber is load-bearing: without it the compiler may
+    // schedule ITS OWN loads inside a hand-counted asm batch, and the
+    // s_waitcnt vmcnt(N) constants then count the wrong instructions
+    // (consumed-before-loaded garbage fed the MFMAs).
     bf16x8_t v;
     asm volatile("global_load_dwordx4 %0, %1, off sc0 sc1"
-                 : "=v"(v) : "v"(p));
+                 : "=v"(v) : "v"(p) : "memory");
     return v;
 }
 typedef unsigned int u32x4_t __attribute__((ext_vector_type(4)));
 
 FMDA_DEV bf16x8_t load16_asm(const void* p) {
     // plain 16-B load via asm: invisible to the compiler's waitcnt
-    // bookkeeping, so it can share a hand-counted batch with the sc1 loads
+    // bookkeeping, so it can share a hand-counted batch with the sc1
+    // loads. "memory" clobber: see load16_sc1.
     bf16x8_t v;
-    asm volatile("global_load_dwordx4 %0, %1, off" : "=v"(v) : "v"(p));
+    asm volatile("global_load_dwordx4 %0, %1, off" : "=v"(v) : "v"(p)
+                 : "memory");
     return v;
 }
 
@@ -1635,21 +1641,20 @@ void gru_fwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
         {
             const int arow = lane & 15;
             const int koff = 8 * (lane >> 4);
-            // issue ALL h fragment loads, then one drain: a wait per
-            // k-slice would expose L2 latency 16x per step
-            bf16x8_t a[KK][MT];
-#pragma unroll
-            for (int kk = 0; kk < KK; ++kk)
-#pragma unroll
-                for (int m = 0; m < MT; ++m) {
-                    const int row = 16 * (wave + NW * m) + arow;
-                    a[kk][m] = load16_sc1(hsrc + (long)row * Hp + 32 * kk +
-                                          koff);
-                }
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            // h fragments via NONTEMPORAL loads: `nt` bypasses the L1
+            // exactly like sc1 (ring slots repeat every 2 steps, so an
+            // L1 hit would be stale) while staying compiler-tracked —
+            // precise waitcnt scheduling for free.
 #pragma unroll
             for (int kk = 0; kk < KK; ++kk) {
                 const int kbase = 32 * kk + koff;
+                bf16x8_t a[MT];
+#pragma unroll
+                for (int m = 0; m < MT; ++m) {
+                    const int row = 16 * (wave + NW * m) + arow;
+                    a[m] = __builtin_nontemporal_load(
+                        (const bf16x8_t*)(hsrc + (long)row * Hp + kbase));
+                }
 #pragma unroll
                 for (int i = 0; i < NCT; ++i) {
                     const bf16x8_t b = *(const bf16x8_t*)&(
@@ -1658,7 +1663,7 @@ void gru_fwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
 #pragma unroll
                     for (int m = 0; m < MT; ++m)
                         acc[i][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            a[kk][m], b, acc[i][m], 0, 0, 0);
+                            a[m], b, acc[i][m], 0, 0, 0);
                 }
             }
         }
@@ -1814,7 +1819,6 @@ void gru_bwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
     constexpr int JT = CS / 16;          // 2 h-column tiles
     constexpr int KK = Hp / 32;          // 16 (recompute GEMM K)
     constexpr int KK2 = 3 * Hp / 32;     // 48 (carry GEMM K)
-    constexpr int KCH = 2;               // carry-GEMM k-slices per chunk
     constexpr int WPITCH = Hp + 8;
     constexpr int GP = 3 * CS;           // LDS gi/stage pitch
 
@@ -1928,65 +1932,32 @@ void gru_bwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
                 for (int m = 0; m < MT; ++m) acc2[j][m] = f32x4_t{0.f};
             const int arow = lane & 15;
             const int koff = 8 * (lane >> 4);
-            // ALL carry-GEMM operands load through hand-counted asm
-            // batches (A = sc1 ring reads, B = plain W^T reads): a
-            // compiler-tracked load in this loop would over-wait the
-            // whole in-flight batch. Two-chunk software pipeline with
-            // COMPILE-TIME buffer selection (a runtime-indexed register
-            // array would spill to scratch — the pointer-array lesson).
-            bf16x8_t a0[KCH][MT], a1[KCH][MT];
-            bf16x8_t b0[KCH][JT], b1[KCH][JT];
-            auto issue = [&](int c, bf16x8_t (&aa)[KCH][MT],
-                             bf16x8_t (&bb)[KCH][JT]) {
+            // Ring reads via NONTEMPORAL loads (L1 bypass, compiler-
+            // tracked — see the forward kernel note); W^T fragments are
+            // plain L2 loads. The compiler schedules and waits both.
+            // Partial unroll: a full KK2=48 unroll batches enough live
+            // loads to spill.
+#pragma unroll 4
+            for (int kk = 0; kk < KK2; ++kk) {
+                const int kbase = 32 * kk + koff;
+                bf16x8_t a[MT];
 #pragma unroll
-                for (int k = 0; k < KCH; ++k) {
-#pragma unroll
-                    for (int m = 0; m < MT; ++m) {
-                        const int row = 16 * (wave + NW * m) + arow;
-                        aa[k][m] = load16_sc1(
-                            asrc + (long)row * 3 * Hp +
-                            32 * (c * KCH + k) + koff);
-                    }
-#pragma unroll
-                    for (int j = 0; j < JT; ++j)
-                        bb[k][j] = load16_asm(
-                            (const __bf16*)wt_dir +
-                            (long)(ct * CS + j * 16 + (lane & 15)) * 3 * Hp +
-                            32 * (c * KCH + k) + koff);
+                for (int m = 0; m < MT; ++m) {
+                    const int row = 16 * (wave + NW * m) + arow;
+                    a[m] = __builtin_nontemporal_load(
+                        (const bf16x8_t*)(asrc + (long)row * 3 * Hp + kbase));
                 }
-            };
-            auto mul = [&](int c, bf16x8_t (&aa)[KCH][MT],
-                           bf16x8_t (&bb)[KCH][JT]) {
-                (void)c;
 #pragma unroll
-                for (int k = 0; k < KCH; ++k)
+                for (int j = 0; j < JT; ++j) {
+                    const bf16x8_t bfr = *(const bf16x8_t*)&(
+                        (const __bf16*)wt_dir)[
+                        (long)(ct * CS + j * 16 + (lane & 15)) * 3 * Hp +
+                        kbase];
 #pragma unroll
-                    for (int j = 0; j < JT; ++j)
-#pragma unroll
-                        for (int m = 0; m < MT; ++m)
-                            acc2[j][m] =
-                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                                    aa[k][m], bb[k][j], acc2[j][m], 0, 0, 0);
-            };
-            constexpr int NCH = KK2 / KCH;     // 12 chunks, even
-            issue(0, a0, b0);
-            for (int c = 0; c < NCH; c += 2) {
-                if (c + 1 < NCH) {
-                    issue(c + 1, a1, b1);
-                    asm volatile("s_waitcnt vmcnt(%0)"
-                                 ::"i"(KCH * (MT + JT)) : "memory");
-                } else {
-                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                    for (int m = 0; m < MT; ++m)
+                        acc2[j][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[m], bfr, acc2[j][m], 0, 0, 0);
                 }
-                mul(c, a0, b0);
-                if (c + 2 < NCH) {
-                    issue(c + 2, a0, b0);
-                    asm volatile("s_waitcnt vmcnt(%0)"
-                                 ::"i"(KCH * (MT + JT)) : "memory");
-                } else {
-                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-                }
-                if (c + 1 < NCH) mul(c + 1, a1, b1);
             }
 #pragma unroll
             for (int j = 0; j < JT; ++j)
