@@ -213,7 +213,7 @@ def main():
     acc = float(acc_sum.item()) / max(n_acc, 1)
 
     if rank == 0:
-        n_gpus = world if use_cuda else 1
+        n_gpus = world if (use_cuda or world > 1) else 1
         seq_per_sec = n_gpus * batch * args.steps / dt_max
         result = {
             "metric": "train seq/sec, biGRU on synthetic order-book",
