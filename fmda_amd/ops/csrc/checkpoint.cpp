@@ -19,6 +19,7 @@
 #include <cstdint>
 #include <cstring>
 #include <fstream>
+#include <functional>
 #include <map>
 #include <sstream>
 #include <string>
@@ -285,18 +286,21 @@ struct PVal {  // tagged value for the mini VM
     torch::Tensor ten;
 };
 
-std::vector<std::pair<std::string, torch::Tensor>> load_state_dict(
-        const std::string& path) {
-    ZipReader z(path);
-    auto [pkl, pkl_n] = z.get("data.pkl");
-
+// Mini pickle VM. `data_of` resolves a storage key to raw bytes; on
+// return, `end_pos` (when non-null) receives the position just past the
+// STOP opcode so several back-to-back pickles can be parsed (legacy
+// container). `keys_out` collects a top-level list of strings if the
+// pickle's result is a list (the legacy storage-key list).
+static std::vector<std::pair<std::string, torch::Tensor>> run_pickle(
+        const uint8_t* pkl, size_t pkl_n,
+        const std::function<std::pair<const uint8_t*, size_t>(
+            const std::string&)>& data_of,
+        size_t* end_pos = nullptr,
+        std::vector<std::string>* keys_out = nullptr) {
     std::vector<PVal> stack;
     std::vector<size_t> marks;
     std::vector<PVal> memo;
     std::vector<std::pair<std::string, torch::Tensor>> result;
-    auto data_of = [&](const std::string& key) {
-        return z.get("data/" + key);
-    };
 
     size_t i = 0;
     auto need = [&](size_t k) { TORCH_CHECK(i + k <= pkl_n, "pickle truncated"); };
@@ -326,10 +330,27 @@ std::vector<std::pair<std::string, torch::Tensor>> load_state_dict(
                         stack.push_back(v); break; }
             case 0x8a: {                                   // LONG1
                 uint8_t n = rd8(); int64_t v = 0;
-                for (int k = 0; k < n; ++k) v |= (int64_t)rd8() << (8 * k);
-                if (n && (pkl[i - 1] & 0x80)) v -= (int64_t)1 << (8 * n);
+                for (int k = 0; k < n; ++k) {                  // >8 bytes:
+                    uint8_t byt = rd8();                        // keep low 64
+                    if (k < 8) v |= (int64_t)byt << (8 * k);
+                }
+                if (n && n <= 8 && (pkl[i - 1] & 0x80))
+                    v -= (int64_t)1 << (8 * (int)n);
                 PVal pv; pv.kind = PVal::INT; pv.i = v; stack.push_back(pv); break;
             }
+            case 'N': { PVal v; stack.push_back(v); break; }   // NONE
+            case 'a': {                                        // APPEND
+                PVal v = stack.back(); stack.pop_back();
+                stack.back().tup.push_back(v); break;
+            }
+            case 'e': {                                        // APPENDS
+                size_t m = marks.back(); marks.pop_back();
+                std::vector<PVal> items(stack.begin() + m + 1, stack.end());
+                stack.resize(m);
+                for (auto& it : items) stack.back().tup.push_back(it);
+                break;
+            }
+            case 0x88 + 0x100: break;  // unreachable
             case 'X': { uint32_t n = rd32v(); need(n);
                         PVal v; v.kind = PVal::STR;
                         v.s.assign((const char*)pkl + i, n); i += n;
@@ -390,10 +411,14 @@ std::vector<std::pair<std::string, torch::Tensor>> load_state_dict(
                     std::vector<int64_t> sizes, strides;
                     for (const auto& v : a[2].tup) sizes.push_back(v.i);
                     for (const auto& v : a[3].tup) strides.push_back(v.i);
-                    torch::Tensor storage = torch::empty(
-                        {(int64_t)(nbytes / dt.esize)},
-                        torch::TensorOptions().dtype(dt.st));
-                    memcpy(storage.data_ptr(), ptr, nbytes);
+                    // nbytes == 0: structural pass over a legacy container
+                    // (layout not known yet) — build a zero storage of the
+                    // persistent id's element count instead.
+                    const int64_t sn =
+                        nbytes ? (int64_t)(nbytes / dt.esize) : st.i;
+                    torch::Tensor storage = torch::zeros(
+                        {sn}, torch::TensorOptions().dtype(dt.st));
+                    if (nbytes) memcpy(storage.data_ptr(), ptr, nbytes);
                     PVal out; out.kind = PVal::TENSOR;
                     out.ten = storage.as_strided(sizes, strides, offset)
                                   .contiguous();
@@ -431,8 +456,14 @@ std::vector<std::pair<std::string, torch::Tensor>> load_state_dict(
                 break;
             }
             case '.':
+                if (end_pos) *end_pos = i;
+                if (keys_out && !stack.empty())
+                    for (const auto& it : stack.back().tup)
+                        keys_out->push_back(it.s);
                 return result;
             case '2': break;                               // DUP (unused)
+            case 'G': { need(8); i += 8;                   // BINFLOAT
+                        PVal v; v.kind = PVal::INT; stack.push_back(v); break; }
             case 0x95: { need(8); i += 8; break; }         // FRAME
             default:
                 {
@@ -442,6 +473,86 @@ std::vector<std::pair<std::string, torch::Tensor>> load_state_dict(
                 }
         }
     }
+}
+
+std::vector<std::pair<std::string, torch::Tensor>> load_state_dict(
+        const std::string& path) {
+    std::ifstream f(path, std::ios::binary | std::ios::ate);
+    TORCH_CHECK(f.good(), "cannot open ", path);
+    size_t n = (size_t)f.tellg();
+    std::vector<uint8_t> buf(n);
+    f.seekg(0);
+    f.read((char*)buf.data(), n);
+
+    if (n >= 4 && buf[0] == 'P' && buf[1] == 'K') {
+        ZipReader z(path);
+        auto [pkl, pkl_n] = z.get("data.pkl");
+        return run_pickle(pkl, pkl_n, [&](const std::string& key) {
+            return z.get("data/" + key);
+        });
+    }
+
+    // Legacy (pre-zip) torch container, as shipped in the reference
+    // model_params.pt: magic pickle, protocol pickle, sys_info pickle,
+    // the object pickle, the storage-key list pickle, then per key an
+    // int64 numel followed by the raw storage bytes.
+    TORCH_CHECK(n >= 2 && buf[0] == 0x80, "not a torch checkpoint: ", path);
+    auto none_data = [](const std::string&)
+        -> std::pair<const uint8_t*, size_t> {
+        TORCH_CHECK(false, "storage requested during structural pass");
+    };
+    size_t pos = 0, next = 0;
+    run_pickle(buf.data() + pos, n - pos, none_data, &next);  // magic
+    pos += next;
+    run_pickle(buf.data() + pos, n - pos, none_data, &next);  // protocol
+    pos += next;
+    run_pickle(buf.data() + pos, n - pos, none_data, &next);  // sys_info
+    pos += next;
+    const size_t obj_pos = pos;
+    {   // structural pass over the object pickle just to find its end;
+        // storages resolve to empty (the VM builds zero tensors then)
+        auto zero_data = [&](const std::string&)
+            -> std::pair<const uint8_t*, size_t> {
+            return {nullptr, 0};
+        };
+        run_pickle(buf.data() + pos, n - pos, zero_data, &next);
+        pos += next;
+    }
+    std::vector<std::string> keys;
+    run_pickle(buf.data() + pos, n - pos, none_data, &next, &keys);
+    pos += next;
+
+    // Data section: per key (in key-list order) an int64 element count,
+    // then numel * esize raw bytes. The element size is uniform per file
+    // in practice; find the E in {4, 8, 2, 1} whose walk lands exactly on
+    // EOF.
+    const size_t data_pos = pos;
+    std::map<std::string, std::pair<size_t, int64_t>> slots;  // pos, numel
+    size_t E_found = 0;
+    for (size_t E : {4, 8, 2, 1}) {
+        size_t p2 = data_pos;
+        std::map<std::string, std::pair<size_t, int64_t>> trial;
+        bool ok = true;
+        for (const auto& k : keys) {
+            if (p2 + 8 > n) { ok = false; break; }
+            int64_t numel;
+            memcpy(&numel, buf.data() + p2, 8);
+            p2 += 8;
+            if (numel < 0 || p2 + (size_t)numel * E > n) { ok = false; break; }
+            trial[k] = {p2, numel};
+            p2 += (size_t)numel * E;
+        }
+        if (ok && p2 == n) { slots = trial; E_found = E; break; }
+    }
+    TORCH_CHECK(E_found, "legacy storage layout not understood");
+    auto data_of2 = [&](const std::string& key)
+        -> std::pair<const uint8_t*, size_t> {
+        auto it = slots.find(key);
+        TORCH_CHECK(it != slots.end(), "unknown storage key ", key);
+        return {buf.data() + it->second.first,
+                (size_t)it->second.second * E_found};
+    };
+    return run_pickle(buf.data() + obj_pos, n - obj_pos, data_of2);
 }
 
 }  // namespace fmda_ckpt
