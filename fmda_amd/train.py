@@ -53,7 +53,8 @@ def chunk_batches(market, chunk_set, dcfg, batch_size):
 
 def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
           tcfg: TrainConfig = None, checkpoint_path: str = "model_params.pt",
-          norm_params_path: str = "norm_params", log=print):
+          norm_params_path: str = "norm_params", log=print,
+          resume: str = None):
     mcfg = mcfg or ModelConfig()
     dcfg = dcfg or DataConfig(n_features=mcfg.n_features)
     tcfg = tcfg or TrainConfig()
@@ -72,6 +73,14 @@ def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
                                            pos_weight=pos_weight.to(device)))
     model.add_optimizer(torch.optim.Adam(model.parameters(), lr=tcfg.lr))
     model.add_device(device)
+    if resume:
+        # mid-training resume (reference only ever saves, predict.py:104
+        # only loads; this is the SURVEY.md section 5 native extension)
+        state = torch.load(resume, map_location=device, weights_only=True)
+        if "model" in state:
+            model.load_state_dict(state["model"])
+        else:
+            model.load_state_dict(state)
 
     history = []
     best_val_acc = -1.0
@@ -105,13 +114,16 @@ def main():
     ap.add_argument("--batch", type=int, default=2)
     ap.add_argument("--device", default="cpu")
     ap.add_argument("--checkpoint", default="model_params.pt")
+    ap.add_argument("--resume", default=None,
+                    help="checkpoint to resume from (model_params.pt format)")
     args = ap.parse_args()
     mcfg = ModelConfig(hidden_size=args.hidden, n_layers=args.layers,
                        spatial_dropout=False, dropout=0.5)
     dcfg = DataConfig(n_rows=args.rows, window=args.window)
     tcfg = TrainConfig(batch_size=args.batch, epochs=args.epochs,
                        device=args.device)
-    train(mcfg, dcfg, tcfg, checkpoint_path=args.checkpoint)
+    train(mcfg, dcfg, tcfg, checkpoint_path=args.checkpoint,
+          resume=args.resume)
 
 
 if __name__ == "__main__":
