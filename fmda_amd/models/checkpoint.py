@@ -19,7 +19,11 @@ def save_checkpoint(model: BiGRU, path: str) -> None:
     Uses the native C++ zip/pickle writer (ops/csrc/checkpoint.cpp) when
     the extension is importable; the output is torch.load-compatible
     either way."""
-    sd = model.state_dict()
+    sd = {k: v for k, v in model.state_dict().items()
+          if not k.startswith("loss_fn.")}
+    # (add_loss_fn attaches the loss module, whose weight/pos_weight
+    # buffers would otherwise leak into the artifact; the reference
+    # model_params.pt holds only gru.* / linear.* entries.)
     try:
         from ..ops import _fmda_hip
         _fmda_hip.save_state_dict_native(path, list(sd.keys()),
