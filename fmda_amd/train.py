@@ -77,10 +77,13 @@ def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
         # mid-training resume (reference only ever saves, predict.py:104
         # only loads; this is the SURVEY.md section 5 native extension)
         state = torch.load(resume, map_location=device, weights_only=True)
-        if "model" in state:
-            model.load_state_dict(state["model"])
-        else:
-            model.load_state_dict(state)
+        sd = state["model"] if "model" in state else state
+        # strict=False: the live model carries the attached loss module,
+        # whose buffers are deliberately absent from checkpoints
+        missing, unexpected = model.load_state_dict(sd, strict=False)
+        bad = [k for k in list(missing) + list(unexpected)
+               if not k.startswith("loss_fn.")]
+        assert not bad, f"resume key mismatch: {bad}" 
 
     history = []
     best_val_acc = -1.0
