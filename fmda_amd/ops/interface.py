@@ -20,20 +20,6 @@ from .blas import chunked_colsum, chunked_outer, enable_tunableop
 
 _ALLOWED_HP = [16, 32, 64, 128, 256, 512]
 
-_side_stream = None
-
-
-def _side():
-    """Side HIP stream for gradient-reduction GEMMs. The persistent BPTT
-    kernel runs 4 waves/CU (latency-bound), leaving the MFMA pipes mostly
-    idle; the dW reductions overlap into that shadow. Every use is
-    self-contained (fork from and join back to the current stream inside
-    one backward call), so no synchronization leaks to callers."""
-    global _side_stream
-    if _side_stream is None:
-        _side_stream = torch.cuda.Stream()
-    return _side_stream
-
 
 def _pad_h(H: int) -> int:
     for hp in _ALLOWED_HP:
@@ -265,24 +251,14 @@ class _BiGRULayer(torch.autograd.Function):
             gi, w_hh_cat, b_hh_cat, out, d_out, d_hlast)
         M = dgi.shape[0] * dgi.shape[1]
 
-        # dW reductions on the side stream, overlapped with the dx GEMM on
-        # the main stream; joined below before returning, so nothing leaks
-        # to callers (DDP hooks / optimizer read grads on the main stream).
-        main = torch.cuda.current_stream()
-        side = _side()
-        side.wait_stream(main)
-        with torch.cuda.stream(side):
-            # dW_hh via the time-shifted dGh; dW_ih for both directions —
-            # hand-split fat-K reductions (see chunked_outer)
-            cross = chunked_outer(dgh.reshape(M, -1), out.reshape(M, -1))
-            dwih_cat = chunked_outer(dgi.reshape(M, -1), x2d)
-            cross.record_stream(side)
-            dwih_cat.record_stream(side)
+        # dW_hh via the time-shifted dGh and one split-K reduction
+        cross = chunked_outer(dgh.reshape(M, -1), out.reshape(M, -1))
+        # dW_ih for both directions in one split-K reduction
+        dwih_cat = chunked_outer(dgi.reshape(M, -1), x2d)
         dx = None
         if need_dx:
             dx = torch.matmul(dgi.reshape(M, -1), w_ih_cat)
             dx = dx.view(dgi.shape[0], dgi.shape[1], -1)
-        main.wait_stream(side)
 
         grads = []
         for d in range(D):
