@@ -870,13 +870,18 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     char* p = smem;
-    __hip_bfloat16* gi_s[2];
-    gi_s[0] = (__hip_bfloat16*)p; p += 2 * BT * GP;
-    gi_s[1] = (__hip_bfloat16*)p; p += 2 * BT * GP;
-    __hip_bfloat16* hb_s[2];
-    hb_s[0] = (__hip_bfloat16*)p; p += 2 * BT * WP;
-    hb_s[1] = (__hip_bfloat16*)p; p += 2 * BT * WP;
+    // Select HELPERS, not pointer arrays: a runtime-indexed local pointer
+    // array defeats LDS address-space inference, so every gi/h access in
+    // the loop compiled to FLAT ops — which wait on vmcnt and made the
+    // compiler drain the in-flight LDS-DMA with a vmcnt(0) before every
+    // gate group (the 2.4x slowdown this comment is a tombstone for).
+    __hip_bfloat16* gi_s0 = (__hip_bfloat16*)p; p += 2 * BT * GP;
+    __hip_bfloat16* gi_s1 = (__hip_bfloat16*)p; p += 2 * BT * GP;
+    __hip_bfloat16* hb_s0 = (__hip_bfloat16*)p; p += 2 * BT * WP;
+    __hip_bfloat16* hb_s1 = (__hip_bfloat16*)p; p += 2 * BT * WP;
     float* bhh_s = (float*)p;
+    auto gi_buf = [&](int b) { return b ? gi_s1 : gi_s0; };
+    auto hb_buf = [&](int b) { return b ? hb_s1 : hb_s0; };
 
     const long gi_row = (long)Tseq * n_dir * 3 * Hp;
     const long out_row = (long)Tseq * n_dir * Hp;
@@ -885,7 +890,7 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 
     for (int c = tid; c < 3 * Hp; c += NT)
         bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
-    zero_tile<__hip_bfloat16, BT, NT>(hb_s[0], WP, tid);
+    zero_tile<__hip_bfloat16, BT, NT>(hb_s0, WP, tid);
 
     // W_hh B-fragments, register-resident for the whole sequence.
     const __hip_bfloat16* wdir = w + (long)dir * 3 * Hp * Hp;
@@ -921,7 +926,7 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
             const int c = e % GP;
             if (r >= rows_valid) r = 0;
             glds16(src_t + (long)r * gi_row + c,
-                   (char*)gi_s[buf] + (long)(wavu * PW + k) * 1024);
+                   (char*)gi_buf(buf) + (long)(wavu * PW + k) * 1024);
         }
     };
 
@@ -943,7 +948,7 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
         if (u > 0) {
             const int ttp = rev ? (Tseq - u) : (u - 1);
             store_tile<__hip_bfloat16, BT, NT>(
-                hb_s[pb], out_b + (long)ttp * n_dir * Hp, Hp, WP, out_row,
+                hb_buf(pb), out_b + (long)ttp * n_dir * Hp, Hp, WP, out_row,
                 rows_valid, tid);
         }
 
@@ -965,7 +970,8 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 #pragma unroll
                 for (int m = 0; m < MT; ++m)
                     a[m] = *(const bf16x8_t*)&(
-                        (const __bf16*)hb_s[pb])[(16 * m + arow) * WP + kbase];
+                        (const __bf16*)hb_buf(pb))[(16 * m + arow) * WP +
+                                                   kbase];
 #pragma unroll
                 for (int g = 0; g < 3; ++g)
 #pragma unroll
@@ -985,8 +991,8 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
         for (int i = 0; i < CPW; ++i) {
             const int ct = wave + NW * i;
             const int j = ct * 16 + (lane & 15);
-            const __bf16* gprow = (const __bf16*)gi_s[pb];
-            __bf16* hrow = (__bf16*)hb_s[1 - pb];
+            const __bf16* gprow = (const __bf16*)gi_buf(pb);
+            __bf16* hrow = (__bf16*)hb_buf(1 - pb);
 #pragma unroll
             for (int m = 0; m < MT; ++m) {
 #pragma unroll
@@ -1023,7 +1029,7 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
         const int pl = Tseq & 1;
         const int ttl = rev ? 0 : (Tseq - 1);
         store_tile<__hip_bfloat16, BT, NT>(
-            hb_s[pl], out_b + (long)ttl * n_dir * Hp, Hp, WP, out_row,
+            hb_buf(pl), out_b + (long)ttl * n_dir * Hp, Hp, WP, out_row,
             rows_valid, tid);
         float* hl = hlast + ((long)dir * B + b0) * Hp;
 #pragma unroll
