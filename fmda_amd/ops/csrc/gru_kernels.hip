@@ -2502,14 +2502,11 @@ static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
                              const float* dhT, void* dgi, void* dgh,
                              float* dh0, float* dbhh, int B, int Tseq,
                              int n_dir, hipStream_t stream) {
-    constexpr int BT = 32, Hp = 128, NT = 256;
+    constexpr int BT = 32, Hp = 128, NT = 512;
     const size_t lds = 3 * 2 * BT * 3 * Hp + 2 * 2 * BT * Hp +
                        2 * 2 * BT * Hp + 2 * 2 * BT * (3 * Hp + 8) +
                        4 * 3 * Hp;
-    // 4 waves at 1 wave/SIMD: this shape is spill-free (NT=512 at 2
-    // waves/SIMD spills 3 VGPRs, and every spill reload re-triggers the
-    // glds vmcnt(0) drain).
-    auto k = gru_bwd_v3_kernel<BT, Hp, NT, 1>;
+    auto k = gru_bwd_v3_kernel<BT, Hp, NT, 2>;
     (void)hipFuncSetAttribute((const void*)k,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     const dim3 grid((B + BT - 1) / BT, n_dir);
