@@ -290,3 +290,51 @@ def test_bigru_layer_bf16_all_grads_vs_fp32_reference(B, H):
         else:
             rel = (a - ref).norm() / ref.norm().clamp(min=1e-2)
             assert rel < 8e-2, f"{nm}: rel-L2={rel}"
+
+
+def test_training_step_reproducible():
+    """Two identical training steps from identical state produce the same
+    parameters within atomic-accumulation jitter (SURVEY.md section 4 (d);
+    db_hh/db_ih use atomicAdd, so bitwise equality is not expected)."""
+    from fmda_amd.optim import FusedClipAdam
+    results = []
+    for _ in range(2):
+        torch.manual_seed(77)
+        from fmda_amd.models import BiGRU
+        m = BiGRU(128, 96, 4, n_layers=2, spatial_dropout=False,
+                  dropout=0.0).cuda()
+        opt = FusedClipAdam(m.parameters(), lr=1e-3, clip=50.0)
+        g = torch.Generator().manual_seed(5)
+        x = torch.randn(64, 30, 96, generator=g).bfloat16().cuda()
+        y = (torch.rand(64, 4, generator=g) < 0.3).float().cuda()
+        for _ in range(3):
+            opt.zero_grad(set_to_none=False)
+            logits = m(x)
+            loss = nn.functional.binary_cross_entropy_with_logits(
+                logits.float(), y)
+            loss.backward()
+            opt.step()
+        results.append([p.detach().clone() for p in m.parameters()])
+    for a, b in zip(*results):
+        assert torch.allclose(a, b, atol=1e-5, rtol=1e-4), \
+            (a - b).abs().max()
+
+
+def test_reference_train_api_on_gpu():
+    """The reference train_model/evaluate_model epoch loops run on the HIP
+    engine (biGRU_model.py:162-286 semantics on cuda)."""
+    from fmda_amd.models import BiGRU
+    torch.manual_seed(3)
+    m = BiGRU(32, 16, 4, n_layers=1, spatial_dropout=False,
+              dropout=0.2).cuda()
+    m.add_loss_fn(nn.BCEWithLogitsLoss())
+    m.add_optimizer(torch.optim.Adam(m.parameters(), lr=1e-3))
+    m.add_device(torch.device("cuda"))
+    g = torch.Generator().manual_seed(9)
+    batches = [(torch.randn(8, 12, 16, generator=g),
+                (torch.rand(8, 1, 4, generator=g) < 0.3).float())
+               for _ in range(4)]
+    acc, ham, loss, fbeta = m.train_model(iter(batches))
+    assert 0.0 <= acc <= 1.0 and 0.0 <= ham <= 1.0
+    acc2, ham2, fb2, pred_tot, tgt_tot = m.evaluate_model(iter(batches))
+    assert pred_tot.shape == (32, 4) and tgt_tot.shape == (32, 4)
