@@ -2181,23 +2181,35 @@ __global__ void pool_fwd_kernel(const T* __restrict__ out,
                                 float* __restrict__ avgv,
                                 int* __restrict__ amax, int B, int Tseq,
                                 int H, int n_dir) {
+    // one thread per (b, h-pair): 4-byte loads keep the row reads at full
+    // cache-line efficiency (2-byte-per-lane halves it)
     const int idx = blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx >= B * H) return;
-    const int b = idx / H;
-    const int h = idx % H;
+    const int HP2 = H / 2;
+    if (idx >= B * HP2) return;
+    const int b = idx / HP2;
+    const int h2 = (idx % HP2) * 2;
     const int HD = n_dir * H;
-    const T* p = out + (long)b * Tseq * HD + h;
-    float mx = -3.4e38f, s = 0.0f;
-    int im = 0;
+    const T* p = out + (long)b * Tseq * HD + h2;
+    float mx0 = -3.4e38f, mx1 = -3.4e38f, s0 = 0.0f, s1 = 0.0f;
+    int im0 = 0, im1 = 0;
     for (int t = 0; t < Tseq; ++t) {
-        float v = to_f32<T>(p[(long)t * HD]);
-        if (n_dir == 2) v += to_f32<T>(p[(long)t * HD + H]);
-        s += v;
-        if (v > mx) { mx = v; im = t; }
+        const T* pt = p + (long)t * HD;
+        struct alignas(2 * sizeof(T)) Pair { T x, y; };
+        const Pair v2 = *(const Pair*)pt;
+        float v0 = to_f32<T>(v2.x), v1 = to_f32<T>(v2.y);
+        if (n_dir == 2) {
+            const Pair w2 = *(const Pair*)(pt + H);
+            v0 += to_f32<T>(w2.x);
+            v1 += to_f32<T>(w2.y);
+        }
+        s0 += v0; s1 += v1;
+        if (v0 > mx0) { mx0 = v0; im0 = t; }
+        if (v1 > mx1) { mx1 = v1; im1 = t; }
     }
-    maxv[idx] = mx;
-    avgv[idx] = s / (float)Tseq;
-    amax[idx] = im;
+    const int o = b * H + h2;
+    maxv[o] = mx0; maxv[o + 1] = mx1;
+    avgv[o] = s0 / (float)Tseq; avgv[o + 1] = s1 / (float)Tseq;
+    amax[o] = im0; amax[o + 1] = im1;
 }
 
 template <typename T>
@@ -2225,7 +2237,7 @@ __global__ void pool_bwd_kernel(const float* __restrict__ dmax,
 extern "C" int fmda_pool_fwd_launch(int is_bf16, const void* out, float* maxv,
                                     float* avgv, int* amax, int B, int Tseq,
                                     int H, int n_dir, hipStream_t stream) {
-    const int n = B * H;
+    const int n = B * (H / 2);   // thread per h-pair (H is always even)
     const dim3 grid((n + 255) / 256);
     if (is_bf16)
         pool_fwd_kernel<__hip_bfloat16><<<grid, 256, 0, stream>>>(
@@ -2483,6 +2495,21 @@ using bf16_t = __hip_bfloat16;
 static int launch_fwd_v3_128(const void* gi, const void* w, const float* bhh,
                              void* out, float* hlast, int B, int Tseq,
                              int n_dir, hipStream_t stream) {
+    static const bool big = getenv("FMDA_FWD_BT32") != nullptr;
+    if (big) {   // A/B: one 8-wave block per CU instead of two 4-wave
+        constexpr int BT = 32, Hp = 128, NT = 512;
+        const size_t lds = 2 * 2 * BT * 3 * Hp + 2 * 2 * BT * (Hp + 8) +
+                           4 * 3 * Hp;
+        auto k2 = gru_fwd_v3_kernel<BT, Hp, NT, 2>;
+        (void)hipFuncSetAttribute((const void*)k2,
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+        const dim3 grid((B + BT - 1) / BT, n_dir);
+        k2<<<grid, NT, lds, stream>>>((const __hip_bfloat16*)gi,
+                                      (const __hip_bfloat16*)w, bhh,
+                                      (__hip_bfloat16*)out, hlast, B, Tseq,
+                                      n_dir);
+        return 0;
+    }
     constexpr int BT = 16, Hp = 128, NT = 256;
     const size_t lds = 2 * 2 * BT * 3 * Hp + 2 * 2 * BT * (Hp + 8) +
                        4 * 3 * Hp;
