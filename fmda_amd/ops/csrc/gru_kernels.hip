@@ -1482,27 +1482,7 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 // not apply).
 // ===========================================================================
 
-FMDA_DEV bf16x8_t load16_sc1(const void* p) {
-    // The "memory" clobber is load-bearing: without it the compiler may
-    // schedule ITS OWN loads inside a hand-counted asm batch, and the
-    // s_waitcnt vmcnt(N) constants then count the wrong instructions
-    // (consumed-before-loaded garbage fed the MFMAs).
-    bf16x8_t v;
-    asm volatile("global_load_dwordx4 %0, %1, off sc0 sc1"
-                 : "=v"(v) : "v"(p) : "memory");
-    return v;
-}
 typedef unsigned int u32x4_t __attribute__((ext_vector_type(4)));
-
-FMDA_DEV bf16x8_t load16_asm(const void* p) {
-    // plain 16-B load via asm: invisible to the compiler's waitcnt
-    // bookkeeping, so it can share a hand-counted batch with the sc1
-    // loads. "memory" clobber: see load16_sc1.
-    bf16x8_t v;
-    asm volatile("global_load_dwordx4 %0, %1, off" : "=v"(v) : "v"(p)
-                 : "memory");
-    return v;
-}
 
 FMDA_DEV void store16_sc1(void* p, u32x4_t v) {
     asm volatile("global_store_dwordx4 %0, %1, off sc0 sc1\ns_nop 1"

@@ -21,8 +21,7 @@ struct OptChunk {
     int n;
 };
 
-constexpr int OPT_CHUNK = 1 << 12;  // elements per chunk (small => enough blocks for 256 CUs)
-constexpr int OPT_NT = 256;
+constexpr int OPT_NT = 256;  // chunking lives host-side (fmda_amd/optim.py)
 
 __global__ void norm2_kernel(const OptChunk* __restrict__ chunks, int n_chunks,
                              float* __restrict__ out) {
