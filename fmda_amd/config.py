@@ -109,9 +109,11 @@ BENCH_CONFIGS = {
     # BASELINE.json configs[0]: CPU plumbing check
     "cpu": BenchConfig(name="cpu", hidden_size=32, n_layers=1, seq_len=60,
                        n_features=64, batch_per_gpu=8, dtype="fp32"),
-    # configs[1]/[2]: repo config, 1 and 8 GPU
+    # configs[1]/[2]: repo config, 1 and 8 GPU. batch 8192/GPU measured
+    # ~3% faster than 4096 (better amortization of the latency-bound
+    # recurrence steps); both fill the grid.
     "repo": BenchConfig(name="repo", hidden_size=128, n_layers=2, seq_len=120,
-                        n_features=96, batch_per_gpu=4096, dtype="bf16"),
+                        n_features=96, batch_per_gpu=8192, dtype="bf16"),
     # configs[3]: stress. batch 2048 fills the chip: the recurrence
     # kernels parallelize over (batch-tile, direction) blocks, and H=512
     # at batch 256 dispatches only 32 workgroups on 256 CUs.
