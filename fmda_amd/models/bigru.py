@@ -95,6 +95,10 @@ class BiGRU(nn.Module):
             input_seq = input_seq.permute(0, 2, 1)
             input_seq = self.spatial_dropout1d(input_seq)
             input_seq = input_seq.permute(0, 2, 1)
+        elif (self.training and input_seq.is_cuda
+              and input_seq.dtype == torch.bfloat16 and self.dropout_p > 0):
+            from ..ops.interface import fused_dropout
+            input_seq = fused_dropout(input_seq, self.dropout_p)
         else:
             input_seq = self.dropout(input_seq)
 

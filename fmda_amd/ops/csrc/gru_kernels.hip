@@ -2149,6 +2149,65 @@ void gru_bwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
 }
 
 // ===========================================================================
+// Counter-based dropout (the reference's nn.Dropout call
+// sites, biGRU_model.py:50-52,87-94). The mask is a pure function of
+// (seed, element index) via a splitmix64 mix, so backward RECOMPUTES it
+// instead of loading a saved mask — half the memory traffic of the eager
+// pair (no mask tensor exists at all). Keep/scale semantics match
+// torch.nn.functional.dropout (scale 1/(1-p) on kept elements).
+// ===========================================================================
+FMDA_DEV unsigned long long mix64(unsigned long long x) {
+    x += 0x9E3779B97F4A7C15ull;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+    return x ^ (x >> 31);
+}
+
+template <typename T>
+__global__ void dropout_kernel(const T* __restrict__ x, T* __restrict__ y,
+                               long n, float p, float scale,
+                               unsigned long long seed) {
+    // 8 elements per thread; the mask bits come from one mix64 per octet
+    const long o = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (o >= n) return;
+    const unsigned long long r = mix64(seed ^ (unsigned long long)(o >> 3));
+    const unsigned int thr = (unsigned int)(p * 4294967296.0f);
+    if (o + 8 <= n) {
+        T v[8];
+        *(chunk16*)v = *(const chunk16*)(x + o);  // bf16 x8 = 16 B
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+            const unsigned int u =
+                (unsigned int)(mix64(r + k) >> 32);
+            v[k] = (u < thr) ? from_f32<T>(0.0f)
+                             : from_f32<T>(to_f32<T>(v[k]) * scale);
+        }
+        *(chunk16*)(y + o) = *(const chunk16*)v;
+    } else {
+        for (long i = o; i < n; ++i) {
+            const unsigned int u =
+                (unsigned int)(mix64(r + (i - o)) >> 32);
+            y[i] = (u < thr) ? from_f32<T>(0.0f)
+                             : from_f32<T>(to_f32<T>(x[i]) * scale);
+        }
+    }
+}
+
+extern "C" int fmda_dropout_launch(int is_bf16, const void* x, void* y,
+                                   long n, float p, unsigned long long seed,
+                                   hipStream_t stream) {
+    const float scale = 1.0f / (1.0f - p);
+    const long threads = (n + 7) / 8;
+    const dim3 grid((threads + 255) / 256);
+    if (is_bf16)
+        dropout_kernel<__hip_bfloat16><<<grid, 256, 0, stream>>>(
+            (const __hip_bfloat16*)x, (__hip_bfloat16*)y, n, p, scale, seed);
+    else
+        return -2;  // bf16-only: the fp32 paths use torch's dropout
+    return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
+// ===========================================================================
 // Fused direction-sum + temporal max/avg pooling (biGRU_model.py:108-133
 // semantics: gru_out = fwd_dir + bwd_dir; max over T with argmax; sum/T).
 // One thread per (b, h): replaces four eager kernels (direction add, max

@@ -126,6 +126,33 @@ class _IHProjection(torch.autograd.Function):
         return dx, dw, db
 
 
+class _FusedDropout(torch.autograd.Function):
+    """Counter-based dropout: the mask is a pure function of (seed, index),
+    so backward recomputes it instead of materializing a mask tensor
+    (reference nn.Dropout call sites, biGRU_model.py:50-52,87-94)."""
+
+    @staticmethod
+    def forward(ctx, x, p, seed):
+        ext = load_extension()
+        ctx.p = p
+        ctx.seed = seed
+        return ext.dropout_fused(x.contiguous(), p, seed)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_extension()
+        return ext.dropout_fused(dy.contiguous(), ctx.p, ctx.seed), None, None
+
+
+def fused_dropout(x: torch.Tensor, p: float) -> torch.Tensor:
+    """Training-mode dropout on the HIP engine (bf16 CUDA tensors);
+    falls back to torch for other dtypes/devices."""
+    if not (x.is_cuda and x.dtype == torch.bfloat16 and 0.0 < p < 1.0):
+        return torch.nn.functional.dropout(x, p=p, training=True)
+    seed = int(torch.empty((), dtype=torch.int64).random_())
+    return _FusedDropout.apply(x, p, seed)
+
+
 class _DirSumPool(torch.autograd.Function):
     """Fused direction-sum + max/avg temporal pooling on the HIP engine
     (biGRU_model.py:108-133 semantics). Returns (max (B,H), avg (B,H)) in
@@ -320,7 +347,7 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
 
         inp = out
         if training and dropout_p > 0 and layer < n_layers - 1:
-            inp = torch.nn.functional.dropout(inp, p=dropout_p, training=True)
+            inp = fused_dropout(inp, dropout_p)
 
     h_n = torch.cat(h_n_parts, dim=0)  # (L*D, B, H)
     return inp, h_n

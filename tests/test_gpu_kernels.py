@@ -353,3 +353,22 @@ def test_unidirectional_model_on_gpu():
     ref = m(x)
     got = m.cuda()(x.cuda()).cpu()
     assert torch.allclose(got, ref, atol=5e-3), (got - ref).abs().max()
+
+
+def test_fused_dropout_statistics_and_backward():
+    """Counter-based dropout: keep-rate ~ 1-p, kept elements scaled by
+    1/(1-p), and backward applies the IDENTICAL mask (recomputed)."""
+    from fmda_amd.ops.interface import _FusedDropout
+    torch.manual_seed(1)
+    x = torch.ones(4096, 257, device="cuda").bfloat16().requires_grad_(True)
+    p = 0.4
+    y = _FusedDropout.apply(x, p, 12345)
+    kept = (y != 0)
+    rate = kept.float().mean().item()
+    assert abs(rate - (1 - p)) < 0.01, rate
+    scale = y.float()[kept].mean().item()
+    assert abs(scale - 1 / (1 - p)) < 0.02, scale
+    g = torch.ones_like(y)
+    y.backward(g)
+    # backward mask identical to forward mask
+    assert torch.equal((x.grad != 0), kept)
