@@ -2170,15 +2170,17 @@ __global__ void dropout_kernel(const T* __restrict__ x, T* __restrict__ y,
     // 8 elements per thread; the mask bits come from one mix64 per octet
     const long o = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
     if (o >= n) return;
+    // ONE mix64 per octet: its 8 bytes are 8 per-element draws against an
+    // 8-bit threshold (p quantized to 1/256 — immaterial for dropout and
+    // 9x less hashing, which made the kernel compute-bound).
     const unsigned long long r = mix64(seed ^ (unsigned long long)(o >> 3));
-    const unsigned int thr = (unsigned int)(p * 4294967296.0f);
+    const unsigned int thr = (unsigned int)(p * 256.0f);
     if (o + 8 <= n) {
         T v[8];
         *(chunk16*)v = *(const chunk16*)(x + o);  // bf16 x8 = 16 B
 #pragma unroll
         for (int k = 0; k < 8; ++k) {
-            const unsigned int u =
-                (unsigned int)(mix64(r + k) >> 32);
+            const unsigned int u = (unsigned int)(r >> (8 * k)) & 0xFF;
             v[k] = (u < thr) ? from_f32<T>(0.0f)
                              : from_f32<T>(to_f32<T>(v[k]) * scale);
         }
@@ -2186,7 +2188,7 @@ __global__ void dropout_kernel(const T* __restrict__ x, T* __restrict__ y,
     } else {
         for (long i = o; i < n; ++i) {
             const unsigned int u =
-                (unsigned int)(mix64(r + (i - o)) >> 32);
+                (unsigned int)(r >> (8 * (i - o))) & 0xFF;
             y[i] = (u < thr) ? from_f32<T>(0.0f)
                              : from_f32<T>(to_f32<T>(x[i]) * scale);
         }
