@@ -48,3 +48,27 @@ def batch_metrics(target: torch.Tensor, pred: torch.Tensor,
                   beta: float = 0.5) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     return (subset_accuracy(target, pred), hamming(target, pred),
             fbeta_per_class(target, pred, beta))
+
+
+def three_class_accuracy(target: torch.Tensor,
+                         pred: torch.Tensor) -> torch.Tensor:
+    """Derived up/down/stall accuracy over the 4-label head.
+
+    The reference README frames the task as 3-class up/down/stall while the
+    implementation is 4-label multilabel (up1, up2, down1, down2 — reference
+    create_database.py:166-190); "stall" is the implicit all-zeros labeling.
+    This collapses the 4 logits-thresholded labels to {up, down, stall}:
+    any up* set -> up, any down* set -> down, both or neither -> stall
+    (conflicting signals carry no direction), and scores exact agreement.
+    """
+    def collapse(x: torch.Tensor) -> torch.Tensor:
+        b = x.to(torch.bool)
+        up = b[:, 0] | b[:, 1]
+        down = b[:, 2] | b[:, 3]
+        # 0 = stall (neither, or conflicting up&down), 1 = up, 2 = down
+        return torch.where(up & ~down, torch.ones_like(up, dtype=torch.long),
+                           torch.where(down & ~up,
+                                       torch.full_like(up, 2, dtype=torch.long),
+                                       torch.zeros_like(up, dtype=torch.long)))
+
+    return (collapse(target) == collapse(pred)).float().mean()

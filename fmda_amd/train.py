@@ -95,9 +95,13 @@ def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
             chunk_batches(market, train_set, dcfg, tcfg.batch_size))
         va = model.evaluate_model(
             chunk_batches(market, val_set, dcfg, tcfg.batch_size))
+        from .metrics import three_class_accuracy
         rec = {"epoch": epoch, "train_acc": float(tr[0]),
                "train_hamming": float(tr[1]), "train_loss": float(tr[2]),
                "val_acc": float(va[0]), "val_hamming": float(va[1]),
+               # derived up/down/stall accuracy over the 4-label head
+               # (SURVEY.md section 6 labeling nuance)
+               "val_acc3": float(three_class_accuracy(va[4], va[3])),
                "sec": time.time() - t0}
         history.append(rec)
         log(json.dumps(rec))

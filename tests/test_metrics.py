@@ -28,3 +28,24 @@ def test_zero_division_matches_sklearn():
     ref = fbeta_score(t, p, beta=0.5, average=None, zero_division=0)
     ours = fbeta_per_class(t, p, beta=0.5).numpy()
     assert np.allclose(ours, ref)
+
+
+def test_three_class_accuracy():
+    from fmda_amd.metrics import three_class_accuracy
+    import torch
+    # rows: up-match, down-match, stall-match, up-vs-down, conflict->stall
+    t = torch.tensor([[1, 0, 0, 0], [0, 0, 1, 0], [0, 0, 0, 0],
+                      [0, 1, 0, 0], [1, 0, 1, 0]])
+    p = torch.tensor([[0, 1, 0, 0],   # up2 vs up1: still up -> match
+                      [0, 0, 0, 1],   # down -> match
+                      [1, 0, 1, 0],   # conflict collapses to stall -> match
+                      [0, 0, 1, 0],   # up vs down -> miss
+                      [0, 0, 0, 0]])  # stall vs conflict-stall -> match
+    assert abs(float(three_class_accuracy(t, p)) - 0.8) < 1e-6
+
+
+def test_three_class_all_match_is_one():
+    from fmda_amd.metrics import three_class_accuracy
+    import torch
+    t = (torch.rand(64, 4) < 0.3).long()
+    assert float(three_class_accuracy(t, t)) == 1.0
