@@ -50,6 +50,20 @@ def batch_metrics(target: torch.Tensor, pred: torch.Tensor,
             fbeta_per_class(target, pred, beta))
 
 
+def multilabel_confusion(target: torch.Tensor,
+                         pred: torch.Tensor) -> torch.Tensor:
+    """sklearn.metrics.multilabel_confusion_matrix: per-class 2x2
+    [[tn, fp], [fn, tp]] (the notebook's per-class confusion matrices,
+    reference training notebook cells 31/37)."""
+    t = target.to(torch.float32)
+    p = pred.to(torch.float32)
+    tp = (t * p).sum(dim=0)
+    fp = ((1 - t) * p).sum(dim=0)
+    fn = (t * (1 - p)).sum(dim=0)
+    tn = ((1 - t) * (1 - p)).sum(dim=0)
+    return torch.stack([tn, fp, fn, tp], dim=1).reshape(-1, 2, 2).long()
+
+
 def three_class_accuracy(target: torch.Tensor,
                          pred: torch.Tensor) -> torch.Tensor:
     """Derived up/down/stall accuracy over the 4-label head.

@@ -108,6 +108,23 @@ def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
         if rec["val_acc"] >= best_val_acc:
             best_val_acc = rec["val_acc"]
             save_checkpoint(model, checkpoint_path)
+
+    # Held-out test evaluation with per-class confusion matrices
+    # (reference notebook cells 33-37).
+    from .metrics import multilabel_confusion, three_class_accuracy
+    _, _, test_set = make_epoch_sets(market, dcfg, norm_params_path)
+    te = model.evaluate_model(
+        chunk_batches(market, test_set, dcfg, tcfg.batch_size))
+    import numpy as np
+    has_rows = te[3].numel() > 0  # tiny configs can have an empty test split
+    test_rec = {"test_acc": float(te[0]), "test_hamming": float(te[1]),
+                "test_fbeta": [float(v) for v in np.atleast_1d(te[2])],
+                "test_acc3": (float(three_class_accuracy(te[4], te[3]))
+                              if has_rows else None),
+                "confusion": (multilabel_confusion(te[4], te[3]).tolist()
+                              if has_rows else None)}
+    history.append(test_rec)
+    log(json.dumps(test_rec))
     return model, history
 
 

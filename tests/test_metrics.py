@@ -49,3 +49,15 @@ def test_three_class_all_match_is_one():
     import torch
     t = (torch.rand(64, 4) < 0.3).long()
     assert float(three_class_accuracy(t, t)) == 1.0
+
+
+def test_multilabel_confusion_matches_sklearn():
+    from fmda_amd.metrics import multilabel_confusion
+    from sklearn.metrics import multilabel_confusion_matrix
+    import torch
+    g = torch.Generator().manual_seed(7)
+    t = (torch.rand(50, 4, generator=g) < 0.3).long()
+    p = (torch.rand(50, 4, generator=g) < 0.3).long()
+    ours = multilabel_confusion(t, p).numpy()
+    ref = multilabel_confusion_matrix(t.numpy(), p.numpy())
+    assert (ours == ref).all()
