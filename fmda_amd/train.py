@@ -123,7 +123,6 @@ def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
                               if has_rows else None),
                 "confusion": (multilabel_confusion(te[4], te[3]).tolist()
                               if has_rows else None)}
-    history.append(test_rec)
     log(json.dumps(test_rec))
     return model, history
 
