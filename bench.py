@@ -31,14 +31,8 @@ from fmda_amd.models import BiGRU
 
 def three_class_accuracy(target: torch.Tensor, pred: torch.Tensor) -> float:
     """Derived up/down/stall accuracy (SURVEY.md section 6 labeling note)."""
-    def to3(lbl):
-        up = lbl[:, 0:2].any(dim=1)
-        dn = lbl[:, 2:4].any(dim=1)
-        cls = torch.zeros(lbl.shape[0], dtype=torch.long, device=lbl.device)
-        cls[up & ~dn] = 1
-        cls[dn & ~up] = 2
-        return cls
-    return float((to3(target.bool()) == to3(pred.bool())).float().mean())
+    from fmda_amd.metrics import three_class_accuracy as _acc3
+    return float(_acc3(target, pred))
 
 
 def bench_predict(cfg, args, use_cuda):
@@ -161,10 +155,20 @@ def main():
     acc3_sum = 0.0
     n_acc = 0
 
+    # Optional roctx ranges (FMDA_ROCTX=1): torch's nvtx API emits roctx
+    # markers on ROCm, letting `rocprofv3 --marker-trace` segment the step
+    # into forward / backward / optimizer phases.
+    if os.environ.get("FMDA_ROCTX") == "1" and use_cuda:
+        rng_push, rng_pop = torch.cuda.nvtx.range_push, torch.cuda.nvtx.range_pop
+    else:
+        rng_push = rng_pop = None
+
     def step(i, measure_quality=False):
         nonlocal acc3_sum, n_acc
         x, y = pool[i % len(pool)]
         opt.zero_grad(set_to_none=False)
+        if rng_push:
+            rng_push("forward")
         if use_cuda:
             # fused head GEMM + BCEWithLogitsLoss(weight, pos_weight) kernel
             from fmda_amd.ops.interface import fused_head_loss
@@ -175,10 +179,18 @@ def main():
         else:
             logits = model(x)
             loss = loss_fn(logits.float(), y)
+        if rng_push:
+            rng_pop()
+            rng_push("backward")
         loss.backward()
         if engine is not None:
             engine.finalize()
+        if rng_push:
+            rng_pop()
+            rng_push("optimizer")
         opt.step()  # fused clip(50) + Adam
+        if rng_push:
+            rng_pop()
         pred = torch.sigmoid(logits.detach().float()) > 0.5
         acc_sum.add_(subset_accuracy(y, pred))  # stays on device, no sync
         if measure_quality:
