@@ -47,11 +47,15 @@ class FusedClipAdam:
             }
 
     def zero_grad(self, set_to_none: bool = True):
-        for p in self.params:
-            if set_to_none:
+        if set_to_none:
+            for p in self.params:
                 p.grad = None
-            elif p.grad is not None:
-                p.grad.zero_()
+            return
+        grads = [p.grad for p in self.params if p.grad is not None]
+        if grads:
+            # one (or a few) multi-tensor launches instead of a per-tensor
+            # FillFunctor each — ~25 fewer kernel launches per step
+            torch._foreach_zero_(grads)
 
     def _build_chunks(self):
         rows = []
