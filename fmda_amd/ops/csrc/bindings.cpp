@@ -20,13 +20,6 @@ extern "C" int fmda_gru_bwd_cs_launch(const void* gi, const void* w,
                                       float* dh0, float* dbhh, void* gpub,
                                       unsigned int* cnt, int B, int Tseq,
                                       int n_dir, hipStream_t stream);
-extern "C" int fmda_gru_bwd_fused_launch(const void* gi, const void* w,
-                                         const void* wt, const float* bhh,
-                                         const void* out, const void* dout,
-                                         const float* dhT, void* dgi,
-                                         float* dwpart, float* dh0,
-                                         float* dbhh, int B, int Tseq,
-                                         int n_dir, hipStream_t stream);
 extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const void* wt,
                                    const float* bhh,
@@ -192,48 +185,6 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
     return {dgi, dgh, dh0, dbhh.contiguous(), dbih};
 }
 
-std::vector<torch::Tensor> gru_bwd_fused(torch::Tensor gi, torch::Tensor w,
-                                         torch::Tensor bhh, torch::Tensor out,
-                                         torch::Tensor dout,
-                                         torch::Tensor dhT) {
-    // v4 path (bf16, Hp=128): dW_hh accumulated in-kernel from the LDS
-    // tiles; no dGh materialization, no host split-K GEMM. Returns
-    // {dgi, dwhh (n_dir, 3Hp, Hp) fp32, dh0, dbhh, dbih}.
-    int B, T, n_dir, Hp;
-    bool is_bf16;
-    check_common(gi, w, bhh, B, T, n_dir, Hp, is_bf16);
-    TORCH_CHECK(is_bf16 && Hp == 128,
-                "gru_bwd_fused: bf16 Hp=128 only, got Hp=", Hp);
-    TORCH_CHECK(out.is_contiguous() && dout.is_contiguous() &&
-                dhT.is_contiguous(),
-                "fmda gru_bwd_fused: tensors must be contiguous");
-    TORCH_CHECK(dhT.scalar_type() == torch::kFloat32, "dhT must be fp32");
-    auto dgi = getenv("FMDA_ZERO_GRADS") ? torch::zeros_like(gi)
-                                         : torch::empty_like(gi);
-    const int BT = 32;
-    const int GX = (B + BT - 1) / BT;
-    auto dwpart = torch::empty({(int64_t)n_dir * GX, 3 * Hp, Hp},
-                               gi.options().dtype(torch::kFloat32));
-    auto dh0 = torch::empty({n_dir, B, Hp},
-                            gi.options().dtype(torch::kFloat32));
-    auto dbsum = torch::zeros({n_dir, 4 * Hp},
-                              gi.options().dtype(torch::kFloat32));
-    auto wt = w.transpose(1, 2).contiguous();
-    auto stream = at::hip::getCurrentHIPStream();
-    int rc = fmda_gru_bwd_fused_launch(
-        gi.data_ptr(), w.data_ptr(), wt.data_ptr(), bhh.data_ptr<float>(),
-        out.data_ptr(), dout.data_ptr(), dhT.data_ptr<float>(),
-        dgi.data_ptr(), dwpart.data_ptr<float>(), dh0.data_ptr<float>(),
-        dbsum.data_ptr<float>(), B, T, n_dir, stream.stream());
-    TORCH_CHECK(rc == 0, "fmda gru_bwd_fused launch failed rc=", rc);
-    auto dwhh = dwpart.view({n_dir, GX, 3 * Hp, Hp}).sum(1);
-    using torch::indexing::Slice;
-    auto dbhh = dbsum.index({Slice(), Slice(0, 3 * Hp)});
-    auto dbih = torch::cat({dbsum.index({Slice(), Slice(0, 2 * Hp)}),
-                            dbsum.index({Slice(), Slice(3 * Hp, 4 * Hp)})}, 1);
-    return {dgi, dwhh, dh0, dbhh.contiguous(), dbih};
-}
-
 torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor Bm) {
     TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16);
     TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}) &&
@@ -360,8 +311,6 @@ torch::Tensor dropout_fused(torch::Tensor x, double p, int64_t seed) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gru_fwd", &gru_fwd, "fused biGRU recurrence forward (HIP/CDNA4)");
     m.def("gru_bwd", &gru_bwd, "fused biGRU recurrence backward (HIP/CDNA4)");
-    m.def("gru_bwd_fused", &gru_bwd_fused,
-          "biGRU backward with in-kernel dW_hh accumulation (bf16 Hp=128)");
     m.def("mfma_selftest", &mfma_selftest, "mfma fragment layout self-test");
     m.def("dropout_fused", &dropout_fused,
           "counter-based dropout (mask recomputed in backward)");

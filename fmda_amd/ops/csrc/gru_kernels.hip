@@ -41,7 +41,6 @@
 //   n = tanh(i_n + r * (W_hn h + b_hn))
 //   h' = (1 - z) * n + z * h
 #include <hip/hip_runtime.h>
-#include <cstdlib>
 #include <hip/hip_bf16.h>
 
 #define FMDA_DEV __device__ __forceinline__
@@ -1456,449 +1455,6 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 }
 
 
-template <int BT, int Hp, int NT, int WPE>
-__global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
-                          amdgpu_waves_per_eu(WPE, WPE)))
-void gru_bwd_v4_kernel(const __hip_bfloat16* __restrict__ gi,
-                       const __hip_bfloat16* __restrict__ w,
-                       const __hip_bfloat16* __restrict__ wt,
-                       const float* __restrict__ bhh,
-                       const __hip_bfloat16* __restrict__ out,
-                       const __hip_bfloat16* __restrict__ dout,
-                       const float* __restrict__ dhT,
-                       __hip_bfloat16* __restrict__ dgi,
-                       float* __restrict__ dwpart, float* __restrict__ dh0,
-                       float* __restrict__ dbhh, int B, int Tseq, int n_dir) {
-    constexpr int NW = NT / 64;
-    constexpr int MT = BT / 16;
-    constexpr int NCT = Hp / 16;
-    constexpr int CPW = NCT / NW;
-    constexpr int KK = Hp / 32;
-    constexpr int KK2 = (3 * Hp) / 32;
-    constexpr int GP = 3 * Hp;
-    constexpr int HROW = Hp * 2;         // h/dout tile row bytes (unpadded)
-    constexpr int GP3 = 3 * Hp + 8;      // padded dgh_s pitch (MFMA A reads)
-    constexpr int PW = (BT * GP * 2) / 1024 / NW;   // gi glds per wave
-    constexpr int DOP = (BT * HROW) / 1024 / NW;    // dout glds per wave
-    constexpr int HBP = DOP;                        // h glds per wave
-    constexpr int NST = ((BT * GP * 2) / 16) / NT;  // dGi store instrs only
-    static_assert(NCT % NW == 0 && (BT * GP * 2) % (1024 * NW) == 0 &&
-                  (BT * HROW) % (1024 * NW) == 0, "tiling mismatch");
-
-    const int tid = threadIdx.x;
-    const int wave = tid >> 6;
-    const int lane = tid & 63;
-    const int b0 = blockIdx.x * BT;
-    const int dir = blockIdx.y;
-    const bool rev = (dir == 1);
-    const int rows_valid = min(BT, B - b0);
-
-    extern __shared__ __attribute__((aligned(16))) char smem[];
-    char* p = smem;
-    // Plain pointers + select helpers, NOT runtime-indexed local arrays:
-    // a dynamically indexed private pointer array lands in scratch, and
-    // every scratch reload is an ordinary VMEM load that triggers the
-    // glds vmcnt(0) drain (this alone cost 3x on this kernel).
-    __hip_bfloat16* gi_s0 = (__hip_bfloat16*)p; p += 2 * BT * GP;
-    __hip_bfloat16* gi_s1 = (__hip_bfloat16*)p; p += 2 * BT * GP;
-    __hip_bfloat16* gi_s2 = (__hip_bfloat16*)p; p += 2 * BT * GP;
-    __hip_bfloat16* hb_s0 = (__hip_bfloat16*)p; p += BT * HROW;
-    __hip_bfloat16* hb_s1 = (__hip_bfloat16*)p; p += BT * HROW;
-    __hip_bfloat16* do_s0 = (__hip_bfloat16*)p; p += BT * HROW;
-    __hip_bfloat16* do_s1 = (__hip_bfloat16*)p; p += BT * HROW;
-    __hip_bfloat16* dgh_s0 = (__hip_bfloat16*)p; p += 2 * BT * GP3;
-    __hip_bfloat16* dgh_s1 = (__hip_bfloat16*)p; p += 2 * BT * GP3;
-    float* bhh_s = (float*)p;
-    auto dgh_buf = [&](int b) { return b ? dgh_s1 : dgh_s0; };
-    auto gi_slot = [&](int sl) {
-        return sl == 0 ? gi_s0 : (sl == 1 ? gi_s1 : gi_s2);
-    };
-    auto hb_buf = [&](int b) { return b ? hb_s1 : hb_s0; };
-    auto do_buf = [&](int b) { return b ? do_s1 : do_s0; };
-
-    const long gi_row = (long)Tseq * n_dir * 3 * Hp;
-    const long out_row = (long)Tseq * n_dir * Hp;
-    const __hip_bfloat16* gi_b = gi + (long)b0 * gi_row + (long)dir * 3 * Hp;
-    __hip_bfloat16* dgi_b = dgi + (long)b0 * gi_row + (long)dir * 3 * Hp;
-    const __hip_bfloat16* out_b = out + (long)b0 * out_row + (long)dir * Hp;
-    const __hip_bfloat16* dout_b = dout + (long)b0 * out_row + (long)dir * Hp;
-
-    for (int c = tid; c < 3 * Hp; c += NT)
-        bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
-
-    // Both W fragment sets, register-resident; loaded ONCE here (ordinary
-    // loads are legal before any glds is issued).
-    const __hip_bfloat16* wdir = w + (long)dir * 3 * Hp * Hp;
-    const __hip_bfloat16* wt_dir = wt + (long)dir * Hp * 3 * Hp;
-    bf16x8_t wA[CPW][3][KK];
-    bf16x8_t wB[CPW][KK2];
-#pragma unroll
-    for (int i = 0; i < CPW; ++i) {
-        const int ct = wave + NW * i;
-        const int jcol = ct * 16 + (lane & 15);
-        const __bf16* wtrow = (const __bf16*)wt_dir + (long)jcol * 3 * Hp;
-#pragma unroll
-        for (int g = 0; g < 3; ++g)
-#pragma unroll
-            for (int kk = 0; kk < KK; ++kk)
-                wA[i][g][kk] = load_wfragA<Hp>(wdir, Hp, ct, g, kk, lane);
-#pragma unroll
-        for (int kk = 0; kk < KK2; ++kk)
-            wB[i][kk] = *(const bf16x8_t*)&wtrow[32 * kk + 8 * (lane >> 4)];
-    }
-
-    float dhreg[CPW][MT][4];   // dh carry (fp32, lane-owned)
-    // in-kernel dW_hh partial: wave w owns gate-row tiles 3w..3w+2 x all
-    // Hp/16 h-column tiles; accumulated in fp32 across all timesteps and
-    // written once in the epilogue (replaces the 1.5 GB dGh materialization
-    // + its split-K GEMM on the host). Requires NT=512, Hp=128.
-    constexpr int JT = GP / 16 / NW;   // gate-row tiles per wave (=3)
-    constexpr int KT = Hp / 16;        // h-column tiles (=8)
-    static_assert(JT * NW * 16 == GP, "gate tiling mismatch");
-    f32x4_t dwacc[JT][KT];
-#pragma unroll
-    for (int jt = 0; jt < JT; ++jt)
-#pragma unroll
-        for (int kt = 0; kt < KT; ++kt) dwacc[jt][kt] = f32x4_t{0.f};
-    // 4 running sums: dr, dz, dhn (-> db_hh) and dn (-> db_ih; dr/dz are
-    // shared between the two bias gradients).
-    float dbacc[CPW][4];
-#pragma unroll
-    for (int i = 0; i < CPW; ++i) {
-#pragma unroll
-        for (int g = 0; g < 4; ++g) dbacc[i][g] = 0.0f;
-        const int ct = wave + NW * i;
-        const int j = ct * 16 + (lane & 15);
-        const float* hT = dhT + ((long)dir * B + b0) * Hp;
-#pragma unroll
-        for (int m = 0; m < MT; ++m)
-#pragma unroll
-            for (int e = 0; e < 4; ++e) {
-                const int b = 16 * m + 4 * (lane >> 4) + e;
-                dhreg[i][m][e] =
-                    (b < rows_valid) ? hT[(long)b * Hp + j] : 0.0f;
-            }
-    }
-
-    const int wavu = __builtin_amdgcn_readfirstlane(wave);
-    auto glds_gi = [&](int tt, int sl) {
-        const __hip_bfloat16* src_t = gi_b + (long)tt * n_dir * 3 * Hp;
-        __hip_bfloat16* dst = gi_slot(sl);
-#pragma unroll
-        for (int k = 0; k < PW; ++k) {
-            const int off = (wavu * PW + k) * 1024 + lane * 16;
-            const int e = off >> 1;
-            int r = e / GP;
-            const int c = e % GP;
-            if (r >= rows_valid) r = 0;
-            glds16(src_t + (long)r * gi_row + c,
-                   (char*)dst + (long)(wavu * PW + k) * 1024);
-        }
-    };
-    auto glds_do = [&](int tt, int buf) {
-        const __hip_bfloat16* src_t = dout_b + (long)tt * n_dir * Hp;
-#pragma unroll
-        for (int k = 0; k < DOP; ++k) {
-            const int off = (wavu * DOP + k) * 1024 + lane * 16;
-            const int e = off >> 1;
-            int r = e / Hp;
-            const int c = e % Hp;
-            if (r >= rows_valid) r = 0;
-            glds16(src_t + (long)r * out_row + c,
-                   (char*)do_buf(buf) + (long)(wavu * DOP + k) * 1024);
-        }
-    };
-    // XOR-swizzled source: LDS[r][k16] = out[r][k16 ^ (r & 15)] so the
-    // MFMA A-fragment reads are bank-conflict-free at the unpadded pitch.
-    auto glds_hb = [&](int tt, int buf) {
-        const __hip_bfloat16* src_t = out_b + (long)tt * n_dir * Hp;
-#pragma unroll
-        for (int k = 0; k < HBP; ++k) {
-            const int off = (wavu * HBP + k) * 1024 + lane * 16;
-            int r = off / HROW;
-            const int k16 = (off % HROW) >> 4;
-            const int csrc = (k16 ^ (r & 15)) << 4;  // byte col in row
-            if (r >= rows_valid) r = 0;
-            glds16((const char*)(src_t + (long)r * out_row) + csrc,
-                   (char*)hb_buf(buf) + (long)(wavu * HBP + k) * 1024);
-        }
-    };
-    auto zero_hb = [&](int buf) {
-        for (int c = tid; c < BT * Hp; c += NT)
-            ((__bf16*)hb_buf(buf))[c] = (__bf16)0.0f;
-    };
-    // swizzled scalar read of h_prev[b][j]
-    auto hb_read = [&](const __hip_bfloat16* hb, int b, int j) -> float {
-        const int k16 = (j * 2) >> 4;           // 16B slot of column j
-        const int ksw = k16 ^ (b & 15);
-        const int within = (j * 2) & 15;
-        return (float)*(const __bf16*)((const char*)hb + b * HROW +
-                                       ksw * 16 + within);
-    };
-
-    {   // prologue: stage step Tseq-1 inputs
-        const int u = Tseq - 1;
-        const int tt = rev ? 0 : u;
-        glds_gi(tt, u % 3);
-        glds_do(tt, u & 1);
-        if (u > 0) {
-            const int ttp = rev ? 1 : (u - 1);
-            glds_hb(ttp, u & 1);
-        } else {
-            zero_hb(u & 1);
-        }
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
-    }
-
-    for (int u = Tseq - 1; u >= 0; --u) {
-        const int q = u & 1;
-        const int slot = u % 3;
-        const int tt = rev ? (Tseq - 1 - u) : u;
-        const bool have_next = (u > 0);
-        const bool have_prev = (u + 1 < Tseq);  // a step was processed before
-
-        // ---- phase A: DMA issues, then one-step-late stores ----
-        if (have_next) {
-            const int ttn = rev ? (Tseq - u) : (u - 1);
-            glds_gi(ttn, (u - 1) % 3);
-            glds_do(ttn, 1 - q);
-            if (u >= 2)
-                glds_hb(rev ? (Tseq + 1 - u) : (u - 2), 1 - q);
-            else
-                zero_hb(1 - q);
-        }
-        if (have_prev) {
-            // one-step-late stores; dgh_s is double-buffered so these
-            // reads never race phase B's writes and no A-end barrier is
-            // needed (the single per-step barrier is at B-end, where it
-            // also rendezvouses this step's DMA).
-            const int ttp = rev ? (Tseq - 2 - u) : (u + 1);
-            store_tile<__hip_bfloat16, BT, NT>(
-                gi_slot((u + 1) % 3), dgi_b + (long)ttp * n_dir * 3 * Hp, GP, GP,
-                gi_row, rows_valid, tid);
-        }
-
-        // ---- phase B: recompute GEMM, then fused gate gradients ----
-        const __hip_bfloat16* hbq = hb_buf(q);
-        f32x4_t acc[CPW][3][MT];
-#pragma unroll
-        for (int i = 0; i < CPW; ++i)
-#pragma unroll
-            for (int g = 0; g < 3; ++g)
-#pragma unroll
-                for (int m = 0; m < MT; ++m) acc[i][g][m] = f32x4_t{0.f};
-#pragma unroll
-        for (int i = 0; i < CPW; ++i) {
-            const int arow = lane & 15;
-#pragma unroll
-            for (int kk = 0; kk < KK; ++kk) {
-                bf16x8_t a[MT];
-#pragma unroll
-                for (int m = 0; m < MT; ++m) {
-                    const int row = 16 * m + arow;
-                    const int k16 = (4 * kk + (lane >> 4)) ^ (row & 15);
-                    a[m] = *(const bf16x8_t*)((const char*)hbq +
-                                              row * HROW + k16 * 16);
-                }
-#pragma unroll
-                for (int g = 0; g < 3; ++g)
-#pragma unroll
-                    for (int m = 0; m < MT; ++m)
-                        acc[i][g][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            a[m], wA[i][g][kk], acc[i][g][m], 0, 0, 0);
-            }
-        }
-#pragma unroll
-        for (int i = 0; i < CPW; ++i) {
-            const int ct = wave + NW * i;
-            const int j = ct * 16 + (lane & 15);
-            __bf16* grow = (__bf16*)gi_slot(slot);
-            __bf16* dgrow = (__bf16*)dgh_buf(q);
-            const __bf16* dorow = (const __bf16*)do_buf(q);
-#pragma unroll
-            for (int m = 0; m < MT; ++m) {
-#pragma unroll
-                for (int e = 0; e < 4; ++e) {
-                    const int b = 16 * m + 4 * (lane >> 4) + e;
-                    const bool live = (b < rows_valid);
-                    const float gr = acc[i][0][m][e] + bhh_s[j];
-                    const float gz = acc[i][1][m][e] + bhh_s[Hp + j];
-                    const float hn = acc[i][2][m][e] + bhh_s[2 * Hp + j];
-                    const float ir = (float)grow[b * GP + j];
-                    const float iz = (float)grow[b * GP + Hp + j];
-                    const float in_ = (float)grow[b * GP + 2 * Hp + j];
-                    const float r = sigmoidf(ir + gr);
-                    const float z = sigmoidf(iz + gz);
-                    const float n = fast_tanh(in_ + r * hn);
-                    const float hprev = hb_read(hbq, b, j);
-                    const float dht =
-                        dhreg[i][m][e] + (live ? (float)dorow[b * Hp + j]
-                                               : 0.0f);
-                    // live-gating every product: 0 * inf = NaN would leak
-                    // into dbacc from uninitialized dead-row LDS.
-                    const float dz_pre =
-                        live ? dht * (hprev - n) * z * (1.0f - z) : 0.0f;
-                    const float dn_pre =
-                        live ? dht * (1.0f - z) * (1.0f - n * n) : 0.0f;
-                    const float dr_pre =
-                        live ? dn_pre * hn * r * (1.0f - r) : 0.0f;
-                    const float dhn = live ? dn_pre * r : 0.0f;
-                    grow[b * GP + j] = (__bf16)__float2bfloat16(dr_pre);
-                    grow[b * GP + Hp + j] = (__bf16)__float2bfloat16(dz_pre);
-                    grow[b * GP + 2 * Hp + j] =
-                        (__bf16)__float2bfloat16(dn_pre);
-                    dgrow[b * GP3 + j] = (__bf16)__float2bfloat16(dr_pre);
-                    dgrow[b * GP3 + Hp + j] = (__bf16)__float2bfloat16(dz_pre);
-                    dgrow[b * GP3 + 2 * Hp + j] =
-                        (__bf16)__float2bfloat16(dhn);
-                    dhreg[i][m][e] = live ? dht * z : 0.0f;
-                    dbacc[i][0] += dr_pre;
-                    dbacc[i][1] += dz_pre;
-                    dbacc[i][2] += dhn;
-                    dbacc[i][3] += dn_pre;
-                }
-            }
-        }
-        // single per-step barrier: publishes dgh_s[q] for phase C AND
-        // rendezvouses this step's DMA (tiles for step u-1, read at
-        // B(u-1)), leaving the trailing dGi/dGh stores in flight.
-        if (have_next) {
-            if (have_prev)
-                asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NST) : "memory");
-            else
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        } else {
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
-
-        // ---- phase C: carry GEMM dh += dGh W — registers + LDS only ----
-        {
-            f32x4_t acc2[CPW][MT];
-#pragma unroll
-            for (int i = 0; i < CPW; ++i)
-#pragma unroll
-                for (int m = 0; m < MT; ++m) acc2[i][m] = f32x4_t{0.f};
-#pragma unroll
-            for (int i = 0; i < CPW; ++i) {
-                const int arow = lane & 15;
-                const int koff = 8 * (lane >> 4);
-#pragma unroll
-                for (int kk = 0; kk < KK2; ++kk) {
-                    const int kbase = 32 * kk + koff;
-                    bf16x8_t a[MT];
-#pragma unroll
-                    for (int m = 0; m < MT; ++m)
-                        a[m] = *(const bf16x8_t*)&((const __bf16*)
-                                   dgh_buf(q))[(16 * m + arow) * GP3 + kbase];
-#pragma unroll
-                    for (int m = 0; m < MT; ++m)
-                        acc2[i][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            a[m], wB[i][kk], acc2[i][m], 0, 0, 0);
-                }
-            }
-#pragma unroll
-            for (int i = 0; i < CPW; ++i)
-#pragma unroll
-                for (int m = 0; m < MT; ++m)
-#pragma unroll
-                    for (int e = 0; e < 4; ++e)
-                        dhreg[i][m][e] += acc2[i][m][e];
-        }
-        // ---- phase C2: dW_hh partial accumulation (off the serial chain).
-        // dW[j][k] += sum_b dGh_u[b][j] * h_{u-1}[b][k]; both operands are
-        // the LDS tiles phase B just published (dgh_buf(q), hbq) — the
-        // step-u pairing is exactly the time-shifted host GEMM's. MFMA
-        // M=GP(j) N=Hp(k) K=BT(b): one 16x16x32 per tile covers all 32
-        // batch rows; transposed A reads are scalar ds_read_u16 (conflict
-        // cost irrelevant: this phase overlaps the next step's latency).
-        {
-            const __bf16* dsrc = (const __bf16*)dgh_buf(q);
-            const int bb = 8 * (lane >> 4);
-#pragma unroll
-            for (int jt = 0; jt < JT; ++jt) {
-                bf16x8_t afrag;
-                const int j = (JT * wave + jt) * 16 + (lane & 15);
-#pragma unroll
-                for (int e = 0; e < 8; ++e)
-                    afrag[e] = dsrc[(bb + e) * GP3 + j];
-#pragma unroll
-                for (int kt = 0; kt < KT; ++kt) {
-                    bf16x8_t bfrag;
-                    const int k = kt * 16 + (lane & 15);
-                    const int k16 = (k * 2) >> 4;
-                    const int within = (k * 2) & 15;
-#pragma unroll
-                    for (int e = 0; e < 8; ++e) {
-                        const int b = bb + e;
-                        bfrag[e] = *(const __bf16*)((const char*)hbq +
-                            b * HROW + (k16 ^ (b & 15)) * 16 + within);
-                    }
-                    // "+a" pins the accumulator into the AGPR half of the
-                    // unified register file: the 96 accumulation registers
-                    // must not raise arch-VGPR pressure (the wA/wB hoists
-                    // already fill it), and the compiler will not allocate
-                    // AGPRs on its own on gfx950.
-                    asm volatile("v_mfma_f32_16x16x32_bf16 %0, %1, %2, %0"
-                                 : "+a"(dwacc[jt][kt])
-                                 : "v"(afrag), "v"(bfrag));
-                }
-            }
-        }
-        // no barrier: the next step's A-end rendezvous orders everything.
-    }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
-
-    {   // epilogue: last dGi store, boundary zero dGh slot, dh0, db_hh
-        const int ttl = rev ? (Tseq - 1) : 0;
-        store_tile<__hip_bfloat16, BT, NT>(
-            gi_s0, dgi_b + (long)ttl * n_dir * 3 * Hp, GP, GP, gi_row,
-            rows_valid, tid);
-        // per-block dW_hh partial -> (dir, bx) slice; host sums over bx.
-        float* wp = dwpart +
-            ((long)dir * gridDim.x + blockIdx.x) * GP * Hp;
-#pragma unroll
-        for (int jt = 0; jt < JT; ++jt)
-#pragma unroll
-            for (int kt = 0; kt < KT; ++kt)
-#pragma unroll
-                for (int v = 0; v < 4; ++v) {
-                    const int j = (JT * wave + jt) * 16 + 4 * (lane >> 4) + v;
-                    const int k = kt * 16 + (lane & 15);
-                    wp[(long)j * Hp + k] = dwacc[jt][kt][v];
-                }
-        float* d0 = dh0 + ((long)dir * B + b0) * Hp;
-#pragma unroll
-        for (int i = 0; i < CPW; ++i) {
-            const int ct = wave + NW * i;
-            const int j = ct * 16 + (lane & 15);
-#pragma unroll
-            for (int m = 0; m < MT; ++m)
-#pragma unroll
-                for (int e = 0; e < 4; ++e) {
-                    const int b = 16 * m + 4 * (lane >> 4) + e;
-                    if (b < rows_valid) d0[(long)b * Hp + j] = dhreg[i][m][e];
-                }
-#pragma unroll
-            for (int g = 0; g < 4; ++g) {
-                // dbhh layout (n_dir, 4Hp): [dr, dz, dhn | dn] — db_hh =
-                // slots 0..2, db_ih = slots 0,1,3 (assembled host-side).
-                float v = dbacc[i][g];
-                v += __shfl_xor(v, 16);
-                v += __shfl_xor(v, 32);
-                if ((lane >> 4) == 0)
-                    atomicAdd(&dbhh[(long)dir * 4 * Hp + g * Hp + ct * 16 +
-                                    (lane & 15)], v);
-            }
-        }
-    }
-}
-
-
 // ===========================================================================
 // Column-split persistent forward for LARGE hidden sizes (Hp = 512).
 //
@@ -3029,39 +2585,6 @@ static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
         (__hip_bfloat16*)dgi, (__hip_bfloat16*)dgh, dh0, dbhh, B, Tseq,
         n_dir);
     return 0;
-}
-
-// v4: dW_hh accumulated in-kernel (no dGh materialization); bf16 Hp=128.
-extern "C" int fmda_gru_bwd_fused_launch(const void* gi, const void* w,
-                                         const void* wt, const float* bhh,
-                                         const void* out, const void* dout,
-                                         const float* dhT, void* dgi,
-                                         float* dwpart, float* dh0,
-                                         float* dbhh, int B, int Tseq,
-                                         int n_dir, hipStream_t stream) {
-    constexpr int BT = 32, Hp = 128;
-    // FMDA_BWD_V4_NT512=1 selects the 512-thread variant (2 waves/SIMD but
-    // register-starved: ~790 B/lane scratch); default is the 256-thread
-    // 1-wave/SIMD variant where everything fits the 512-slot unified file.
-    static const bool nt512 = [] {
-        const char* e = getenv("FMDA_BWD_V4_NT512");
-        return e && e[0] == '1';
-    }();
-    const int NT = nt512 ? 512 : 256;
-    const size_t lds = 3 * 2 * BT * 3 * Hp + 2 * 2 * BT * Hp +
-                       2 * 2 * BT * Hp + 2 * 2 * BT * (3 * Hp + 8) +
-                       4 * 3 * Hp;
-    auto k = nt512 ? gru_bwd_v4_kernel<BT, Hp, 512, 2>
-                   : gru_bwd_v4_kernel<BT, Hp, 256, 1>;
-    (void)hipFuncSetAttribute((const void*)k,
-        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
-    const dim3 grid((B + BT - 1) / BT, n_dir);
-    k<<<grid, NT, lds, stream>>>(
-        (const __hip_bfloat16*)gi, (const __hip_bfloat16*)w,
-        (const __hip_bfloat16*)wt, bhh,
-        (const __hip_bfloat16*)out, (const __hip_bfloat16*)dout, dhT,
-        (__hip_bfloat16*)dgi, dwpart, dh0, dbhh, B, Tseq, n_dir);
-    return hipGetLastError() == hipSuccess ? 0 : -4;
 }
 
 extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,

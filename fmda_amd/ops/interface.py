@@ -11,7 +11,6 @@ compose naturally.
 Replaces the reference's `self.gru(input_seq)` cuDNN RNN call site
 (biGRU_model.py:102).
 """
-import os
 from typing import List, Optional, Tuple
 
 import torch
@@ -275,22 +274,12 @@ class _BiGRULayer(torch.autograd.Function):
         need_dx = ctx.needs_input_grad[0]
         d_out = d_out.contiguous().to(gi.dtype)
         d_hlast = d_hlast.contiguous().float()
-        fused = (Hp == 128 and gi.dtype == torch.bfloat16
-                 and hasattr(ext, "gru_bwd_fused")
-                 and os.environ.get("FMDA_BWD_V4", "1") == "1")
-        if fused:
-            # v4 kernel: dW_hh partials accumulated in-kernel from the LDS
-            # tiles — no dGh materialization, no host split-K GEMM.
-            dgi, dwhh_all, _dh0, dbhh, dbih = ext.gru_bwd_fused(
-                gi, w_hh_cat, b_hh_cat, out, d_out, d_hlast)
-        else:
-            dgi, dgh, _dh0, dbhh, dbih = ext.gru_bwd(
-                gi, w_hh_cat, b_hh_cat, out, d_out, d_hlast)
+        dgi, dgh, _dh0, dbhh, dbih = ext.gru_bwd(
+            gi, w_hh_cat, b_hh_cat, out, d_out, d_hlast)
         M = dgi.shape[0] * dgi.shape[1]
 
-        if not fused:
-            # dW_hh via the time-shifted dGh and one split-K reduction
-            cross = chunked_outer(dgh.reshape(M, -1), out.reshape(M, -1))
+        # dW_hh via the time-shifted dGh and one split-K reduction
+        cross = chunked_outer(dgh.reshape(M, -1), out.reshape(M, -1))
         # dW_ih for both directions in one split-K reduction
         dwih_cat = chunked_outer(dgi.reshape(M, -1), x2d)
         dx = None
@@ -301,11 +290,8 @@ class _BiGRULayer(torch.autograd.Function):
         grads = []
         for d in range(D):
             dwih = dwih_cat[d * 3 * Hp:(d + 1) * 3 * Hp].float()
-            if fused:
-                dwhh = dwhh_all[d]
-            else:
-                dwhh = cross[d * 3 * Hp:(d + 1) * 3 * Hp,
-                             d * Hp:(d + 1) * Hp].float()
+            dwhh = cross[d * 3 * Hp:(d + 1) * 3 * Hp,
+                         d * Hp:(d + 1) * Hp].float()
             grads.append((
                 _unpad_gate_rows(dwih, H, Hp),
                 _unpad_gate_rows(dwhh, H, Hp)[:, :H],

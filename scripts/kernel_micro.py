@@ -40,13 +40,7 @@ def main():
     def run_bwd():
         return ext.gru_bwd(gi, w, bhh, out, dout, dhT)
 
-    def run_bwd_fused():
-        return ext.gru_bwd_fused(gi, w, bhh, out, dout, dhT)
-
-    candidates = [("fwd", run_fwd), ("bwd", run_bwd)]
-    if Hp == 128 and hasattr(ext, "gru_bwd_fused"):
-        candidates.append(("bwd_fused", run_bwd_fused))
-    for name, fn in candidates:
+    for name, fn in [("fwd", run_fwd), ("bwd", run_bwd)]:
         if args.what not in ("both", name):
             continue
         for _ in range(5):
