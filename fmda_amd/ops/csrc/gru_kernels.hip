@@ -1067,6 +1067,143 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 // Phases: A[glds issues + one-step-late dGi/dGh stores] -> rendezvous ->
 // B[recompute GEMM + fused gate grads] -> barrier -> C[carry GEMM from
 // dgh_s, register-only] -> (no barrier).
+// ===========================================================================
+// Batch-1 forward for streaming inference (the predict path).
+//
+// At batch 1 the batch-tiled forward launches 2 blocks on 256 CUs and pays
+// a per-step LDS-DMA rendezvous for a tile that contains one live row.
+// This kernel instead makes the whole problem LDS/register-resident:
+// the ENTIRE gi sequence (T x 3Hp bf16, ~94 KB at T=120) is staged into
+// LDS once, W_hh fragments are register-hoisted as in v3, and h crosses
+// timesteps through a 2-slot LDS ring — one __syncthreads per step and
+// zero global traffic in the loop except the 256 B/step `out` write.
+// MFMA layout: batch is the M dimension (A-fragment rows; only row 0 is
+// live, supplied by the l%16==0 lanes), so each output lane l<16 holds
+// r, z and n for the SAME h column across its three accumulators and the
+// gate math needs no cross-lane traffic. Latency-bound by design: 24
+// MFMAs + one barrier per step per wave.
+// ===========================================================================
+
+template <int Hp, int NT>
+__global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
+                          amdgpu_waves_per_eu(1, 1)))
+void gru_fwd_b1_kernel(const __hip_bfloat16* __restrict__ gi,
+                       const __hip_bfloat16* __restrict__ w,
+                       const float* __restrict__ bhh,
+                       __hip_bfloat16* __restrict__ out,
+                       float* __restrict__ hlast, int Tseq, int n_dir) {
+    constexpr int NW = NT / 64;      // 4 waves
+    constexpr int CT = Hp / 16;      // 8 h-column tiles
+    constexpr int CPW = CT / NW;     // 2 per wave
+    constexpr int KK = Hp / 32;
+    constexpr int GIP = 3 * Hp + 8;  // padded gi row pitch (elements)
+    constexpr int HRP = Hp + 8;      // h ring pitch
+    static_assert(CT % NW == 0, "tiling mismatch");
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int dir = blockIdx.x;
+    const bool rev = (dir == 1);
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    char* p = smem;
+    __hip_bfloat16* gi_s = (__hip_bfloat16*)p; p += 2 * (long)Tseq * GIP;
+    __hip_bfloat16* hr0 = (__hip_bfloat16*)p; p += 2 * HRP;
+    __hip_bfloat16* hr1 = (__hip_bfloat16*)p; p += 2 * HRP;
+    float* bhh_s = (float*)p;
+    auto hring = [&](int s) { return s ? hr1 : hr0; };
+
+    // stage the whole gi sequence (batch row 0, this direction)
+    const __hip_bfloat16* gi_b = gi + (long)dir * 3 * Hp;
+    const long gi_trow = (long)n_dir * 3 * Hp;
+    for (int idx = tid; idx < Tseq * (3 * Hp / 8); idx += NT) {
+        const int t = idx / (3 * Hp / 8);
+        const int c8 = idx % (3 * Hp / 8);
+        *(bf16x8_t*)(gi_s + (long)t * GIP + c8 * 8) =
+            *(const bf16x8_t*)(gi_b + t * gi_trow + c8 * 8);
+    }
+    for (int c = tid; c < 3 * Hp; c += NT)
+        bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
+    for (int c = tid; c < HRP; c += NT)
+        ((__bf16*)hr0)[c] = (__bf16)0.0f;
+
+    const __hip_bfloat16* wdir = w + (long)dir * 3 * Hp * Hp;
+    bf16x8_t wA[CPW][3][KK];
+#pragma unroll
+    for (int i = 0; i < CPW; ++i) {
+        const int ct = wave + NW * i;
+#pragma unroll
+        for (int g = 0; g < 3; ++g)
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk)
+                wA[i][g][kk] = load_wfragA<Hp>(wdir, Hp, ct, g, kk, lane);
+    }
+    float hreg[CPW];                 // this lane's h column (lanes 0..15)
+#pragma unroll
+    for (int i = 0; i < CPW; ++i) hreg[i] = 0.0f;
+
+    __hip_bfloat16* out_b = out + (long)dir * Hp;   // batch row 0
+    const long out_trow = (long)n_dir * Hp;
+    __syncthreads();
+
+    for (int u = 0; u < Tseq; ++u) {
+        const int tt = rev ? (Tseq - 1 - u) : u;
+        const __bf16* hs = (const __bf16*)hring(u & 1);
+
+        f32x4_t acc[CPW][3];
+#pragma unroll
+        for (int i = 0; i < CPW; ++i)
+#pragma unroll
+            for (int g = 0; g < 3; ++g) acc[i][g] = f32x4_t{0.f};
+#pragma unroll
+        for (int kk = 0; kk < KK; ++kk) {
+            bf16x8_t a;
+            if ((lane & 15) == 0)
+                a = *(const bf16x8_t*)&hs[32 * kk + 8 * (lane >> 4)];
+            else
+#pragma unroll
+                for (int e = 0; e < 8; ++e) a[e] = (__bf16)0.0f;
+#pragma unroll
+            for (int i = 0; i < CPW; ++i)
+#pragma unroll
+                for (int g = 0; g < 3; ++g)
+                    acc[i][g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a, wA[i][g][kk], acc[i][g], 0, 0, 0);
+        }
+
+        // gates: batch row 0 lives in C row 0 -> lanes 0..15, element 0
+        __bf16* hw = (__bf16*)hring(1 - (u & 1));
+        if ((lane >> 4) == 0) {
+#pragma unroll
+            for (int i = 0; i < CPW; ++i) {
+                const int ct = wave + NW * i;
+                const int j = ct * 16 + lane;
+                const float gr = acc[i][0][0] + bhh_s[j];
+                const float gz = acc[i][1][0] + bhh_s[Hp + j];
+                const float hn = acc[i][2][0] + bhh_s[2 * Hp + j];
+                const __bf16* gt = (const __bf16*)gi_s + (long)tt * GIP;
+                const float r = sigmoidf((float)gt[j] + gr);
+                const float z = sigmoidf((float)gt[Hp + j] + gz);
+                const float n = fast_tanh((float)gt[2 * Hp + j] + r * hn);
+                const float hnew = (1.0f - z) * n + z * hreg[i];
+                hreg[i] = hnew;
+                const __bf16 hb = (__bf16)__float2bfloat16(hnew);
+                hw[j] = hb;
+                ((__bf16*)out_b)[(long)tt * out_trow + j] = hb;
+            }
+        }
+        __syncthreads();
+    }
+
+    if ((lane >> 4) == 0) {
+        float* hl = hlast + (long)dir * Hp;   // (n_dir, B=1, Hp)
+#pragma unroll
+        for (int i = 0; i < CPW; ++i)
+            hl[(wave + NW * i) * 16 + lane] = hreg[i];
+    }
+}
+
 template <int BT, int Hp, int NT, int WPE>
 __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
                           amdgpu_waves_per_eu(WPE, WPE)))
@@ -2587,6 +2724,23 @@ static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
     return 0;
 }
 
+extern "C" int fmda_gru_fwd_b1_launch(const void* gi, const void* w,
+                                      const float* bhh, void* out,
+                                      float* hlast, int Tseq, int n_dir,
+                                      hipStream_t stream) {
+    constexpr int Hp = 128, NT = 256;
+    const size_t lds = 2 * (size_t)Tseq * (3 * Hp + 8) + 2 * 2 * (Hp + 8) +
+                       4 * 3 * Hp;
+    if (lds > 150 * 1024) return -5;   // sequence too long for LDS residency
+    auto k = gru_fwd_b1_kernel<Hp, NT>;
+    (void)hipFuncSetAttribute((const void*)k,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    k<<<dim3(n_dir), NT, lds, stream>>>(
+        (const __hip_bfloat16*)gi, (const __hip_bfloat16*)w, bhh,
+        (__hip_bfloat16*)out, hlast, Tseq, n_dir);
+    return hipGetLastError() == hipSuccess ? 0 : -4;
+}
+
 extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const float* bhh, void* out,
                                    float* hlast, int B, int Tseq, int n_dir,
@@ -2603,6 +2757,10 @@ extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
             case 32: F(bf16_t, 32, 32, true, 256, false); break;
             case 64: F(bf16_t, 32, 64, true, 256, false); break;
             case 128:
+                if (B == 1 &&
+                    fmda_gru_fwd_b1_launch(gi, w, bhh, out, hlast, Tseq,
+                                           n_dir, stream) == 0)
+                    break;   // LDS-resident batch-1 kernel took it
                 launch_fwd_v3_128(gi, w, bhh, out, hlast, B, Tseq, n_dir,
                                   stream);
                 break;
