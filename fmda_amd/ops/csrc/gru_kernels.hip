@@ -2412,10 +2412,75 @@ __global__ void pool_bwd_kernel(const float* __restrict__ dmax,
     }
 }
 
+// Small-batch variant (the streaming predict path: B=1 would give the
+// thread-per-pair kernel a 2-wave grid looping serially over T): one WAVE
+// per (b, h-pair), lanes strided over T, max/argmax/sum combined with
+// shfl_xor reductions. Argmax tie-break keeps the smallest t — exactly the
+// sequential kernel's first-strictly-greater scan.
+template <typename T>
+__global__ void pool_fwd_small_kernel(const T* __restrict__ out,
+                                      float* __restrict__ maxv,
+                                      float* __restrict__ avgv,
+                                      int* __restrict__ amax, int B, int Tseq,
+                                      int H, int n_dir) {
+    const int HP2 = H / 2;
+    const int wave = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const int idx = blockIdx.x * (blockDim.x >> 6) + wave;
+    if (idx >= B * HP2) return;
+    const int b = idx / HP2;
+    const int h2 = (idx % HP2) * 2;
+    const int HD = n_dir * H;
+    const T* p = out + (long)b * Tseq * HD + h2;
+    float mx0 = -3.4e38f, mx1 = -3.4e38f, s0 = 0.0f, s1 = 0.0f;
+    int im0 = Tseq, im1 = Tseq;
+    for (int t = lane; t < Tseq; t += 64) {
+        const T* pt = p + (long)t * HD;
+        struct alignas(2 * sizeof(T)) Pair { T x, y; };
+        const Pair v2 = *(const Pair*)pt;
+        float v0 = to_f32<T>(v2.x), v1 = to_f32<T>(v2.y);
+        if (n_dir == 2) {
+            const Pair w2 = *(const Pair*)(pt + H);
+            v0 += to_f32<T>(w2.x);
+            v1 += to_f32<T>(w2.y);
+        }
+        s0 += v0; s1 += v1;
+        if (v0 > mx0) { mx0 = v0; im0 = t; }
+        if (v1 > mx1) { mx1 = v1; im1 = t; }
+    }
+#pragma unroll
+    for (int d = 1; d < 64; d <<= 1) {
+        const float omx0 = __shfl_xor(mx0, d), omx1 = __shfl_xor(mx1, d);
+        const int oim0 = __shfl_xor(im0, d), oim1 = __shfl_xor(im1, d);
+        s0 += __shfl_xor(s0, d);
+        s1 += __shfl_xor(s1, d);
+        if (omx0 > mx0 || (omx0 == mx0 && oim0 < im0)) { mx0 = omx0; im0 = oim0; }
+        if (omx1 > mx1 || (omx1 == mx1 && oim1 < im1)) { mx1 = omx1; im1 = oim1; }
+    }
+    if (lane == 0) {
+        const int o = b * H + h2;
+        maxv[o] = mx0; maxv[o + 1] = mx1;
+        avgv[o] = s0 / (float)Tseq; avgv[o + 1] = s1 / (float)Tseq;
+        amax[o] = im0; amax[o + 1] = im1;
+    }
+}
+
 extern "C" int fmda_pool_fwd_launch(int is_bf16, const void* out, float* maxv,
                                     float* avgv, int* amax, int B, int Tseq,
                                     int H, int n_dir, hipStream_t stream) {
     const int n = B * (H / 2);   // thread per h-pair (H is always even)
+    if (n < 4096) {
+        // small grid (streaming predict): wave per pair, lanes over T
+        const dim3 grid((n + 3) / 4);
+        if (is_bf16)
+            pool_fwd_small_kernel<__hip_bfloat16><<<grid, 256, 0, stream>>>(
+                (const __hip_bfloat16*)out, maxv, avgv, amax, B, Tseq, H,
+                n_dir);
+        else
+            pool_fwd_small_kernel<float><<<grid, 256, 0, stream>>>(
+                (const float*)out, maxv, avgv, amax, B, Tseq, H, n_dir);
+        return hipGetLastError() == hipSuccess ? 0 : -1;
+    }
     const dim3 grid((n + 255) / 256);
     if (is_bf16)
         pool_fwd_kernel<__hip_bfloat16><<<grid, 256, 0, stream>>>(

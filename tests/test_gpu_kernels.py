@@ -208,11 +208,13 @@ def test_model_train_step_gpu_bf16():
 
 
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
-def test_dirsum_pool_matches_eager(dtype):
-    """Fused pooling kernel vs the plain torch ops (fwd + bwd)."""
+@pytest.mark.parametrize("B", [1, 33, 700])
+def test_dirsum_pool_matches_eager(dtype, B):
+    """Fused pooling kernel vs the plain torch ops (fwd + bwd). B=1/33 hit
+    the wave-per-pair small-grid variant, B=700 the thread-per-pair one."""
     from fmda_amd.ops.interface import dirsum_pool
     torch.manual_seed(7)
-    B, T, H = 33, 19, 24
+    T, H = 19, 24
     out = (torch.randn(B, T, 2 * H) * 2).to(dtype).cuda().requires_grad_(True)
     mx, av = dirsum_pool(out, 2)
     gm = torch.randn(B, H).to(dtype).cuda()
