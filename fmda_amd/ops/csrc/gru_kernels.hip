@@ -1193,7 +1193,10 @@ void gru_fwd_b1_kernel(const __hip_bfloat16* __restrict__ gi,
                 ((__bf16*)out_b)[(long)tt * out_trow + j] = hb;
             }
         }
-        __syncthreads();
+        // raw barrier + lgkm-only wait: __syncthreads would also drain
+        // vmcnt and stall every step on the out stores' retirement
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
     }
 
     if ((lane >> 4) == 0) {

@@ -95,6 +95,21 @@ def test_gru_fwd_batch1_unidirectional():
     assert (hlast - hlast_ref).abs().max() < 0.05
 
 
+def test_gru_fwd_batch1_long_sequence_fallback():
+    """T past the LDS-residency cap must fall back to the batch-tiled
+    kernel with identical semantics."""
+    ext = _ext()
+    torch.manual_seed(9)
+    H, T, n_dir = 128, 256, 2   # 256*(3*128+8)*2 B > 150 KB -> v3 path
+    gi = (torch.randn(1, T, n_dir * 3 * H) * 0.5).cuda()
+    w = (torch.randn(n_dir, 3 * H, H) * 0.2).cuda()
+    bhh = (torch.randn(n_dir, 3 * H) * 0.1).cuda()
+    out, hlast = ext.gru_fwd(gi.bfloat16(), w.bfloat16(), bhh)
+    out_ref, hlast_ref = _gru_ref_from_gi(gi, w, bhh)
+    assert (out.float() - out_ref).abs().max() < 0.06
+    assert (hlast - hlast_ref).abs().max() < 0.06
+
+
 def _gru_ref_from_gi(gi, w, bhh):
     """Golden recurrence on precomputed input projections (fp32 torch)."""
     B, T, _ = gi.shape
