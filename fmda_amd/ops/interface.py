@@ -304,6 +304,40 @@ class _BiGRULayer(torch.autograd.Function):
 
 
 
+# Bumped by FusedClipAdam.step(): its HIP kernel mutates the masters via
+# raw pointers, which does NOT advance torch's per-tensor _version counter,
+# so the packed-weight cache must also key on this epoch.
+_PACK_EPOCH = 0
+
+
+def _packed_inference_weights(gru_module, layer, params, Hp, H, dtype):
+    """Inference-path cache of the per-layer packed/cast weights.
+
+    Training re-packs every step (Adam mutates the masters), but in
+    eval/no-grad the ~8 cat/cast kernels per layer are pure overhead —
+    at batch=1 streaming they were ~40% of the captured predict graph.
+    Keyed by the params' in-place version counters, so an optimizer step
+    (in-place update) invalidates the cache automatically."""
+    key = f"_fmda_pack_l{layer}_{dtype}"
+    flat = [t for four in params for t in four]
+    versions = (_PACK_EPOCH,) + tuple(t._version for t in flat)
+    ent = getattr(gru_module, key, None)
+    if ent is not None and ent[0] == versions:
+        return ent[1]
+    w_ihs, w_hhs, b_ihs, b_hhs = [], [], [], []
+    for (w_ih, w_hh, b_ih, b_hh) in params:
+        w_ihs.append(_pad_gate_rows(w_ih, H, Hp))
+        w_hhs.append(_pad_cols(_pad_gate_rows(w_hh, H, Hp), H, Hp))
+        b_ihs.append(_pad_gate_rows(b_ih, H, Hp))
+        b_hhs.append(_pad_gate_rows(b_hh, H, Hp))
+    packed = (torch.cat(w_ihs, dim=0).to(dtype),
+              torch.cat(b_ihs, dim=0).to(dtype),
+              torch.stack(w_hhs, dim=0).to(dtype).contiguous(),
+              torch.stack(b_hhs, dim=0).float().contiguous())
+    setattr(gru_module, key, (versions, packed))
+    return packed
+
+
 def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
                 bidirectional: bool, dropout_p: float, training: bool,
                 hidden: Optional[torch.Tensor] = None
@@ -330,7 +364,16 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
                     getattr(gru_module, f"weight_hh_{sfx}"),
                     getattr(gru_module, f"bias_ih_{sfx}"),
                     getattr(gru_module, f"bias_hh_{sfx}"))
-        if D == 2:
+        if not training and not torch.is_grad_enabled():
+            # inference fast path: cached packed weights, direct kernel call
+            ext = load_extension()
+            w_ih_cat, b_ih_cat, w_hh_cat, b_hh_cat = _packed_inference_weights(
+                gru_module, layer, [p[d] for d in range(D)], Hp, H, x.dtype)
+            B, T, F = inp.shape
+            gi = torch.addmm(b_ih_cat, inp.reshape(B * T, F),
+                             w_ih_cat.t()).view(B, T, -1)
+            out_pad, h_last = ext.gru_fwd(gi, w_hh_cat, b_hh_cat)
+        elif D == 2:
             out_pad, h_last = _BiGRULayer.apply(inp, Hp, *p[0], *p[1])
         else:
             out_pad, h_last = _BiGRULayer.apply(inp, Hp, *p[0],

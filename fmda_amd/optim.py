@@ -96,6 +96,10 @@ class FusedClipAdam:
             self._cpu_table[:, 1] = g[self._gidx] + self._goff
             self._chunks.copy_(self._cpu_table, non_blocking=True)
             self._ptrs = ptrs
+        # the HIP kernel writes params through raw pointers: bump the
+        # packed-weight cache epoch (torch _version counters don't move)
+        from .ops import interface as _iface
+        _iface._PACK_EPOCH += 1
         self.last_norm2 = ext.fused_clip_adam(
             self._chunks, self._chunks.shape[0], float(self.clip),
             float(self.lr), float(self.beta1), float(self.beta2),
