@@ -31,16 +31,69 @@ BARS_PER_DAY = 78
 BARS_PER_WEEK = BARS_PER_DAY * 5
 
 
+def sanitize_keys(d: dict) -> dict:
+    """Key sanitizer with the reference's `change_keys` semantics
+    (getMarketData.py:10-35): strips the Alpha-Vantage-style numbered-dot
+    prefixes' separators — spaces, dots and dashes become underscores so
+    keys are valid SQL column / dict identifiers."""
+    out = {}
+    for k, v in d.items():
+        kk = str(k).replace(". ", "_").replace(".", "_")
+        kk = kk.replace(" ", "_").replace("-", "_")
+        out[kk] = sanitize_keys(v) if isinstance(v, dict) else v
+    return out
+
+
+def value_to_number(v):
+    """String->number coercion of the reference `value_to_number`
+    (getMarketData.py:38-58): numeric strings become int/float,
+    K/M/B/% suffixed quantities are scaled, everything else passes
+    through unchanged."""
+    if not isinstance(v, str):
+        return v
+    s = v.strip().replace(",", "")
+    mult = 1.0
+    if s.endswith("%"):
+        s = s[:-1]
+    elif s[-1:] in ("K", "M", "B"):
+        mult = {"K": 1e3, "M": 1e6, "B": 1e9}[s[-1]]
+        s = s[:-1]
+    try:
+        f = float(s) * mult
+        return int(f) if f.is_integer() and "." not in s and mult == 1.0 else f
+    except ValueError:
+        return v
+
+
 class FeedProducers:
     """Publish the five raw topics for each bar of a SyntheticMarket."""
 
     def __init__(self, market, bus: Optional[MessageBus] = None,
-                 freq: float = 300.0, t0: float = 0.0):
+                 freq: float = 300.0, t0: float = 0.0,
+                 registry_path: Optional[str] = None):
         self.market = market
         self.bus = bus or MessageBus()
         self.freq = freq
         self.t0 = t0
         self._col = {name: i for i, name in enumerate(FEATURE_NAMES)}
+        # Indicator dedup registry: the reference pipeline keeps a pickle
+        # of already-published events and only sends new ones
+        # (economic_indicators_spider.py:42-48,94-96; loaded at
+        # producer.py:108-109). Same mechanism, optionally persisted.
+        self.registry_path = registry_path
+        self._ind_registry = {}
+        if registry_path is not None:
+            import os
+            import pickle
+            if os.path.exists(registry_path):
+                with open(registry_path, "rb") as f:
+                    self._ind_registry = pickle.load(f)
+
+    def save_registry(self) -> None:
+        if self.registry_path is not None:
+            import pickle
+            with open(self.registry_path, "wb") as f:
+                pickle.dump(self._ind_registry, f)
 
     def _x(self, i: int, name: str) -> float:
         return float(self.market.X[i, self._col[name]])
@@ -97,11 +150,18 @@ class FeedProducers:
         msg = empty_indicator_message()
         msg["Timestamp"] = ts
         for event in EVENT_LIST_REPL:
-            for value in EVENT_VALUES:
-                v = x(i, f"sd.{event}_{value}")
-                if v != 0.0:
+            vals = tuple(x(i, f"sd.{event}_{v}") for v in EVENT_VALUES)
+            if any(v != 0.0 for v in vals):
+                # dedup: an event row already in the registry is not
+                # re-sent (the registry accumulates every published item,
+                # like the reference items.pickle)
+                seen = self._ind_registry.setdefault(event, set())
+                if vals in seen:
+                    continue
+                seen.add(vals)
+                for value, v in zip(EVENT_VALUES, vals):
                     msg[event][value] = v
-                    fired = True
+                fired = True
         if fired:
             self.bus.publish("ind", msg)
         return ts

@@ -139,14 +139,25 @@ def main():
     ap.add_argument("--checkpoint", default="model_params.pt")
     ap.add_argument("--resume", default=None,
                     help="checkpoint to resume from (model_params.pt format)")
+    ap.add_argument("--log-file", default=None,
+                    help="append per-epoch JSONL records to this file")
     args = ap.parse_args()
     mcfg = ModelConfig(hidden_size=args.hidden, n_layers=args.layers,
                        spatial_dropout=False, dropout=0.5)
     dcfg = DataConfig(n_rows=args.rows, window=args.window)
     tcfg = TrainConfig(batch_size=args.batch, epochs=args.epochs,
                        device=args.device)
+    if args.log_file:
+        fh = open(args.log_file, "a")
+
+        def log(line):
+            print(line)
+            fh.write(line + "\n")
+            fh.flush()
+    else:
+        log = print
     train(mcfg, dcfg, tcfg, checkpoint_path=args.checkpoint,
-          resume=args.resume)
+          resume=args.resume, log=log)
 
 
 if __name__ == "__main__":

@@ -54,3 +54,37 @@ def test_cot_forward_fill_weekly():
     torch.testing.assert_close(X[:, j], market.X[:, j], rtol=1e-6, atol=1e-6)
     # the weekly value actually changes across the week boundary
     assert float(X[0, j]) != float(X[449, j])
+
+
+def test_sanitize_keys_and_value_to_number():
+    from fmda_amd.runtime.producers import sanitize_keys, value_to_number
+    d = sanitize_keys({"1. open": 1, "Fed-Rate Decision": {"a b": 2}})
+    assert d == {"1_open": 1, "Fed_Rate_Decision": {"a_b": 2}}
+    assert value_to_number("1,234") == 1234
+    assert value_to_number("2.5K") == 2500.0
+    assert value_to_number("1.2M") == 1200000.0
+    assert value_to_number("3.1%") == 3.1
+    assert value_to_number("n/a") == "n/a"
+    assert value_to_number(7) == 7
+
+
+def test_indicator_dedup_registry(tmp_path):
+    import pickle
+    from fmda_amd.data.generator import SyntheticMarket
+    from fmda_amd.runtime.bus import MessageBus
+    from fmda_amd.runtime.producers import FeedProducers
+    market = SyntheticMarket(300, seed=11)
+    reg = str(tmp_path / "items.pickle")
+    bus = MessageBus()
+    prod = FeedProducers(market, bus, registry_path=reg)
+    prod.run()
+    prod.save_registry()
+    n_first = bus.topic("ind").end_offset()
+    assert n_first > 0
+    # a second producer with the persisted registry publishes nothing new
+    bus2 = MessageBus()
+    prod2 = FeedProducers(market, bus2, registry_path=reg)
+    prod2.run()
+    assert bus2.topic("ind").end_offset() == 0
+    with open(reg, "rb") as f:
+        assert len(pickle.load(f)) > 0
