@@ -115,10 +115,17 @@ def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
     _, _, test_set = make_epoch_sets(market, dcfg, norm_params_path)
     te = model.evaluate_model(
         chunk_batches(market, test_set, dcfg, tcfg.batch_size))
+    import math
+
     import numpy as np
     has_rows = te[3].numel() > 0  # tiny configs can have an empty test split
-    test_rec = {"test_acc": float(te[0]), "test_hamming": float(te[1]),
-                "test_fbeta": [float(v) for v in np.atleast_1d(te[2])],
+
+    def _num(v):
+        v = float(v)
+        return None if math.isnan(v) else v   # keep the JSONL standard-valid
+
+    test_rec = {"test_acc": _num(te[0]), "test_hamming": _num(te[1]),
+                "test_fbeta": [_num(v) for v in np.atleast_1d(te[2])],
                 "test_acc3": (float(three_class_accuracy(te[4], te[3]))
                               if has_rows else None),
                 "confusion": (multilabel_confusion(te[4], te[3]).tolist()
