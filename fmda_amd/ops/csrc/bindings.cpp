@@ -26,7 +26,10 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* out, const void* dout,
                                    const float* dhT, void* dgi, void* dgh,
                                    float* dh0, float* dbhh, int B, int Tseq,
-                                   int n_dir, hipStream_t stream);
+                                   int n_dir, unsigned int drop_thr,
+                                   float drop_scale,
+                                   unsigned long long drop_seed,
+                                   hipStream_t stream);
 extern "C" int fmda_mfma_selftest_launch(const void* A, const void* Bm,
                                          float* C, hipStream_t stream);
 extern "C" int fmda_pool_fwd_launch(int is_bf16, const void* out, float* maxv,
@@ -125,7 +128,8 @@ std::vector<torch::Tensor> gru_fwd(torch::Tensor gi, torch::Tensor w,
 
 std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
                                    torch::Tensor bhh, torch::Tensor out,
-                                   torch::Tensor dout, torch::Tensor dhT) {
+                                   torch::Tensor dout, torch::Tensor dhT,
+                                   double drop_p, int64_t drop_seed) {
     int B, T, n_dir, Hp;
     bool is_bf16;
     check_common(gi, w, bhh, B, T, n_dir, Hp, is_bf16);
@@ -169,13 +173,24 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
             dbsum.data_ptr<float>(), gpub.data_ptr(),
             (unsigned int*)cnt.data_ptr(), B, T, n_dir, stream.stream());
     } else {
+        // fused inter-layer dropout backward (d_out is w.r.t. the DROPPED
+        // activations; the kernel recomputes the counter-based mask at the
+        // dout read instead of a separate full-tensor pass)
+        const unsigned int drop_thr =
+            (drop_p > 0.0) ? (unsigned int)(drop_p * 256.0) : 0u;
+        const float drop_scale =
+            (drop_p > 0.0) ? (float)(1.0 / (1.0 - drop_p)) : 1.0f;
+        TORCH_CHECK(drop_thr == 0u || (is_bf16 && Hp == 128),
+                    "fused dropout-backward requires bf16 Hp=128");
         rc = fmda_gru_bwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
                                  w.data_ptr(), wt_ptr, bhh.data_ptr<float>(),
                                  out.data_ptr(), dout.data_ptr(),
                                  dhT.data_ptr<float>(), dgi.data_ptr(),
                                  dgh.data_ptr(), dh0.data_ptr<float>(),
                                  dbsum.data_ptr<float>(), B, T,
-                                 n_dir, stream.stream());
+                                 n_dir, drop_thr, drop_scale,
+                                 (unsigned long long)drop_seed,
+                                 stream.stream());
     }
     TORCH_CHECK(rc == 0, "fmda gru_bwd launch failed rc=", rc, " Hp=", Hp);
     using torch::indexing::Slice;
@@ -310,7 +325,10 @@ torch::Tensor dropout_fused(torch::Tensor x, double p, int64_t seed) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gru_fwd", &gru_fwd, "fused biGRU recurrence forward (HIP/CDNA4)");
-    m.def("gru_bwd", &gru_bwd, "fused biGRU recurrence backward (HIP/CDNA4)");
+    m.def("gru_bwd", &gru_bwd, "fused biGRU recurrence backward (HIP/CDNA4)",
+          py::arg("gi"), py::arg("w"), py::arg("bhh"), py::arg("out"),
+          py::arg("dout"), py::arg("dhT"), py::arg("drop_p") = 0.0,
+          py::arg("drop_seed") = 0);
     m.def("mfma_selftest", &mfma_selftest, "mfma fragment layout self-test");
     m.def("dropout_fused", &dropout_fused,
           "counter-based dropout (mask recomputed in backward)");

@@ -220,6 +220,13 @@ struct TilePrefetch {
 // fused gate phase is identical for both dtypes.
 // ---------------------------------------------------------------------------
 
+FMDA_DEV unsigned long long mix64(unsigned long long x) {
+    x += 0x9E3779B97F4A7C15ull;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+    return x ^ (x >> 31);
+}
+
 // B-fragment of W for the gh GEMM (N = gate output column, K = h index).
 template <int Hp>
 FMDA_DEV bf16x8_t load_wfragA(const __hip_bfloat16* __restrict__ w, long pitch,
@@ -1219,7 +1226,9 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                        const float* __restrict__ dhT,
                        __hip_bfloat16* __restrict__ dgi,
                        __hip_bfloat16* __restrict__ dgh, float* __restrict__ dh0,
-                       float* __restrict__ dbhh, int B, int Tseq, int n_dir) {
+                       float* __restrict__ dbhh, int B, int Tseq, int n_dir,
+                       unsigned int drop_thr, float drop_scale,
+                       unsigned long long drop_seed) {
     constexpr int NW = NT / 64;
     constexpr int MT = BT / 16;
     constexpr int NCT = Hp / 16;
@@ -1477,9 +1486,23 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                     const float z = sigmoidf(iz + gz);
                     const float n = fast_tanh(in_ + r * hn);
                     const float hprev = hb_read(hbq, b, j);
-                    const float dht =
-                        dhreg[i][m][e] + (live ? (float)dorow[b * Hp + j]
-                                               : 0.0f);
+                    float doval = live ? (float)dorow[b * Hp + j] : 0.0f;
+                    if (drop_thr != 0u) {
+                        // fused inter-layer dropout backward: dout is the
+                        // grad w.r.t. the DROPPED activations; recompute
+                        // the counter-based mask (same splitmix64 draw as
+                        // dropout_kernel) instead of a separate 1 GB
+                        // read+write pass over d_out.
+                        const long o = ((long)(b0 + b) * Tseq + tt) *
+                                           n_dir * Hp +
+                                       (long)dir * Hp + j;
+                        const unsigned long long rnd =
+                            mix64(drop_seed ^ (unsigned long long)(o >> 3));
+                        const unsigned int u =
+                            (unsigned int)(rnd >> (8 * ((int)o & 7))) & 0xFF;
+                        doval = (u < drop_thr) ? 0.0f : doval * drop_scale;
+                    }
+                    const float dht = dhreg[i][m][e] + doval;
                     // live-gating every product: 0 * inf = NaN would leak
                     // into dbacc from uninitialized dead-row LDS.
                     const float dz_pre =
@@ -2296,12 +2319,6 @@ void gru_bwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
 // pair (no mask tensor exists at all). Keep/scale semantics match
 // torch.nn.functional.dropout (scale 1/(1-p) on kept elements).
 // ===========================================================================
-FMDA_DEV unsigned long long mix64(unsigned long long x) {
-    x += 0x9E3779B97F4A7C15ull;
-    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
-    x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
-    return x ^ (x >> 31);
-}
 
 template <typename T>
 __global__ void dropout_kernel(const T* __restrict__ x, T* __restrict__ y,
@@ -2774,7 +2791,9 @@ static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
                              const void* out, const void* dout,
                              const float* dhT, void* dgi, void* dgh,
                              float* dh0, float* dbhh, int B, int Tseq,
-                             int n_dir, hipStream_t stream) {
+                             int n_dir, unsigned int drop_thr,
+                             float drop_scale, unsigned long long drop_seed,
+                             hipStream_t stream) {
     constexpr int BT = 32, Hp = 128, NT = 512;
     const size_t lds = 3 * 2 * BT * 3 * Hp + 2 * 2 * BT * Hp +
                        2 * 2 * BT * Hp + 2 * 2 * BT * (3 * Hp + 8) +
@@ -2788,7 +2807,7 @@ static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
         (const __hip_bfloat16*)wt, bhh,
         (const __hip_bfloat16*)out, (const __hip_bfloat16*)dout, dhT,
         (__hip_bfloat16*)dgi, (__hip_bfloat16*)dgh, dh0, dbhh, B, Tseq,
-        n_dir);
+        n_dir, drop_thr, drop_scale, drop_seed);
     return 0;
 }
 
@@ -2881,8 +2900,13 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* out, const void* dout,
                                    const float* dhT, void* dgi, void* dgh,
                                    float* dh0, float* dbhh, int B, int Tseq,
-                                   int n_dir, hipStream_t stream) {
+                                   int n_dir, unsigned int drop_thr,
+                                   float drop_scale,
+                                   unsigned long long drop_seed,
+                                   hipStream_t stream) {
     if (!is_bf16 && Hp > 256) return -3;
+    if (drop_thr != 0u && !(is_bf16 && Hp == 128))
+        return -6;   // fused dropout-backward only on the bf16 Hp=128 path
     const LaunchCfg c = bwd_cfg(is_bf16, Hp);
     const size_t lds = bwd_lds_bytes(is_bf16, Hp, c.bt, c.wlds);
     if (lds > 160 * 1024) return -2;
@@ -2897,7 +2921,8 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
             case 64: G(bf16_t, 32, 64, true, 256, false); break;
             case 128:
                 launch_bwd_v3_128(gi, w, wt, bhh, out, dout, dhT, dgi, dgh,
-                                  dh0, dbhh, B, Tseq, n_dir, stream);
+                                  dh0, dbhh, B, Tseq, n_dir, drop_thr,
+                                  drop_scale, drop_seed, stream);
                 break;
             case 256: G(bf16_t, 32, 256, false, 512, false); break;
             case 512: G(bf16_t, 16, 512, false, 512, false); break;

@@ -144,6 +144,23 @@ class _FusedDropout(torch.autograd.Function):
         return ext.dropout_fused(dy.contiguous(), ctx.p, ctx.seed), None, None
 
 
+class _DeferredDropout(torch.autograd.Function):
+    """Inter-layer dropout whose BACKWARD is fused into the producing
+    layer's BPTT kernel: forward applies the counter-based mask as usual,
+    backward passes the gradient through untouched — the (p, seed) pair is
+    handed to the upstream _BiGRULayer, whose gru_bwd applies the same
+    mask at its d_out read (saving a full-tensor read+write pass)."""
+
+    @staticmethod
+    def forward(ctx, x, p, seed):
+        ext = load_extension()
+        return ext.dropout_fused(x.contiguous(), p, seed)
+
+    @staticmethod
+    def backward(ctx, dy):
+        return dy, None, None
+
+
 def fused_dropout(x: torch.Tensor, p: float) -> torch.Tensor:
     """Training-mode dropout on the HIP engine (bf16 CUDA tensors);
     falls back to torch for other dtypes/devices."""
@@ -239,7 +256,8 @@ class _BiGRULayer(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, Hp, w_ih0, w_hh0, b_ih0, b_hh0,
-                w_ih1, w_hh1, b_ih1, b_hh1):
+                w_ih1, w_hh1, b_ih1, b_hh1, out_drop_p=0.0,
+                out_drop_seed=0):
         ext = load_extension()
         D = 2 if w_ih1 is not None else 1
         H = w_hh0.shape[1]
@@ -263,19 +281,20 @@ class _BiGRULayer(torch.autograd.Function):
         gi = torch.addmm(b_ih_cat, x2d, w_ih_cat.t()).view(B, T, -1)
         out, h_last = ext.gru_fwd(gi, w_hh_cat, b_hh_cat)
         ctx.save_for_backward(x2d, w_ih_cat, w_hh_cat, b_hh_cat, gi, out)
-        ctx.meta = (D, H, Hp)
+        ctx.meta = (D, H, Hp, out_drop_p, out_drop_seed)
         return out, h_last
 
     @staticmethod
     def backward(ctx, d_out, d_hlast):
         ext = load_extension()
         x2d, w_ih_cat, w_hh_cat, b_hh_cat, gi, out = ctx.saved_tensors
-        D, H, Hp = ctx.meta
+        D, H, Hp, drop_p, drop_seed = ctx.meta
         need_dx = ctx.needs_input_grad[0]
         d_out = d_out.contiguous().to(gi.dtype)
         d_hlast = d_hlast.contiguous().float()
         dgi, dgh, _dh0, dbhh, dbih = ext.gru_bwd(
-            gi, w_hh_cat, b_hh_cat, out, d_out, d_hlast)
+            gi, w_hh_cat, b_hh_cat, out, d_out, d_hlast,
+            drop_p, drop_seed)
         M = dgi.shape[0] * dgi.shape[1]
 
         # dW_hh via the time-shifted dGh and one split-K reduction
@@ -300,7 +319,7 @@ class _BiGRULayer(torch.autograd.Function):
         if D == 1:
             grads.append((None, None, None, None))
         (a0, b0, c0, e0), (a1, b1, c1, e1) = grads
-        return dx, None, a0, b0, c0, e0, a1, b1, c1, e1
+        return (dx, None, a0, b0, c0, e0, a1, b1, c1, e1, None, None)
 
 
 
@@ -364,6 +383,16 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
                     getattr(gru_module, f"weight_hh_{sfx}"),
                     getattr(gru_module, f"bias_ih_{sfx}"),
                     getattr(gru_module, f"bias_hh_{sfx}"))
+        # inter-layer dropout planned for this layer's output? choose the
+        # seed NOW so the producing layer's BPTT kernel can apply the same
+        # counter-based mask at its d_out read (backward pass fused away)
+        drop_here = (training and dropout_p > 0 and layer < n_layers - 1)
+        defer = (drop_here and Hp == H == 128
+                 and x.dtype == torch.bfloat16 and x.is_cuda)
+        seed = (int(torch.empty((), dtype=torch.int64).random_())
+                if defer else 0)
+        dp = (dropout_p, seed) if defer else (0.0, 0)
+
         if not training and not torch.is_grad_enabled():
             # inference fast path: cached packed weights, direct kernel call
             ext = load_extension()
@@ -374,10 +403,10 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
                              w_ih_cat.t()).view(B, T, -1)
             out_pad, h_last = ext.gru_fwd(gi, w_hh_cat, b_hh_cat)
         elif D == 2:
-            out_pad, h_last = _BiGRULayer.apply(inp, Hp, *p[0], *p[1])
+            out_pad, h_last = _BiGRULayer.apply(inp, Hp, *p[0], *p[1], *dp)
         else:
             out_pad, h_last = _BiGRULayer.apply(inp, Hp, *p[0],
-                                                None, None, None, None)
+                                                None, None, None, None, *dp)
 
         if Hp == H:
             out = out_pad
@@ -389,8 +418,11 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
         h_n_parts.append(h_last[:, :, :H].to(x.dtype))
 
         inp = out
-        if training and dropout_p > 0 and layer < n_layers - 1:
-            inp = fused_dropout(inp, dropout_p)
+        if drop_here:
+            if defer:
+                inp = _DeferredDropout.apply(inp, dropout_p, seed)
+            else:
+                inp = fused_dropout(inp, dropout_p)
 
     h_n = torch.cat(h_n_parts, dim=0)  # (L*D, B, H)
     return inp, h_n

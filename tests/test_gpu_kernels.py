@@ -405,6 +405,46 @@ def test_unidirectional_model_on_gpu():
     assert torch.allclose(got, ref, atol=5e-3), (got - ref).abs().max()
 
 
+def test_deferred_dropout_backward_matches_explicit():
+    """The dropout mask applied inside gru_bwd's d_out read (deferred
+    path) must produce the same layer gradients as the explicit
+    _FusedDropout backward pass (same seed, same counter-based mask)."""
+    from fmda_amd.ops.interface import (_BiGRULayer, _DeferredDropout,
+                                        _FusedDropout)
+    torch.manual_seed(3)
+    B, T, F, H = 48, 24, 64, 128
+    p_drop, seed = 0.3, 987654321
+    masters = []
+    for _ in range(2):
+        masters += [torch.randn(3 * H, F if _ == 0 or True else F,
+                                device="cuda") * 0.2,
+                    torch.randn(3 * H, H, device="cuda") * 0.2,
+                    torch.randn(3 * H, device="cuda") * 0.1,
+                    torch.randn(3 * H, device="cuda") * 0.1]
+    x = (torch.randn(B, T, F, device="cuda") * 0.5).bfloat16()
+    gup = torch.randn(B, T, 2 * H, device="cuda").bfloat16()
+    ghl = torch.randn(2, B, H, device="cuda")
+
+    def run(deferred):
+        ms = [m.detach().clone().requires_grad_(True) for m in masters]
+        if deferred:
+            out, hl = _BiGRULayer.apply(x, H, *ms[:4], *ms[4:], p_drop, seed)
+            dropped = _DeferredDropout.apply(out, p_drop, seed)
+        else:
+            out, hl = _BiGRULayer.apply(x, H, *ms[:4], *ms[4:], 0.0, 0)
+            dropped = _FusedDropout.apply(out, p_drop, seed)
+        ((dropped.float() * gup.float()).sum()
+         + (hl * ghl).sum()).backward()
+        return [m.grad.clone() for m in ms], dropped.detach()
+
+    ga, da = run(True)
+    gb, db = run(False)
+    assert torch.equal(da, db)  # forward mask identical by construction
+    for i, (a, b) in enumerate(zip(ga, gb)):
+        rel = (a - b).norm() / b.norm().clamp(min=1e-12)
+        assert rel < 2e-2, (i, float(rel))
+
+
 def test_fused_dropout_statistics_and_backward():
     """Counter-based dropout: keep-rate ~ 1-p, kept elements scaled by
     1/(1-p), and backward applies the IDENTICAL mask (recomputed)."""
