@@ -8,8 +8,8 @@ moving-average periods, Bollinger / stochastic settings, and the
 API tokens, Kafka brokers and MySQL credentials have no equivalent here:
 transport is an in-process queue and storage is an in-memory feature table.
 """
-from dataclasses import dataclass, field
-from typing import List, Optional
+from dataclasses import dataclass
+
 
 
 # Number of order book price levels to include (reference config.py:36-37).

@@ -26,7 +26,7 @@ All of it is vectorized torch on CPU; generation is one-time per dataset
 (the training hot path never touches it).
 """
 import math
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 

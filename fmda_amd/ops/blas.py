@@ -15,7 +15,7 @@ gpurun_out/gemm_ab_*.json):
    150-250 us. `chunked_outer` implements that formulation.
 """
 import os
-from typing import Optional
+
 
 import torch
 

@@ -24,7 +24,7 @@ from typing import Optional
 
 from ..config import (ASK_LEVELS, BID_LEVELS, EVENT_LIST_REPL, EVENT_VALUES,
                       empty_indicator_message)
-from ..features import FEATURE_NAMES, index_of
+from ..features import FEATURE_NAMES
 from .bus import MessageBus
 
 BARS_PER_DAY = 78

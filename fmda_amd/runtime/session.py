@@ -12,10 +12,9 @@ Time is virtual by default (no sleeping); `pace` > 0 plays bars back in
 real time for demos.
 """
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
-import torch
 
 from ..data.generator import SyntheticMarket
 from .bus import MessageBus

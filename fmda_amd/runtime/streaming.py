@@ -20,7 +20,7 @@ from typing import Dict, List, Optional
 
 import torch
 
-from ..features import FEATURE_NAMES, TARGET_NAMES
+from ..features import TARGET_NAMES
 from ..models.bigru import BiGRU
 from .bus import MessageBus
 
