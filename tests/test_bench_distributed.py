@@ -1,0 +1,28 @@
+"""The driver launches bench.py via torch.distributed.run for N>1; this
+runs the REAL entrypoint with world_size=2 over gloo on CPU so rendezvous,
+env parsing, GradAllReduce wiring and the JSON contract are exercised
+end-to-end before any multi-GPU run."""
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_under_torchrun_world2_cpu():
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29571", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--config", "cpu", "--steps", "2", "--warmup", "1"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=420)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout  # exactly one JSON line from rank 0
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2 and rec["config"]["parallelism"] == "dp2"
+    assert rec["value"] > 0 and rec["scaling"] == "weak"
+    assert rec["config"]["global_batch"] == 2 * 8  # per-rank CPU batch cap
