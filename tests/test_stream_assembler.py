@@ -88,3 +88,67 @@ def test_indicator_dedup_registry(tmp_path):
     assert bus2.topic("ind").end_offset() == 0
     with open(reg, "rb") as f:
         assert len(pickle.load(f)) > 0
+
+
+def test_out_of_order_and_late_messages():
+    """The reference handles late data with Spark watermarks
+    (spark_consumer.py per-stream withWatermark); the in-process assembler
+    must tolerate out-of-order bars and late joins keyed by the 5-minute
+    floor."""
+    from fmda_amd.data.generator import SyntheticMarket
+    from fmda_amd.runtime.assembler import StreamAssembler
+    from fmda_amd.runtime.bus import MessageBus
+    from fmda_amd.runtime.producers import FeedProducers
+
+    market = SyntheticMarket(60, seed=21)
+    bus = MessageBus()
+    asm = StreamAssembler(bus, emit_signal=False)
+    prod = FeedProducers(market, bus)
+    # publish bars out of order: evens first, then odds (vix/cot of a bar
+    # arrive with their own bar's deep, so forward-fill sees jumps)
+    for i in range(0, 60, 2):
+        prod.publish_bar(i)
+    for i in range(1, 60, 2):
+        prod.publish_bar(i)
+    X = asm.table()
+    assert X.shape == market.X.shape
+    col = {n: j for j, n in enumerate(FEATURE_NAMES)}
+    # bar-keyed (non-forward-filled) columns must still match exactly
+    for name in ("sd.4_close", "sd.5_volume", "sd.bid_0_size",
+                 "sd.session_start"):
+        assert torch.equal(X[:, col[name]], market.X[:, col[name]]), name
+    # windowed features are computed on the sorted table -> exact too
+    torch.testing.assert_close(X[:, col["p.price_MA20"]],
+                               market.X[:, col["p.price_MA20"]],
+                               rtol=1e-5, atol=1e-5)
+
+
+def test_assembler_tolerates_missing_optional_topics():
+    """Rows assemble even when vix/cot/ind never arrive (zeros like the
+    reference table's IFNULL(...,0) fetch)."""
+    from fmda_amd.data.generator import SyntheticMarket
+    from fmda_amd.runtime.assembler import StreamAssembler
+    from fmda_amd.runtime.bus import MessageBus
+    from fmda_amd.runtime.producers import FeedProducers
+
+    market = SyntheticMarket(40, seed=22)
+    bus = MessageBus()
+    asm = StreamAssembler(bus, emit_signal=False)
+    prod = FeedProducers(market, bus)
+    for i in range(40):
+        ts = prod.t0 + i * prod.freq
+        # deep + volume only
+        full_bus = MessageBus()
+        p2 = FeedProducers(market, full_bus)
+        p2.publish_bar(i)
+        deep = full_bus.topic("deep")._buf[0]
+        vol = full_bus.topic("volume")._buf[0]
+        bus.publish("deep", deep)
+        bus.publish("volume", vol)
+    X = asm.table()
+    assert X.shape[0] == 40
+    col = {n: j for j, n in enumerate(FEATURE_NAMES)}
+    assert torch.all(X[:, col["sd.VIX"]] == 0)
+    assert torch.all(X[:, col["sd.Asset_long_pos"]] == 0)
+    assert torch.equal(X[:, col["sd.4_close"]],
+                       market.X[:, col["sd.4_close"]])
