@@ -36,6 +36,20 @@ def main():
     ck = os.path.join(args.outdir, "model_params.pt")
     np_path = os.path.join(args.outdir, "norm_params")
 
+    print("== raw feed assembly ==")
+    # the reference's producer -> Kafka -> Spark-join chain, in-process:
+    # per-topic raw messages rebuilt into the joined 108-feature table
+    from fmda_amd.data import SyntheticMarket as _SM
+    from fmda_amd.runtime import FeedProducers, StreamAssembler
+    feed_market = _SM(300, seed=13)
+    feed_bus = MessageBus()
+    assembler = StreamAssembler(feed_bus, emit_signal=False)
+    FeedProducers(feed_market, feed_bus).run()
+    X_assembled = assembler.table()
+    dev = (X_assembled - feed_market.X).abs().max()
+    print(f"assembled {tuple(X_assembled.shape)} from raw topics; "
+          f"max deviation vs direct table: {float(dev):.2e}")
+
     print("== training ==")
     mcfg = ModelConfig(hidden_size=16, spatial_dropout=False, dropout=0.3)
     dcfg = DataConfig(n_rows=args.rows, chunk_size=100, window=20)
