@@ -11,6 +11,7 @@ Run: python -m fmda_amd.train [--epochs 25] [--device cuda]
 """
 import argparse
 import json
+import os
 import time
 from typing import Optional
 
@@ -85,9 +86,26 @@ def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
                if not k.startswith("loss_fn.")]
         assert not bad, f"resume key mismatch: {bad}" 
 
+    start_epoch = 1
+    if resume:
+        # full training-state sidecar (optimizer moments, epoch, RNG): a
+        # TRUE mid-training resume — the reference only ever saves weights
+        # (notebook cell 39) and reloads them in predict.py:104
+        ts_path = resume + ".train_state.pt"
+        if os.path.exists(ts_path):
+            ts = torch.load(ts_path, map_location=device,
+                            weights_only=False)
+            # the sidecar's weights are the LAST epoch's (the main
+            # checkpoint holds the BEST-val epoch — the wrong pairing for
+            # the last-epoch optimizer moments)
+            model.load_state_dict(ts["model"], strict=False)
+            model.optimizer.load_state_dict(ts["optimizer"])
+            start_epoch = ts["epoch"] + 1
+            torch.set_rng_state(ts["rng"])
+
     history = []
     best_val_acc = -1.0
-    for epoch in range(1, tcfg.epochs + 1):
+    for epoch in range(start_epoch, tcfg.epochs + 1):
         t0 = time.time()
         train_set, val_set, _ = make_epoch_sets(market, dcfg,
                                                 norm_params_path)
@@ -108,6 +126,15 @@ def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
         if rec["val_acc"] >= best_val_acc:
             best_val_acc = rec["val_acc"]
             save_checkpoint(model, checkpoint_path)
+        # training-state sidecar for exact resume (always current epoch;
+        # includes the CURRENT weights — the main checkpoint may hold an
+        # earlier best-val epoch)
+        sd = {k: v for k, v in model.state_dict().items()
+              if not k.startswith("loss_fn.")}
+        torch.save({"model": sd,
+                    "optimizer": model.optimizer.state_dict(),
+                    "epoch": epoch, "rng": torch.get_rng_state()},
+                   checkpoint_path + ".train_state.pt")
 
     # Held-out test evaluation with per-class confusion matrices
     # (reference notebook cells 33-37).

@@ -69,3 +69,39 @@ def test_train_resume(tmp_path):
     sd = torch.load(ck, weights_only=True)
     model3 = type(model2)(8, 16, 4, n_layers=1, spatial_dropout=False)
     model3.load_state_dict(sd)
+
+
+def test_exact_midtraining_resume(tmp_path):
+    """train(4 epochs) == train(2) + resume(2 more): the sidecar restores
+    optimizer moments, epoch counter and RNG state, so the continued run
+    reproduces the straight run's records exactly."""
+    import json
+
+    from fmda_amd.config import DataConfig, ModelConfig, TrainConfig
+    from fmda_amd.train import train
+
+    mcfg = ModelConfig(hidden_size=8, n_layers=1, spatial_dropout=False,
+                       dropout=0.3)
+    dcfg = DataConfig(n_rows=400, chunk_size=80, window=10)
+    ck_a = str(tmp_path / "a.pt")
+    ck_b = str(tmp_path / "b.pt")
+
+    recs_a = []
+    train(mcfg, dcfg, TrainConfig(batch_size=8, epochs=4),
+          checkpoint_path=ck_a, norm_params_path=str(tmp_path / "na"),
+          log=lambda l: recs_a.append(json.loads(l)))
+
+    recs_b = []
+    train(mcfg, dcfg, TrainConfig(batch_size=8, epochs=2),
+          checkpoint_path=ck_b, norm_params_path=str(tmp_path / "nb"),
+          log=lambda l: recs_b.append(json.loads(l)))
+    train(mcfg, dcfg, TrainConfig(batch_size=8, epochs=4),
+          checkpoint_path=ck_b, norm_params_path=str(tmp_path / "nb"),
+          resume=ck_b, log=lambda l: recs_b.append(json.loads(l)))
+
+    ep_a = [r for r in recs_a if "epoch" in r]
+    ep_b = [r for r in recs_b if "epoch" in r]
+    assert [r["epoch"] for r in ep_b] == [1, 2, 3, 4]
+    for ra, rb in zip(ep_a[2:], ep_b[2:]):   # epochs 3 and 4
+        assert abs(ra["train_loss"] - rb["train_loss"]) < 1e-6, (ra, rb)
+        assert ra["train_acc"] == rb["train_acc"]
