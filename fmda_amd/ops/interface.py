@@ -337,7 +337,7 @@ def _packed_inference_weights(gru_module, layer, params, Hp, H, dtype):
     at batch=1 streaming they were ~40% of the captured predict graph.
     Keyed by the params' in-place version counters, so an optimizer step
     (in-place update) invalidates the cache automatically."""
-    key = f"_fmda_pack_l{layer}_{dtype}"
+    key = f"_fmda_pack_l{layer}_{dtype}_{params[0][0].device}"
     flat = [t for four in params for t in four]
     versions = (_PACK_EPOCH,) + tuple(t._version for t in flat)
     ent = getattr(gru_module, key, None)
