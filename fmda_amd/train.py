@@ -151,8 +151,14 @@ def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
         v = float(v)
         return None if math.isnan(v) else v   # keep the JSONL standard-valid
 
+    from .metrics import fbeta_per_class
+    # the notebook's held-out test metrics use beta=1 (cells 33-36), while
+    # the epoch loops use beta=0.5 (biGRU_model.py:221,279)
+    fb1 = (fbeta_per_class(te[4], te[3], beta=1.0).tolist()
+           if has_rows else [None])
     test_rec = {"test_acc": _num(te[0]), "test_hamming": _num(te[1]),
-                "test_fbeta": [_num(v) for v in np.atleast_1d(te[2])],
+                "test_fbeta": fb1,
+                "test_fbeta_beta05": [_num(v) for v in np.atleast_1d(te[2])],
                 "test_acc3": (float(three_class_accuracy(te[4], te[3]))
                               if has_rows else None),
                 "confusion": (multilabel_confusion(te[4], te[3]).tolist()
