@@ -95,6 +95,12 @@ def main():
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    # --gpus documents the intended N; the actual world size comes from the
+    # torchrun env. Catch a mismatched launch early instead of reporting a
+    # wrong aggregate.
+    if world > 1 and args.gpus not in (1, world):
+        raise SystemExit(f"--gpus {args.gpus} != WORLD_SIZE {world}: launch "
+                         "with torchrun --nproc-per-node matching --gpus")
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     use_cuda = torch.cuda.is_available()
