@@ -66,3 +66,34 @@ def test_mixed_dtypes_roundtrip(tmp_path):
     for k in sd:
         assert torch.equal(sd[k], got[k]), k
         assert torch.equal(sd[k], sd2[k]), k
+
+
+def test_native_reader_rejects_malformed_files(tmp_path):
+    """The C++ reader must raise clean RuntimeErrors (never crash) on
+    malformed input: garbage, truncated zip containers, truncated legacy
+    pickles."""
+    import pytest
+    import torch
+
+    from fmda_amd.ops import load_extension
+    ext = load_extension()
+    if ext is None:
+        pytest.skip("extension not built")
+
+    cases = [b"", b"\x00\x01junk" * 16,
+             b"PK\x03\x04" + b"\x00" * 20,
+             (0x1950a86a20f9469c).to_bytes(8, "little")]
+    sd = {"gru.weight_ih_l0": torch.randn(6, 4)}
+    zp = tmp_path / "z.pt"
+    torch.save(sd, str(zp))
+    cases.append(zp.read_bytes()[: zp.stat().st_size // 2])
+    lp = tmp_path / "l.pt"
+    torch.save(sd, str(lp), _use_new_zipfile_serialization=False)
+    cases.append(lp.read_bytes()[: int(lp.stat().st_size * 0.4)])
+    cases.append(lp.read_bytes()[: int(lp.stat().st_size * 0.9)])
+
+    for i, data in enumerate(cases):
+        p = tmp_path / f"bad_{i}"
+        p.write_bytes(data)
+        with pytest.raises(RuntimeError):
+            ext.load_state_dict_native(str(p))
