@@ -97,3 +97,50 @@ def test_native_reader_rejects_malformed_files(tmp_path):
         p.write_bytes(data)
         with pytest.raises(RuntimeError):
             ext.load_state_dict_native(str(p))
+
+
+def test_native_roundtrip_property(tmp_path):
+    """Property test: random state dicts (shapes, dtypes, key names) round
+    trip through BOTH directions — native writer -> torch.load and
+    torch.save -> native reader — bit-exactly."""
+    import torch
+    from hypothesis import given, settings, strategies as st
+
+    from fmda_amd.ops import load_extension
+    ext = load_extension()
+
+    dtypes = [torch.float32, torch.float64, torch.int64, torch.int32,
+              torch.float16, torch.bfloat16, torch.bool]
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.lists(
+        st.tuples(st.text(alphabet="abc_.019", min_size=1, max_size=12),
+                  st.lists(st.integers(0, 7), min_size=0, max_size=3),
+                  st.integers(0, len(dtypes) - 1)),
+        min_size=1, max_size=5, unique_by=lambda t: t[0]))
+    def check(spec):
+        g = torch.Generator().manual_seed(7)
+        sd = {}
+        for name, shape, di in spec:
+            dt = dtypes[di]
+            if dt.is_floating_point:
+                t = torch.randn(shape, generator=g).to(dt)
+            elif dt == torch.bool:
+                t = torch.rand(shape, generator=g) < 0.5
+            else:
+                t = (torch.rand(shape, generator=g) * 100).to(dt)
+            sd[name] = t
+        p1 = str(tmp_path / "n.pt")
+        ext.save_state_dict_native(p1, list(sd.keys()), list(sd.values()))
+        back = dict(torch.load(p1, weights_only=True))
+        assert set(back) == set(sd)
+        for k in sd:
+            assert back[k].dtype == sd[k].dtype
+            assert torch.equal(back[k], sd[k]), k
+        p2 = str(tmp_path / "t.pt")
+        torch.save(sd, p2)
+        back2 = dict(ext.load_state_dict_native(p2))
+        for k in sd:
+            assert torch.equal(back2[k], sd[k]), k
+
+    check()
