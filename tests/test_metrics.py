@@ -61,3 +61,38 @@ def test_multilabel_confusion_matches_sklearn():
     ours = multilabel_confusion(t, p).numpy()
     ref = multilabel_confusion_matrix(t.numpy(), p.numpy())
     assert (ours == ref).all()
+
+
+def test_metrics_match_sklearn_property():
+    """Property test vs sklearn on random multilabel matrices, including
+    degenerate columns (all-positive / all-negative)."""
+    import torch
+    from hypothesis import given, settings, strategies as st
+    from sklearn.metrics import (accuracy_score, fbeta_score, hamming_loss,
+                                 multilabel_confusion_matrix)
+
+    from fmda_amd.metrics import (fbeta_per_class, hamming,
+                                  multilabel_confusion, subset_accuracy)
+
+    @settings(max_examples=30, deadline=None)
+    # c >= 2: sklearn reinterprets an (n, 1) indicator matrix as BINARY
+    # labels (one 2-class problem), not one-column multilabel — different
+    # semantics than the per-column formulas (the reference always has 4)
+    @given(st.integers(1, 40), st.integers(2, 6), st.integers(0, 2 ** 31 - 1),
+           st.sampled_from([0.5, 1.0, 2.0]))
+    def check(n, c, seed, beta):
+        g = torch.Generator().manual_seed(seed)
+        t = (torch.rand(n, c, generator=g) < 0.4).long()
+        p = (torch.rand(n, c, generator=g) < 0.4).long()
+        assert abs(float(subset_accuracy(t, p))
+                   - accuracy_score(t.numpy(), p.numpy())) < 1e-6
+        assert abs(float(hamming(t, p))
+                   - hamming_loss(t.numpy(), p.numpy())) < 1e-6
+        ours = fbeta_per_class(t, p, beta=beta).numpy()
+        ref = fbeta_score(t.numpy(), p.numpy(), beta=beta, average=None,
+                          zero_division=0)
+        assert abs(ours - ref).max() < 1e-6
+        assert (multilabel_confusion(t, p).numpy()
+                == multilabel_confusion_matrix(t.numpy(), p.numpy())).all()
+
+    check()
