@@ -78,3 +78,34 @@ def test_split_sizes():
     assert n_tr == int(0.8 * total)
     assert n_va == int(0.1 * total) + 1
     assert n_tr + n_va + n_te <= total + 2
+
+
+def test_chunk_window_coverage_property():
+    """Property: across all chunks, the window END ids are contiguous with
+    no duplicates and reach the last row — the chunk overlap (window-1)
+    exists exactly so that no training sample is lost or double-served at
+    chunk boundaries (reference sql_pytorch_dataloader.py:72-78)."""
+    import torch
+    from hypothesis import given, settings, strategies as st
+
+    from fmda_amd.data import ChunkLoader
+
+    @settings(max_examples=30, deadline=None)
+    @given(st.integers(2, 30), st.integers(1, 200))
+    def check(window, extra):
+        chunk_size = window * 2 + 7
+        db_length = chunk_size + extra
+        X = torch.randn(db_length, 5)
+        cl = ChunkLoader(X, chunk_size, window,
+                         feature_names=[f"f{i}" for i in range(5)])
+        ends = []
+        for ci in range(len(cl)):
+            ids, _ = cl[ci]
+            if len(ids) >= window:
+                ends.extend(ids[window - 1:])
+        assert len(ends) == len(set(ends)), "duplicate window ends"
+        assert ends == sorted(ends)
+        assert ends[-1] == db_length, (ends[-1], db_length)
+        assert ends == list(range(ends[0], db_length + 1)), "gap in coverage"
+
+    check()
