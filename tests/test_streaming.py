@@ -93,3 +93,47 @@ def test_calendar_gate():
     assert cal.is_open(0)            # Monday
     assert not cal.is_open(5 * 78)   # Saturday
     assert MarketCalendar(forex_fallback=True).is_open(5 * 78)
+
+
+def test_bus_threaded_producer_consumer():
+    """Topic reads block until publish (condition variable) and deliver in
+    order across threads — the Kafka-consumer-loop semantics predict.py
+    relies on."""
+    import threading
+    import time
+
+    from fmda_amd.runtime.bus import MessageBus
+
+    bus = MessageBus()
+    got = []
+
+    def consume():
+        for msg in bus.consume("predict_timestamp", from_end=False,
+                               timeout=5.0):
+            got.append(msg["i"])
+            if msg["i"] == 99:
+                return
+
+    t = threading.Thread(target=consume)
+    t.start()
+    for i in range(100):
+        bus.publish("predict_timestamp", {"i": i})
+        if i % 25 == 0:
+            time.sleep(0.002)   # let the consumer interleave
+    t.join(timeout=10)
+    assert not t.is_alive()
+    assert got == list(range(100))
+
+
+def test_bus_seek_to_end_semantics():
+    """from_end=True must skip everything already published (the reference
+    consumer's seek_to_end, predict.py:30)."""
+    from fmda_amd.runtime.bus import MessageBus
+
+    bus = MessageBus()
+    for i in range(5):
+        bus.publish("vix", {"i": i})
+    it = bus.consume("vix", from_end=True, timeout=0.05)
+    bus.publish("vix", {"i": 99})
+    first = next(it)
+    assert first["i"] == 99
