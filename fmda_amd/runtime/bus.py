@@ -63,12 +63,20 @@ class MessageBus:
     def consume(self, topic: str, from_end: bool = True,
                 timeout: Optional[float] = None) -> Iterator[Any]:
         """Iterator over messages, starting at the end like the reference
-        consumer's seek_to_end (predict.py:30)."""
+        consumer's seek_to_end (predict.py:30). The end offset is captured
+        HERE (consumer creation), not lazily at the first next() — a
+        generator body would only run on first iteration and would skip
+        messages published in between."""
         t = self.topic(topic)
         offset = t.end_offset() if from_end else 0
-        while True:
-            msg = t.read(offset, timeout=timeout)
-            if msg is None:
-                return
-            offset += 1
-            yield msg
+
+        def _iter():
+            nonlocal offset
+            while True:
+                msg = t.read(offset, timeout=timeout)
+                if msg is None:
+                    return
+                offset += 1
+                yield msg
+
+        return _iter()
