@@ -137,3 +137,23 @@ def test_bus_seek_to_end_semantics():
     bus.publish("vix", {"i": 99})
     first = next(it)
     assert first["i"] == 99
+
+
+def test_market_calendar_gates_bars():
+    """Closed days are skipped (producer.py:159-165 market-closed path);
+    the FOREX fallback keeps a 24h session (producer.py:239-243)."""
+    from fmda_amd.data.generator import SyntheticMarket
+    from fmda_amd.runtime import MarketCalendar, MarketSession, MessageBus
+
+    market = SyntheticMarket(78 * 7, seed=3)   # one full week of bars
+    cal = MarketCalendar(bars_per_day=78, open_days=(0, 1))  # Mon+Tue only
+    bus = MessageBus()
+    s = MarketSession(market, bus=bus, calendar=cal)
+    s.run()
+    assert s.published == 78 * 2          # two open days of the seven
+    assert bus.topic("deep").end_offset() == 78 * 2
+
+    cal24 = MarketCalendar(forex_fallback=True)
+    s2 = MarketSession(market, bus=MessageBus(), calendar=cal24)
+    s2.run()
+    assert s2.published == 78 * 7         # nothing skipped
