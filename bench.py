@@ -238,7 +238,7 @@ def main():
         n_gpus = world if (use_cuda or world > 1) else 1
         seq_per_sec = n_gpus * batch * args.steps / dt_max
         result = {
-            "metric": "train seq/sec, biGRU on synthetic order-book",
+            "metric": "train seq/sec + 3-class acc, biGRU on synthetic order-book",
             "value": seq_per_sec,
             "unit": "seq/s",
             "n_gpus": n_gpus,
