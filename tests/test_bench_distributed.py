@@ -26,3 +26,11 @@ def test_bench_under_torchrun_world2_cpu():
     assert rec["n_gpus"] == 2 and rec["config"]["parallelism"] == "dp2"
     assert rec["value"] > 0 and rec["scaling"] == "weak"
     assert rec["config"]["global_batch"] == 2 * 8  # per-rank CPU batch cap
+    # full driver-contract key set
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in rec, key
+    assert rec["data"] == "synthetic" and rec["higher_is_better"] is True
+    for ckey in ("model", "global_batch", "seq_len", "parallelism"):
+        assert ckey in rec["config"], ckey
