@@ -181,6 +181,9 @@ def main():
                     help="checkpoint to resume from (model_params.pt format)")
     ap.add_argument("--log-file", default=None,
                     help="append per-epoch JSONL records to this file")
+    ap.add_argument("--plots", default=None,
+                    help="directory for learning-curve / confusion plots "
+                         "(the notebook's cells 30-31 figures)")
     args = ap.parse_args()
     mcfg = ModelConfig(hidden_size=args.hidden, n_layers=args.layers,
                        spatial_dropout=False, dropout=0.5)
@@ -196,8 +199,28 @@ def main():
             fh.flush()
     else:
         log = print
+    records = []
+
+    def log2(line):
+        records.append(line)
+        log(line)
+
     train(mcfg, dcfg, tcfg, checkpoint_path=args.checkpoint,
-          resume=args.resume, log=log)
+          resume=args.resume, log=log2)
+    if args.plots:
+        import json as _json
+        import os as _os
+
+        from .plots import confusion_heatmaps, learning_curves
+        _os.makedirs(args.plots, exist_ok=True)
+        recs = [_json.loads(r) for r in records]
+        learning_curves(recs, _os.path.join(args.plots,
+                                            "learning_curves.png"))
+        test_rec = next((r for r in recs if "confusion" in r), None)
+        if test_rec and test_rec["confusion"]:
+            confusion_heatmaps(test_rec["confusion"],
+                               _os.path.join(args.plots,
+                                             "confusion_matrix.png"))
 
 
 if __name__ == "__main__":

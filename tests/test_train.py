@@ -105,3 +105,26 @@ def test_exact_midtraining_resume(tmp_path):
     for ra, rb in zip(ep_a[2:], ep_b[2:]):   # epochs 3 and 4
         assert abs(ra["train_loss"] - rb["train_loss"]) < 1e-6, (ra, rb)
         assert ra["train_acc"] == rb["train_acc"]
+
+
+def test_training_plots(tmp_path, monkeypatch):
+    """--plots produces the notebook's learning-curve and confusion-matrix
+    figures (cells 30-31)."""
+    import os
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+    out = subprocess.run(
+        [sys.executable, "-m", "fmda_amd.train", "--epochs", "2",
+         "--rows", "600", "--checkpoint", str(tmp_path / "m.pt"),
+         "--plots", str(tmp_path / "plots")],
+        capture_output=True, text=True, timeout=300, cwd=str(tmp_path),
+        env=env)
+    assert out.returncode == 0, out.stderr[-1500:]
+    curves = tmp_path / "plots" / "learning_curves.png"
+    conf = tmp_path / "plots" / "confusion_matrix.png"
+    assert curves.exists() and curves.stat().st_size > 5000
+    assert conf.exists() and conf.stat().st_size > 5000
