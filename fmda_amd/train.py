@@ -65,6 +65,13 @@ def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
 
     market = SyntheticMarket(dcfg.n_rows, seed=dcfg.seed)
     weight, pos_weight = class_weights(market.Y)
+    # class-imbalance accounting (notebook cells 13-14 print the dataset
+    # size and per-class positive counts before choosing loss weights)
+    log(json.dumps({"n_rows": dcfg.n_rows,
+                    "positives": [int(v) for v in market.Y.sum(dim=0)],
+                    "loss_weight": [round(float(v), 3) for v in weight],
+                    "pos_weight": [round(float(v), 3)
+                                   for v in pos_weight]}))
 
     model = BiGRU(mcfg.hidden_size, mcfg.n_features, mcfg.output_size,
                   n_layers=mcfg.n_layers, clip=mcfg.clip,
