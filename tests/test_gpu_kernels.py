@@ -416,8 +416,7 @@ def test_deferred_dropout_backward_matches_explicit():
     p_drop, seed = 0.3, 987654321
     masters = []
     for _ in range(2):
-        masters += [torch.randn(3 * H, F if _ == 0 or True else F,
-                                device="cuda") * 0.2,
+        masters += [torch.randn(3 * H, F, device="cuda") * 0.2,
                     torch.randn(3 * H, H, device="cuda") * 0.2,
                     torch.randn(3 * H, device="cuda") * 0.1,
                     torch.randn(3 * H, device="cuda") * 0.1]
