@@ -66,3 +66,17 @@ def test_streaming_predictor_runs_reference_model():
     pred_dict = pred.predict_window()
     assert len(pred_dict["probabilities"]) == 4
     assert all(0.0 <= p <= 1.0 for p in pred_dict["probabilities"])
+
+
+def test_reference_ranges_match_real_artifact():
+    """feature_ranges.REFERENCE_RANGES (the generator's value bounds) must
+    equal the real norm_params artifact's recorded MIN/MAX."""
+    import torch
+
+    from fmda_amd.data.norm import load_norm_params
+    from fmda_amd.features import REFERENCE_RANGES
+    names, x_min, x_max = load_norm_params(os.path.join(REF, "norm_params"))
+    ours_lo = torch.tensor([REFERENCE_RANGES[n][0] for n in names])
+    ours_hi = torch.tensor([REFERENCE_RANGES[n][1] for n in names])
+    torch.testing.assert_close(ours_lo, x_min, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(ours_hi, x_max, rtol=1e-5, atol=1e-5)
