@@ -20,6 +20,12 @@ produced with PySpark Structured Streaming + MariaDB:
   shared rolling helpers of the generator;
 - a `predict_timestamp` signal per assembled row (spark_consumer.py:490-502).
 
+Threading contract: handlers run synchronously in the publisher's thread
+(the in-process bus delivers on publish). Ingestion is single-writer;
+call `table()` after the feed drains (or under external synchronization)
+— matching the reference, where the Spark job owns all mutation and the
+dataloader reads the warehouse afterwards.
+
 `table()` returns the assembled (n, 108) float32 tensor in canonical
 registry order — bit-comparable (to float tolerance) with the
 `SyntheticMarket` table the raw feeds were derived from, which is exactly

@@ -43,3 +43,23 @@ def test_graph_capture_happens_once():
     g1 = p._graph
     p.predict_window()
     assert p._graph is g1 and g1 is not None
+
+
+@pytest.mark.gpu
+def test_end_to_end_demo_on_gpu(tmp_path):
+    """examples/end_to_end.py --device cuda: GPU training through the
+    reference train_model API, checkpoint save, and the hipGraph streaming
+    session, end to end."""
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "examples", "end_to_end.py"),
+         "--device", "cuda", "--epochs", "1", "--rows", "500",
+         "--outdir", str(tmp_path)],
+        capture_output=True, text=True, timeout=300, cwd=repo)
+    assert out.returncode == 0, (out.stdout[-800:], out.stderr[-1200:])
+    assert "predictions" in out.stdout
+    assert (tmp_path / "model_params.pt").exists()
