@@ -176,8 +176,11 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
         // fused inter-layer dropout backward (d_out is w.r.t. the DROPPED
         // activations; the kernel recomputes the counter-based mask at the
         // dout read instead of a separate full-tensor pass)
+        // quantized EXACTLY like dropout_kernel's forward threshold
+        // ((unsigned)(float_p * 256.0f)) so the recomputed backward mask
+        // can never disagree with the forward one at a rounding boundary
         const unsigned int drop_thr =
-            (drop_p > 0.0) ? (unsigned int)(drop_p * 256.0) : 0u;
+            (drop_p > 0.0) ? (unsigned int)((float)drop_p * 256.0f) : 0u;
         const float drop_scale =
             (drop_p > 0.0) ? (float)(1.0 / (1.0 - drop_p)) : 1.0f;
         TORCH_CHECK(drop_thr == 0u || (is_bf16 && Hp == 128),

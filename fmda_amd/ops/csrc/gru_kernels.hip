@@ -1925,7 +1925,7 @@ extern "C" int fmda_gru_fwd_cs_launch(const void* gi, const void* w,
                                  (__hip_bfloat16*)out, hlast,
                                  (__hip_bfloat16*)hpub, cnt, B, Tseq, n_dir,
                                  GB);
-    return 0;
+    return hipGetLastError() == hipSuccess ? 0 : -1;
 }
 
 // ===========================================================================
@@ -2891,7 +2891,7 @@ extern "C" int fmda_gru_bwd_cs_launch(const void* gi, const void* w,
         (const __hip_bfloat16*)dout, dhT, (__hip_bfloat16*)dgi,
         (__hip_bfloat16*)dgh, dh0, dbhh, (__hip_bfloat16*)gpub, cnt, B,
         Tseq, n_dir, GB);
-    return 0;
+    return hipGetLastError() == hipSuccess ? 0 : -1;
 }
 
 extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,

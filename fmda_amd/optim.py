@@ -76,7 +76,17 @@ class FusedClipAdam:
                 goff.append(4 * off)
                 off += c
         cpu = torch.tensor(rows, dtype=torch.int64)
-        self._cpu_table = cpu.pin_memory() if self.params[0].is_cuda else cpu
+        on_gpu = self.params[0].is_cuda
+        # DOUBLE-buffered pinned staging: the H2D copy is async, so the
+        # buffer a pending copy reads must never be the one being rewritten
+        # for the next step. Two buffers alternate, and each is guarded by
+        # an event recorded after its copy — waited on (host-side) before
+        # the buffer is mutated again.
+        self._cpu_tables = [cpu.pin_memory() if on_gpu else cpu,
+                            cpu.clone().pin_memory() if on_gpu else cpu.clone()]
+        self._copy_events = [torch.cuda.Event(), torch.cuda.Event()] \
+            if on_gpu else [None, None]
+        self._buf_i = 0
         self._gidx = torch.tensor(gidx, dtype=torch.int64)
         self._goff = torch.tensor(goff, dtype=torch.int64)
         self._chunks = torch.empty_like(cpu, device=self.params[0].device)
@@ -92,9 +102,16 @@ class FusedClipAdam:
         if self._chunks is None:
             self._build_static()
         if ptrs != self._ptrs:
+            i = self._buf_i
+            ev = self._copy_events[i]
+            if ev is not None:
+                ev.synchronize()  # prior async copy from THIS buffer done
             g = torch.tensor(ptrs, dtype=torch.int64)
-            self._cpu_table[:, 1] = g[self._gidx] + self._goff
-            self._chunks.copy_(self._cpu_table, non_blocking=True)
+            self._cpu_tables[i][:, 1] = g[self._gidx] + self._goff
+            self._chunks.copy_(self._cpu_tables[i], non_blocking=True)
+            if ev is not None:
+                ev.record()
+            self._buf_i = 1 - i
             self._ptrs = ptrs
         # the HIP kernel writes params through raw pointers: bump the
         # packed-weight cache epoch (torch _version counters don't move)

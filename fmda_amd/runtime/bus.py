@@ -31,7 +31,8 @@ class Topic:
             cb(value)
 
     def subscribe(self, callback: Callable[[Any], None]) -> None:
-        self._subscribers.append(callback)
+        with self._cv:
+            self._subscribers = self._subscribers + [callback]
 
     def end_offset(self) -> int:
         with self._cv:

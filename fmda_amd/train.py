@@ -161,7 +161,7 @@ def train(mcfg: ModelConfig = None, dcfg: DataConfig = None,
     from .metrics import fbeta_per_class
     # the notebook's held-out test metrics use beta=1 (cells 33-36), while
     # the epoch loops use beta=0.5 (biGRU_model.py:221,279)
-    fb1 = (fbeta_per_class(te[4], te[3], beta=1.0).tolist()
+    fb1 = ([_num(v) for v in fbeta_per_class(te[4], te[3], beta=1.0).tolist()]
            if has_rows else [None])
     test_rec = {"test_acc": _num(te[0]), "test_hamming": _num(te[1]),
                 "test_fbeta": fb1,
