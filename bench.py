@@ -92,6 +92,10 @@ def main():
                     help="repo|cpu|stress|predict (default: repo on GPU, cpu otherwise)")
     ap.add_argument("--batch", type=int, default=None,
                     help="override per-GPU batch")
+    ap.add_argument("--ddp-mode", default=None, choices=["bucketed", "fused"],
+                    help="gradient sync mode (default: bucketed; fused for "
+                         "the stress config, whose 64 MB payload would "
+                         "otherwise split into ~64 latency-bound collectives)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -136,7 +140,9 @@ def main():
     engine = None
     if world > 1:
         from fmda_amd.parallel import GradAllReduce
-        engine = GradAllReduce(model)
+        ddp_mode = args.ddp_mode or ("fused" if cfg_name == "stress"
+                                     else "bucketed")
+        engine = GradAllReduce(model, mode=ddp_mode)
 
     # Class weighting like the training notebook (cell 16): weight =
     # total/positives, pos_weight = negatives/positives, from the synthetic

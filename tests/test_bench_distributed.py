@@ -10,22 +10,28 @@ import sys
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def test_bench_under_torchrun_world2_cpu():
+import pytest
+
+
+@pytest.mark.parametrize("world,port", [(2, 29575), (4, 29576)])
+def test_bench_under_torchrun_cpu(world, port):
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29571", os.path.join(REPO, "bench.py"),
-         "--gpus", "2", "--config", "cpu", "--steps", "2", "--warmup", "1"],
+         "--nproc-per-node", str(world), "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(REPO, "bench.py"),
+         "--gpus", str(world), "--config", "cpu", "--steps", "2",
+         "--warmup", "1"],
         cwd=REPO, env=env, capture_output=True, text=True, timeout=420)
     assert out.returncode == 0, out.stderr[-2000:]
     lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
     assert len(lines) == 1, out.stdout  # exactly one JSON line from rank 0
     rec = json.loads(lines[0])
-    assert rec["n_gpus"] == 2 and rec["config"]["parallelism"] == "dp2"
+    assert rec["n_gpus"] == world
+    assert rec["config"]["parallelism"] == f"dp{world}"
     assert rec["value"] > 0 and rec["scaling"] == "weak"
-    assert rec["config"]["global_batch"] == 2 * 8  # per-rank CPU batch cap
+    assert rec["config"]["global_batch"] == world * 8  # per-rank CPU cap
     # full driver-contract key set
     for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
                 "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
