@@ -7,7 +7,7 @@
 extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const float* bhh, void* out,
                                    float* hlast, int B, int Tseq, int n_dir,
-                                   hipStream_t stream);
+                                   const float* h0, hipStream_t stream);
 extern "C" int fmda_gru_fwd_cs_launch(const void* gi, const void* w,
                                       const float* bhh, void* out,
                                       float* hlast, void* hpub,
@@ -29,6 +29,7 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    int n_dir, unsigned int drop_thr,
                                    float drop_scale,
                                    unsigned long long drop_seed,
+                                   const float* h0, void* dgh0,
                                    hipStream_t stream);
 extern "C" int fmda_mfma_selftest_launch(const void* A, const void* Bm,
                                          float* C, hipStream_t stream);
@@ -95,16 +96,28 @@ void check_common(const torch::Tensor& gi, const torch::Tensor& w,
 }  // namespace
 
 std::vector<torch::Tensor> gru_fwd(torch::Tensor gi, torch::Tensor w,
-                                   torch::Tensor bhh) {
+                                   torch::Tensor bhh,
+                                   c10::optional<torch::Tensor> h0_opt) {
     int B, T, n_dir, Hp;
     bool is_bf16;
     check_common(gi, w, bhh, B, T, n_dir, Hp, is_bf16);
+    const float* h0 = nullptr;
+    if (h0_opt.has_value()) {
+        auto& h = h0_opt.value();
+        TORCH_CHECK(h.is_cuda() && h.is_contiguous() &&
+                    h.scalar_type() == torch::kFloat32 &&
+                    h.sizes() == torch::IntArrayRef({n_dir, B, Hp}),
+                    "h0 must be contiguous fp32 (n_dir, B, Hp)");
+        h0 = h.data_ptr<float>();
+    }
     auto out = torch::empty({B, T, (int64_t)n_dir * Hp}, gi.options());
     auto hlast = torch::empty({n_dir, B, Hp},
                               gi.options().dtype(torch::kFloat32));
     auto stream = at::hip::getCurrentHIPStream();
     int rc;
-    if (is_bf16 && Hp == 512) {
+    // the column-split Hp=512 kernel pre-publishes h_{-1}=0; with a real
+    // h0 the batch-parallel kernel takes the (rare) call instead
+    if (is_bf16 && Hp == 512 && h0 == nullptr) {
         // column-split persistent kernel: publication ring + group counters
         const int BR = 256;
         const int GB = (B + BR - 1) / BR;
@@ -120,7 +133,7 @@ std::vector<torch::Tensor> gru_fwd(torch::Tensor gi, torch::Tensor w,
         rc = fmda_gru_fwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
                                  w.data_ptr(), bhh.data_ptr<float>(),
                                  out.data_ptr(), hlast.data_ptr<float>(), B, T,
-                                 n_dir, stream.stream());
+                                 n_dir, h0, stream.stream());
     }
     TORCH_CHECK(rc == 0, "fmda gru_fwd launch failed rc=", rc, " Hp=", Hp);
     return {out, hlast};
@@ -129,10 +142,24 @@ std::vector<torch::Tensor> gru_fwd(torch::Tensor gi, torch::Tensor w,
 std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
                                    torch::Tensor bhh, torch::Tensor out,
                                    torch::Tensor dout, torch::Tensor dhT,
-                                   double drop_p, int64_t drop_seed) {
+                                   double drop_p, int64_t drop_seed,
+                                   c10::optional<torch::Tensor> h0_opt) {
     int B, T, n_dir, Hp;
     bool is_bf16;
     check_common(gi, w, bhh, B, T, n_dir, Hp, is_bf16);
+    const float* h0 = nullptr;
+    torch::Tensor dgh0;
+    void* dgh0_ptr = nullptr;
+    if (h0_opt.has_value()) {
+        auto& h = h0_opt.value();
+        TORCH_CHECK(h.is_cuda() && h.is_contiguous() &&
+                    h.scalar_type() == torch::kFloat32 &&
+                    h.sizes() == torch::IntArrayRef({n_dir, B, Hp}),
+                    "h0 must be contiguous fp32 (n_dir, B, Hp)");
+        h0 = h.data_ptr<float>();
+        dgh0 = torch::empty({n_dir, B, (int64_t)3 * Hp}, gi.options());
+        dgh0_ptr = dgh0.data_ptr();
+    }
     TORCH_CHECK(out.is_contiguous() && dout.is_contiguous() &&
                 dhT.is_contiguous(), "fmda gru_bwd: tensors must be contiguous");
     TORCH_CHECK(dhT.scalar_type() == torch::kFloat32, "dhT must be fp32");
@@ -160,7 +187,7 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
     }
     auto stream = at::hip::getCurrentHIPStream();
     int rc;
-    if (is_bf16 && Hp == 512) {
+    if (is_bf16 && Hp == 512 && h0 == nullptr) {
         const int BR = 256;
         const int GB = (B + BR - 1) / BR;
         const int G = GB * n_dir;
@@ -192,7 +219,7 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
                                  dgh.data_ptr(), dh0.data_ptr<float>(),
                                  dbsum.data_ptr<float>(), B, T,
                                  n_dir, drop_thr, drop_scale,
-                                 (unsigned long long)drop_seed,
+                                 (unsigned long long)drop_seed, h0, dgh0_ptr,
                                  stream.stream());
     }
     TORCH_CHECK(rc == 0, "fmda gru_bwd launch failed rc=", rc, " Hp=", Hp);
@@ -200,6 +227,8 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
     auto dbhh = dbsum.index({Slice(), Slice(0, 3 * Hp)});
     auto dbih = torch::cat({dbsum.index({Slice(), Slice(0, 2 * Hp)}),
                             dbsum.index({Slice(), Slice(3 * Hp, 4 * Hp)})}, 1);
+    if (dgh0_ptr != nullptr)
+        return {dgi, dgh, dh0, dbhh.contiguous(), dbih, dgh0};
     return {dgi, dgh, dh0, dbhh.contiguous(), dbih};
 }
 
@@ -327,11 +356,13 @@ torch::Tensor dropout_fused(torch::Tensor x, double p, int64_t seed) {
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-    m.def("gru_fwd", &gru_fwd, "fused biGRU recurrence forward (HIP/CDNA4)");
+    m.def("gru_fwd", &gru_fwd, "fused biGRU recurrence forward (HIP/CDNA4)",
+          py::arg("gi"), py::arg("w"), py::arg("bhh"),
+          py::arg("h0") = py::none());
     m.def("gru_bwd", &gru_bwd, "fused biGRU recurrence backward (HIP/CDNA4)",
           py::arg("gi"), py::arg("w"), py::arg("bhh"), py::arg("out"),
           py::arg("dout"), py::arg("dhT"), py::arg("drop_p") = 0.0,
-          py::arg("drop_seed") = 0);
+          py::arg("drop_seed") = 0, py::arg("h0") = py::none());
     m.def("mfma_selftest", &mfma_selftest, "mfma fragment layout self-test");
     m.def("dropout_fused", &dropout_fused,
           "counter-based dropout (mask recomputed in backward)");

@@ -139,6 +139,21 @@ FMDA_DEV void zero_tile(T* __restrict__ lds, int lds_pitch, int tid) {
         lds[c] = from_f32<T>(0.0f);
 }
 
+// Initial hidden state h0 (fp32, contiguous (rows, cols)) -> LDS tile of
+// dtype T at `lds_pitch`; dead rows and pad columns are zeroed (nn.GRU h0
+// support, reference biGRU_model.py:102 `self.gru(input_seq, hidden)`).
+template <typename T, int ROWS, int NT>
+FMDA_DEV void stage_h0_tile(T* __restrict__ lds, const float* __restrict__ h0,
+                            int cols, int lds_pitch, int rows_valid, int tid) {
+    for (int c = tid; c < ROWS * lds_pitch; c += NT) {
+        const int r = c / lds_pitch;
+        const int j = c % lds_pitch;
+        const float v = (r < rows_valid && j < cols)
+                            ? h0[(long)r * cols + j] : 0.0f;
+        lds[c] = from_f32<T>(v);
+    }
+}
+
 // fp32 LDS tile += T-typed global tile (dh += dOut[t]); synchronous form.
 template <typename T, int ROWS, int NT>
 FMDA_DEV void accum_tile_f32(float* __restrict__ lds, const T* __restrict__ gp,
@@ -266,7 +281,8 @@ __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
                           amdgpu_waves_per_eu(NT / 256, NT / 256))) void gru_fwd_kernel(
     const T* __restrict__ gi, const T* __restrict__ w,
     const float* __restrict__ bhh, T* __restrict__ out,
-    float* __restrict__ hlast, int B, int Tseq, int n_dir) {
+    float* __restrict__ hlast, int B, int Tseq, int n_dir,
+    const float* __restrict__ h0) {
     constexpr int NW = NT / 64;
     constexpr int MT = BT / 16;
     constexpr int NCT = Hp / 16;
@@ -303,8 +319,16 @@ __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
         stage_tile<T, 3 * Hp, NT>(w_s, wdir, Hp, WP, Hp, 3 * Hp, tid);
     for (int c = tid; c < 3 * Hp; c += NT)
         bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
-    zero_tile<float, BT, NT>(hf_s, HFP, tid);
-    if (IS_BF16) zero_tile<__hip_bfloat16, BT, NT>(hb_s, WP, tid);
+    if (h0 != nullptr) {
+        const float* h0d = h0 + ((long)dir * B + b0) * Hp;
+        stage_h0_tile<float, BT, NT>(hf_s, h0d, Hp, HFP, rows_valid, tid);
+        if (IS_BF16)
+            stage_h0_tile<__hip_bfloat16, BT, NT>(hb_s, h0d, Hp, WP,
+                                                  rows_valid, tid);
+    } else {
+        zero_tile<float, BT, NT>(hf_s, HFP, tid);
+        if (IS_BF16) zero_tile<__hip_bfloat16, BT, NT>(hb_s, WP, tid);
+    }
 
     const long gi_row = (long)Tseq * n_dir * 3 * Hp;
     const long out_row = (long)Tseq * n_dir * Hp;
@@ -479,7 +503,8 @@ __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
     const float* __restrict__ bhh, const T* __restrict__ out,
     const T* __restrict__ dout, const float* __restrict__ dhT,
     T* __restrict__ dgi, T* __restrict__ dgh, float* __restrict__ dh0,
-    float* __restrict__ dbhh, int B, int Tseq, int n_dir) {
+    float* __restrict__ dbhh, int B, int Tseq, int n_dir,
+    const float* __restrict__ h0, T* __restrict__ dgh0) {
     constexpr int NW = NT / 64;
     constexpr int MT = BT / 16;
     constexpr int NCT = Hp / 16;
@@ -546,6 +571,9 @@ __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
             const int ttp = rev ? (Tseq - u) : (u - 1);
             stage_tile<T, BT, NT>(hb_s, out_b + (long)ttp * n_dir * Hp, Hp, WP,
                                   out_row, rows_valid, tid);
+        } else if (h0 != nullptr) {
+            stage_h0_tile<T, BT, NT>(hb_s, h0 + ((long)dir * B + b0) * Hp, Hp,
+                                     WP, rows_valid, tid);
         } else {
             zero_tile<T, BT, NT>(hb_s, WP, tid);
         }
@@ -770,6 +798,10 @@ __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
         if (have_next) {
             if (un > 0)
                 pf_hb.commit(hb_s, Hp, WP, tid);
+            else if (h0 != nullptr)
+                stage_h0_tile<T, BT, NT>(hb_s,
+                                         h0 + ((long)dir * B + b0) * Hp, Hp,
+                                         WP, rows_valid, tid);
             else
                 zero_tile<T, BT, NT>(hb_s, WP, tid);
             pf_do.commit_accum_f32(dh_s, Hp, HFP, rows_valid, tid);
@@ -778,14 +810,23 @@ __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
                               GP3, gi_row, rows_valid, tid);
         {
             const int slot = rev ? (tt + 1) : (tt - 1);
-            if (slot >= 0 && slot < Tseq)
+            if (slot >= 0 && slot < Tseq) {
                 store_tile<T, BT, NT>(dgh_s,
                                       dgh_b + (long)slot * n_dir * 3 * Hp,
                                       3 * Hp, GP3, gi_row, rows_valid, tid);
-            else
+            } else {
+                // u = 0: this dGh pairs with h_{-1}. Its global slot gets
+                // zeros (the dW_hh reduction pairs slots with out[t]); with
+                // an initial hidden state the values go to dgh0 instead so
+                // the host can add the dGh_0 (x) h0 term to dW_hh.
                 store_zero_tile<T, BT, NT>(
                     dgh_b + (long)(rev ? 0 : (Tseq - 1)) * n_dir * 3 * Hp,
                     3 * Hp, gi_row, rows_valid, tid);
+                if (dgh0 != nullptr)
+                    store_tile<T, BT, NT>(
+                        dgh_s, dgh0 + ((long)dir * B + b0) * 3 * Hp, 3 * Hp,
+                        GP3, 3 * Hp, rows_valid, tid);
+            }
         }
         if (have_next)
             pf_gi.commit(gi_s, 3 * Hp, GP3, tid);
@@ -855,7 +896,7 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                        const float* __restrict__ bhh,
                        __hip_bfloat16* __restrict__ out,
                        float* __restrict__ hlast, int B, int Tseq,
-                       int n_dir) {
+                       int n_dir, const float* __restrict__ h0) {
     constexpr int NW = NT / 64;
     constexpr int MT = BT / 16;
     constexpr int NCT = Hp / 16;
@@ -897,7 +938,13 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 
     for (int c = tid; c < 3 * Hp; c += NT)
         bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
-    zero_tile<__hip_bfloat16, BT, NT>(hb_s0, WP, tid);
+    const float* h0d = (h0 != nullptr) ? h0 + ((long)dir * B + b0) * Hp
+                                       : nullptr;
+    if (h0d != nullptr)
+        stage_h0_tile<__hip_bfloat16, BT, NT>(hb_s0, h0d, Hp, WP, rows_valid,
+                                              tid);
+    else
+        zero_tile<__hip_bfloat16, BT, NT>(hb_s0, WP, tid);
 
     // W_hh B-fragments, register-resident for the whole sequence.
     const __hip_bfloat16* wdir = w + (long)dir * 3 * Hp * Hp;
@@ -914,11 +961,18 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 
     float hreg[CPW][MT][4];
 #pragma unroll
-    for (int i = 0; i < CPW; ++i)
+    for (int i = 0; i < CPW; ++i) {
+        const int ct = wave + NW * i;
+        const int j = ct * 16 + (lane & 15);
 #pragma unroll
         for (int m = 0; m < MT; ++m)
 #pragma unroll
-            for (int e = 0; e < 4; ++e) hreg[i][m][e] = 0.0f;
+            for (int e = 0; e < 4; ++e) {
+                const int b = 16 * m + 4 * (lane >> 4) + e;
+                hreg[i][m][e] = (h0d != nullptr && b < rows_valid)
+                                    ? h0d[(long)b * Hp + j] : 0.0f;
+            }
+    }
 
     // Lane-linear LDS-DMA of one timestep's gi tile (per-lane global
     // addresses handle the strided batch rows; dead rows clamp to row 0).
@@ -1098,7 +1152,8 @@ void gru_fwd_b1_kernel(const __hip_bfloat16* __restrict__ gi,
                        const __hip_bfloat16* __restrict__ w,
                        const float* __restrict__ bhh,
                        __hip_bfloat16* __restrict__ out,
-                       float* __restrict__ hlast, int Tseq, int n_dir) {
+                       float* __restrict__ hlast, int Tseq, int n_dir,
+                       const float* __restrict__ h0) {
     constexpr int NW = NT / 64;      // 4 waves
     constexpr int CT = Hp / 16;      // 8 h-column tiles
     constexpr int CPW = CT / NW;     // 2 per wave
@@ -1132,8 +1187,11 @@ void gru_fwd_b1_kernel(const __hip_bfloat16* __restrict__ gi,
     }
     for (int c = tid; c < 3 * Hp; c += NT)
         bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
+    const float* h0d = (h0 != nullptr) ? h0 + (long)dir * Hp : nullptr;
     for (int c = tid; c < HRP; c += NT)
-        ((__bf16*)hr0)[c] = (__bf16)0.0f;
+        ((__bf16*)hr0)[c] = (h0d != nullptr && c < Hp)
+                                ? (__bf16)__float2bfloat16(h0d[c])
+                                : (__bf16)0.0f;
 
     const __hip_bfloat16* wdir = w + (long)dir * 3 * Hp * Hp;
     bf16x8_t wA[CPW][3][KK];
@@ -1148,7 +1206,9 @@ void gru_fwd_b1_kernel(const __hip_bfloat16* __restrict__ gi,
     }
     float hreg[CPW];                 // this lane's h column (lanes 0..15)
 #pragma unroll
-    for (int i = 0; i < CPW; ++i) hreg[i] = 0.0f;
+    for (int i = 0; i < CPW; ++i)
+        hreg[i] = (h0d != nullptr)
+                      ? h0d[(wave + NW * i) * 16 + (lane & 15)] : 0.0f;
 
     __hip_bfloat16* out_b = out + (long)dir * Hp;   // batch row 0
     const long out_trow = (long)n_dir * Hp;
@@ -1228,7 +1288,9 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                        __hip_bfloat16* __restrict__ dgh, float* __restrict__ dh0,
                        float* __restrict__ dbhh, int B, int Tseq, int n_dir,
                        unsigned int drop_thr, float drop_scale,
-                       unsigned long long drop_seed) {
+                       unsigned long long drop_seed,
+                       const float* __restrict__ h0,
+                       __hip_bfloat16* __restrict__ dgh0) {
     constexpr int NW = NT / 64;
     constexpr int MT = BT / 16;
     constexpr int NCT = Hp / 16;
@@ -1372,7 +1434,25 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                    (char*)hb_buf(buf) + (long)(wavu * HBP + k) * 1024);
         }
     };
+    // Boundary h image: zeros, or h0 when an initial hidden state is given.
+    // h0 goes in SWIZZLED (LDS[r][k16] = h[r][k16 ^ (r & 15)]) like the
+    // glds_hb image so hb_read / the A-fragment reads see the same layout.
+    // The h0 loads are ordinary VMEM loads; at the in-loop call site (u==1)
+    // they cost one vmcnt(0) drain of that step's DMA — once per launch.
     auto zero_hb = [&](int buf) {
+        if (h0 != nullptr) {
+            const float* h0d = h0 + ((long)dir * B + b0) * Hp;
+            char* dst = (char*)hb_buf(buf);
+            for (int c = tid; c < BT * Hp; c += NT) {
+                const int r = c / Hp, j = c % Hp;
+                const float v =
+                    (r < rows_valid) ? h0d[(long)r * Hp + j] : 0.0f;
+                const int k16 = ((j * 2) >> 4) ^ (r & 15);
+                *(__bf16*)(dst + r * HROW + k16 * 16 + ((j * 2) & 15)) =
+                    (__bf16)__float2bfloat16(v);
+            }
+            return;
+        }
         for (int c = tid; c < BT * Hp; c += NT)
             ((__bf16*)hb_buf(buf))[c] = (__bf16)0.0f;
     };
@@ -1585,11 +1665,17 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
         store_tile<__hip_bfloat16, BT, NT>(
             gi_s0, dgi_b + (long)ttl * n_dir * 3 * Hp, GP, GP, gi_row,
             rows_valid, tid);
-        // step u=0's dGh pairs with h_{-1}=0: discarded; its global slot
-        // (rev ? 0 : Tseq-1) gets zeros instead.
+        // step u=0's dGh pairs with h_{-1}: its global slot (rev ? 0 :
+        // Tseq-1) gets zeros (the dW_hh reduction pairs slots with out[t]);
+        // with an initial hidden state the values go to dgh0 so the host
+        // adds the dGh_0 (x) h0 term to dW_hh.
         store_zero_tile<__hip_bfloat16, BT, NT>(
             dgh_b + (long)(rev ? 0 : (Tseq - 1)) * n_dir * 3 * Hp, GP, gi_row,
             rows_valid, tid);
+        if (dgh0 != nullptr)
+            store_tile<__hip_bfloat16, BT, NT>(
+                dgh_s0, dgh0 + ((long)dir * B + b0) * GP, GP, GP3, GP,
+                rows_valid, tid);
         float* d0 = dh0 + ((long)dir * B + b0) * Hp;
 #pragma unroll
         for (int i = 0; i < CPW; ++i) {
@@ -2726,13 +2812,13 @@ static inline size_t bwd_lds_bytes(bool bf16, int Hp, int bt, bool wlds) {
 template <typename T, int BT, int Hp, bool WLDS, int NT, bool HOIST>
 static int launch_fwd(const void* gi, const void* w, const float* bhh,
                       void* out, float* hlast, int B, int Tseq, int n_dir,
-                      size_t lds, hipStream_t stream) {
+                      const float* h0, size_t lds, hipStream_t stream) {
     auto k = gru_fwd_kernel<T, BT, Hp, WLDS, NT, HOIST>;
     (void)hipFuncSetAttribute((const void*)k,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     const dim3 grid((B + BT - 1) / BT, n_dir);
     k<<<grid, NT, lds, stream>>>((const T*)gi, (const T*)w, bhh, (T*)out,
-                                 hlast, B, Tseq, n_dir);
+                                 hlast, B, Tseq, n_dir, h0);
     return 0;
 }
 
@@ -2740,14 +2826,16 @@ template <typename T, int BT, int Hp, bool WLDS, int NT, bool HOIST>
 static int launch_bwd(const void* gi, const void* w, const float* bhh,
                       const void* out, const void* dout, const float* dhT,
                       void* dgi, void* dgh, float* dh0, float* dbhh, int B,
-                      int Tseq, int n_dir, size_t lds, hipStream_t stream) {
+                      int Tseq, int n_dir, const float* h0, void* dgh0,
+                      size_t lds, hipStream_t stream) {
     auto k = gru_bwd_kernel<T, BT, Hp, WLDS, NT, HOIST>;
     (void)hipFuncSetAttribute((const void*)k,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     const dim3 grid((B + BT - 1) / BT, n_dir);
     k<<<grid, NT, lds, stream>>>((const T*)gi, (const T*)w, bhh,
                                  (const T*)out, (const T*)dout, dhT, (T*)dgi,
-                                 (T*)dgh, dh0, dbhh, B, Tseq, n_dir);
+                                 (T*)dgh, dh0, dbhh, B, Tseq, n_dir, h0,
+                                 (T*)dgh0);
     return 0;
 }
 
@@ -2757,7 +2845,8 @@ using bf16_t = __hip_bfloat16;
 // v3 launch helpers (bf16 Hp=128).
 static int launch_fwd_v3_128(const void* gi, const void* w, const float* bhh,
                              void* out, float* hlast, int B, int Tseq,
-                             int n_dir, hipStream_t stream) {
+                             int n_dir, const float* h0,
+                             hipStream_t stream) {
     static const bool big = getenv("FMDA_FWD_BT32") != nullptr;
     if (big) {   // A/B: one 8-wave block per CU instead of two 4-wave
         constexpr int BT = 32, Hp = 128, NT = 512;
@@ -2770,7 +2859,7 @@ static int launch_fwd_v3_128(const void* gi, const void* w, const float* bhh,
         k2<<<grid, NT, lds, stream>>>((const __hip_bfloat16*)gi,
                                       (const __hip_bfloat16*)w, bhh,
                                       (__hip_bfloat16*)out, hlast, B, Tseq,
-                                      n_dir);
+                                      n_dir, h0);
         return 0;
     }
     constexpr int BT = 16, Hp = 128, NT = 256;
@@ -2782,7 +2871,8 @@ static int launch_fwd_v3_128(const void* gi, const void* w, const float* bhh,
     const dim3 grid((B + BT - 1) / BT, n_dir);
     k<<<grid, NT, lds, stream>>>((const __hip_bfloat16*)gi,
                                  (const __hip_bfloat16*)w, bhh,
-                                 (__hip_bfloat16*)out, hlast, B, Tseq, n_dir);
+                                 (__hip_bfloat16*)out, hlast, B, Tseq, n_dir,
+                                 h0);
     return 0;
 }
 
@@ -2793,6 +2883,7 @@ static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
                              float* dh0, float* dbhh, int B, int Tseq,
                              int n_dir, unsigned int drop_thr,
                              float drop_scale, unsigned long long drop_seed,
+                             const float* h0, void* dgh0,
                              hipStream_t stream) {
     constexpr int BT = 32, Hp = 128, NT = 512;
     const size_t lds = 3 * 2 * BT * 3 * Hp + 2 * 2 * BT * Hp +
@@ -2807,14 +2898,15 @@ static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
         (const __hip_bfloat16*)wt, bhh,
         (const __hip_bfloat16*)out, (const __hip_bfloat16*)dout, dhT,
         (__hip_bfloat16*)dgi, (__hip_bfloat16*)dgh, dh0, dbhh, B, Tseq,
-        n_dir, drop_thr, drop_scale, drop_seed);
+        n_dir, drop_thr, drop_scale, drop_seed, h0,
+        (__hip_bfloat16*)dgh0);
     return 0;
 }
 
 extern "C" int fmda_gru_fwd_b1_launch(const void* gi, const void* w,
                                       const float* bhh, void* out,
                                       float* hlast, int Tseq, int n_dir,
-                                      hipStream_t stream) {
+                                      const float* h0, hipStream_t stream) {
     constexpr int Hp = 128, NT = 256;
     const size_t lds = 2 * (size_t)Tseq * (3 * Hp + 8) + 2 * 2 * (Hp + 8) +
                        4 * 3 * Hp;
@@ -2824,20 +2916,20 @@ extern "C" int fmda_gru_fwd_b1_launch(const void* gi, const void* w,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     k<<<dim3(n_dir), NT, lds, stream>>>(
         (const __hip_bfloat16*)gi, (const __hip_bfloat16*)w, bhh,
-        (__hip_bfloat16*)out, hlast, Tseq, n_dir);
+        (__hip_bfloat16*)out, hlast, Tseq, n_dir, h0);
     return hipGetLastError() == hipSuccess ? 0 : -4;
 }
 
 extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const float* bhh, void* out,
                                    float* hlast, int B, int Tseq, int n_dir,
-                                   hipStream_t stream) {
+                                   const float* h0, hipStream_t stream) {
     const LaunchCfg c = fwd_cfg(is_bf16, Hp);
     const size_t lds = fwd_lds_bytes(is_bf16, Hp, c.bt, c.wlds);
     if (lds > 160 * 1024) return -2;
 #define F(TY, BTV, HPV, WL, NTV, HO)                                           \
     launch_fwd<TY, BTV, HPV, WL, NTV, HO>(gi, w, bhh, out, hlast, B, Tseq,     \
-                                          n_dir, lds, stream)
+                                          n_dir, h0, lds, stream)
     if (is_bf16) {
         switch (Hp) {
             case 16: F(bf16_t, 32, 16, true, 256, false); break;
@@ -2846,10 +2938,10 @@ extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
             case 128:
                 if (B == 1 &&
                     fmda_gru_fwd_b1_launch(gi, w, bhh, out, hlast, Tseq,
-                                           n_dir, stream) == 0)
+                                           n_dir, h0, stream) == 0)
                     break;   // LDS-resident batch-1 kernel took it
                 launch_fwd_v3_128(gi, w, bhh, out, hlast, B, Tseq, n_dir,
-                                  stream);
+                                  h0, stream);
                 break;
             case 256: F(bf16_t, 32, 256, false, 512, false); break;
             case 512: F(bf16_t, 16, 512, false, 512, false); break;
@@ -2903,6 +2995,7 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    int n_dir, unsigned int drop_thr,
                                    float drop_scale,
                                    unsigned long long drop_seed,
+                                   const float* h0, void* dgh0,
                                    hipStream_t stream) {
     if (!is_bf16 && Hp > 256) return -3;
     if (drop_thr != 0u && !(is_bf16 && Hp == 128))
@@ -2913,7 +3006,7 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
 #define G(TY, BTV, HPV, WL, NTV, HO)                                           \
     launch_bwd<TY, BTV, HPV, WL, NTV, HO>(gi, w, bhh, out, dout, dhT, dgi,     \
                                           dgh, dh0, dbhh, B, Tseq, n_dir,      \
-                                          lds, stream)
+                                          h0, dgh0, lds, stream)
     if (is_bf16) {
         switch (Hp) {
             case 16: G(bf16_t, 32, 16, true, 256, false); break;
@@ -2922,7 +3015,7 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
             case 128:
                 launch_bwd_v3_128(gi, w, wt, bhh, out, dout, dhT, dgi, dgh,
                                   dh0, dbhh, B, Tseq, n_dir, drop_thr,
-                                  drop_scale, drop_seed, stream);
+                                  drop_scale, drop_seed, h0, dgh0, stream);
                 break;
             case 256: G(bf16_t, 32, 256, false, 512, false); break;
             case 512: G(bf16_t, 16, 512, false, 512, false); break;

@@ -62,25 +62,30 @@ class _GRURecurrence(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, gi, w, bhh):
+    def forward(ctx, gi, w, bhh, h0=None):
         ext = load_extension()
         gi = gi.contiguous()
         w = w.contiguous()
         bhh32 = bhh.to(torch.float32).contiguous()
-        out, h_last = ext.gru_fwd(gi, w, bhh32)
+        h0c = (h0.detach().to(torch.float32).contiguous()
+               if h0 is not None else None)
+        out, h_last = ext.gru_fwd(gi, w, bhh32, h0c)
         ctx.save_for_backward(gi, w, bhh32, out)
+        ctx.h0 = h0c
         return out, h_last
 
     @staticmethod
     def backward(ctx, d_out, d_hlast):
         ext = load_extension()
         gi, w, bhh32, out = ctx.saved_tensors
+        h0c = ctx.h0
         B, T, _ = gi.shape
         n_dir, threeHp, Hp = w.shape
         d_out = d_out.contiguous().to(gi.dtype)
         d_hlast = d_hlast.contiguous().to(torch.float32)
-        dgi, dgh, _dh0, dbhh, _dbih = ext.gru_bwd(gi, w, bhh32, out, d_out,
-                                                  d_hlast)
+        res = ext.gru_bwd(gi, w, bhh32, out, d_out, d_hlast, 0.0, 0, h0c)
+        dgi, dgh, dh0_out, dbhh, _dbih = res[:5]
+        dgh0 = res[5] if len(res) > 5 else None
 
         # dW_hh[n, k] = sum_{b,t} dGh_shifted[b,t,n] * out[b,t,k]: the
         # kernel stores dGh time-shifted so slot t pairs with out[t] — one
@@ -93,12 +98,16 @@ class _GRURecurrence(torch.autograd.Function):
         for d in range(n_dir):
             dw[d] = cross[d * threeHp:(d + 1) * threeHp,
                           d * Hp:(d + 1) * Hp]
-        return dgi, dw, dbhh
+            if dgh0 is not None:  # t=0 term pairs with h0, not out
+                dw[d] += (dgh0[d].float().t() @ h0c[d]).to(dw.dtype)
+        dh0_grad = dh0_out if h0c is not None else None
+        return dgi, dw, dbhh, dh0_grad
 
 
-def gru_directions(gi: torch.Tensor, w: torch.Tensor, bhh: torch.Tensor
+def gru_directions(gi: torch.Tensor, w: torch.Tensor, bhh: torch.Tensor,
+                   h0: Optional[torch.Tensor] = None
                    ) -> Tuple[torch.Tensor, torch.Tensor]:
-    return _GRURecurrence.apply(gi, w, bhh)
+    return _GRURecurrence.apply(gi, w, bhh, h0)
 
 
 class _IHProjection(torch.autograd.Function):
@@ -257,7 +266,7 @@ class _BiGRULayer(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, Hp, w_ih0, w_hh0, b_ih0, b_hh0,
                 w_ih1, w_hh1, b_ih1, b_hh1, out_drop_p=0.0,
-                out_drop_seed=0):
+                out_drop_seed=0, h0=None):
         ext = load_extension()
         D = 2 if w_ih1 is not None else 1
         H = w_hh0.shape[1]
@@ -279,8 +288,10 @@ class _BiGRULayer(torch.autograd.Function):
             b_hh_cat = torch.stack(b_hhs, dim=0).float()       # (D, 3Hp)
         x2d = x.reshape(B * T, F)
         gi = torch.addmm(b_ih_cat, x2d, w_ih_cat.t()).view(B, T, -1)
-        out, h_last = ext.gru_fwd(gi, w_hh_cat, b_hh_cat)
+        h0c = h0.detach().contiguous() if h0 is not None else None
+        out, h_last = ext.gru_fwd(gi, w_hh_cat, b_hh_cat, h0c)
         ctx.save_for_backward(x2d, w_ih_cat, w_hh_cat, b_hh_cat, gi, out)
+        ctx.h0 = h0c
         ctx.meta = (D, H, Hp, out_drop_p, out_drop_seed)
         return out, h_last
 
@@ -292,13 +303,21 @@ class _BiGRULayer(torch.autograd.Function):
         need_dx = ctx.needs_input_grad[0]
         d_out = d_out.contiguous().to(gi.dtype)
         d_hlast = d_hlast.contiguous().float()
-        dgi, dgh, _dh0, dbhh, dbih = ext.gru_bwd(
-            gi, w_hh_cat, b_hh_cat, out, d_out, d_hlast,
-            drop_p, drop_seed)
+        h0c = ctx.h0
+        res = ext.gru_bwd(gi, w_hh_cat, b_hh_cat, out, d_out, d_hlast,
+                          drop_p, drop_seed, h0c)
+        dgi, dgh, dh0_out, dbhh, dbih = res[:5]
+        dgh0 = res[5] if len(res) > 5 else None
         M = dgi.shape[0] * dgi.shape[1]
 
         # dW_hh via the time-shifted dGh and one split-K reduction
         cross = chunked_outer(dgh.reshape(M, -1), out.reshape(M, -1))
+        if dgh0 is not None:
+            # t=0 term of dW_hh: dGh_0 (x) h0 (boundary slot holds zeros)
+            for d in range(D):
+                cross[d * 3 * Hp:(d + 1) * 3 * Hp,
+                      d * Hp:(d + 1) * Hp] += \
+                    (dgh0[d].float().t() @ h0c[d]).to(cross.dtype)
         # dW_ih for both directions in one split-K reduction
         dwih_cat = chunked_outer(dgi.reshape(M, -1), x2d)
         dx = None
@@ -319,7 +338,9 @@ class _BiGRULayer(torch.autograd.Function):
         if D == 1:
             grads.append((None, None, None, None))
         (a0, b0, c0, e0), (a1, b1, c1, e1) = grads
-        return (dx, None, a0, b0, c0, e0, a1, b1, c1, e1, None, None)
+        dh0_grad = dh0_out if h0c is not None else None
+        return (dx, None, a0, b0, c0, e0, a1, b1, c1, e1, None, None,
+                dh0_grad)
 
 
 
@@ -366,12 +387,22 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
     Weights come from the nn.GRU parameter container (reference state_dict
     layout); masters stay fp32 and are packed/cast to the compute dtype
     inside the single-layer autograd node. Replaces the reference's
-    `self.gru(input_seq)` call (biGRU_model.py:102)."""
-    assert hidden is None, "initial hidden state not supported on GPU path"
+    `self.gru(input_seq)` call (biGRU_model.py:102); `hidden` is the
+    optional (L*D, B, H) initial state of the nn.GRU signature."""
     enable_tunableop()
     D = 2 if bidirectional else 1
     H = gru_module.hidden_size
     Hp = _pad_h(H)
+
+    h0_layers = [None] * n_layers
+    if hidden is not None:
+        B = x.shape[0]
+        assert hidden.shape == (n_layers * D, B, H), \
+            f"hidden must be ({n_layers * D}, {B}, {H})"
+        hid32 = hidden.float()
+        for layer in range(n_layers):
+            h0_layers[layer] = _pad_cols(
+                hid32[layer * D:(layer + 1) * D], H, Hp).contiguous()
 
     h_n_parts: List[torch.Tensor] = []
     inp = x
@@ -393,6 +424,7 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
                 if defer else 0)
         dp = (dropout_p, seed) if defer else (0.0, 0)
 
+        h0_l = h0_layers[layer]
         if not training and not torch.is_grad_enabled():
             # inference fast path: cached packed weights, direct kernel call
             ext = load_extension()
@@ -401,12 +433,14 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
             B, T, F = inp.shape
             gi = torch.addmm(b_ih_cat, inp.reshape(B * T, F),
                              w_ih_cat.t()).view(B, T, -1)
-            out_pad, h_last = ext.gru_fwd(gi, w_hh_cat, b_hh_cat)
+            out_pad, h_last = ext.gru_fwd(gi, w_hh_cat, b_hh_cat, h0_l)
         elif D == 2:
-            out_pad, h_last = _BiGRULayer.apply(inp, Hp, *p[0], *p[1], *dp)
+            out_pad, h_last = _BiGRULayer.apply(inp, Hp, *p[0], *p[1], *dp,
+                                                h0_l)
         else:
             out_pad, h_last = _BiGRULayer.apply(inp, Hp, *p[0],
-                                                None, None, None, None, *dp)
+                                                None, None, None, None, *dp,
+                                                h0_l)
 
         if Hp == H:
             out = out_pad

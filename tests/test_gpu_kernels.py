@@ -110,14 +110,15 @@ def test_gru_fwd_batch1_long_sequence_fallback():
     assert (hlast - hlast_ref).abs().max() < 0.06
 
 
-def _gru_ref_from_gi(gi, w, bhh):
+def _gru_ref_from_gi(gi, w, bhh, h0=None):
     """Golden recurrence on precomputed input projections (fp32 torch)."""
     B, T, _ = gi.shape
     n_dir, threeH, H = w.shape
     out = torch.zeros(B, T, n_dir * H, device=gi.device)
     hlast = torch.zeros(n_dir, B, H, device=gi.device)
     for d in range(n_dir):
-        h = torch.zeros(B, H, device=gi.device)
+        h = (h0[d].clone() if h0 is not None
+             else torch.zeros(B, H, device=gi.device))
         steps = range(T - 1, -1, -1) if d == 1 else range(T)
         for t in steps:
             g = gi[:, t, d * threeH:(d + 1) * threeH]
@@ -166,13 +167,13 @@ def test_gru_backward_fp32_matches_autograd(H, T, B, n_dir):
         assert ((a - b).abs().max() / scale) < 1e-4, (a - b).abs().max()
 
 
-def _gru_ref_from_gi_autograd(gi, w, bhh):
+def _gru_ref_from_gi_autograd(gi, w, bhh, h0=None):
     B, T, _ = gi.shape
     n_dir, threeH, H = w.shape
     outs = []
     hlast = []
     for d in range(n_dir):
-        h = torch.zeros(B, H, device=gi.device)
+        h = h0[d] if h0 is not None else torch.zeros(B, H, device=gi.device)
         hs = [None] * T
         steps = range(T - 1, -1, -1) if d == 1 else range(T)
         for t in steps:
@@ -508,3 +509,118 @@ def test_fused_dropout_statistics_and_backward():
     y.backward(g)
     # backward mask identical to forward mask
     assert torch.equal((x.grad != 0), kept)
+
+
+# ---------------------------------------------------------------------------
+# Initial hidden state (h0) support — nn.GRU signature parity
+# (reference biGRU_model.py:102 `self.gru(input_seq, hidden)`).
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("H,T,B,n_dir", [
+    (16, 7, 5, 2), (32, 12, 33, 2), (128, 20, 32, 2), (64, 9, 4, 1),
+    (256, 8, 16, 2), (512, 6, 16, 2),
+])
+def test_gru_fwd_h0_fp32_matches_reference(H, T, B, n_dir):
+    ext = _ext()
+    torch.manual_seed(11)
+    gi = torch.randn(B, T, n_dir * 3 * H).cuda()
+    w = torch.randn(n_dir, 3 * H, H).cuda() * 0.3
+    bhh = torch.randn(n_dir, 3 * H).cuda() * 0.1
+    h0 = torch.randn(n_dir, B, H).cuda() * 0.5
+    out, hlast = ext.gru_fwd(gi, w, bhh, h0.contiguous())
+    out_ref, hlast_ref = _gru_ref_from_gi(gi, w, bhh, h0)
+    assert torch.allclose(out, out_ref, atol=1e-4), (out - out_ref).abs().max()
+    assert torch.allclose(hlast, hlast_ref, atol=1e-4)
+
+
+@pytest.mark.parametrize("H,T,B", [(128, 20, 32), (32, 9, 7), (512, 5, 8)])
+def test_gru_fwd_h0_bf16(H, T, B):
+    """bf16 paths with h0: v3 (H=128), v2 (H=32), and the batch-parallel
+    fallback the Hp=512 h0 call routes to."""
+    ext = _ext()
+    torch.manual_seed(12)
+    n_dir = 2
+    gi = (torch.randn(B, T, n_dir * 3 * H) * 0.5).cuda()
+    w = (torch.randn(n_dir, 3 * H, H) * 0.2).cuda()
+    bhh = (torch.randn(n_dir, 3 * H) * 0.1).cuda()
+    h0 = (torch.randn(n_dir, B, H) * 0.5).cuda()
+    out, hlast = ext.gru_fwd(gi.bfloat16(), w.bfloat16(), bhh, h0.contiguous())
+    out_ref, hlast_ref = _gru_ref_from_gi(gi, w, bhh, h0)
+    assert (out.float() - out_ref).abs().max() < 0.05
+    assert (hlast - hlast_ref).abs().max() < 0.05
+
+
+def test_gru_fwd_h0_batch1_kernel():
+    """The LDS-resident batch-1 streaming kernel honors h0."""
+    ext = _ext()
+    torch.manual_seed(13)
+    H, T, n_dir = 128, 40, 2
+    gi = (torch.randn(1, T, n_dir * 3 * H) * 0.5).cuda()
+    w = (torch.randn(n_dir, 3 * H, H) * 0.2).cuda()
+    bhh = (torch.randn(n_dir, 3 * H) * 0.1).cuda()
+    h0 = (torch.randn(n_dir, 1, H) * 0.5).cuda()
+    out, hlast = ext.gru_fwd(gi.bfloat16(), w.bfloat16(), bhh, h0.contiguous())
+    out_ref, hlast_ref = _gru_ref_from_gi(gi, w, bhh, h0)
+    assert (out.float() - out_ref).abs().max() < 0.05
+    assert (hlast - hlast_ref).abs().max() < 0.05
+
+
+@pytest.mark.parametrize("H,T,B,n_dir", [
+    (32, 10, 33, 2), (128, 25, 32, 2), (64, 7, 4, 1),
+])
+def test_gru_backward_h0_fp32_matches_autograd(H, T, B, n_dir):
+    """BPTT with a real initial hidden state: dgi/dW/db AND dh0 against
+    torch autograd (the t=0 dGh (x) h0 term of dW_hh is the tricky bit)."""
+    from fmda_amd.ops.interface import gru_directions
+    torch.manual_seed(14)
+    gi = (torch.randn(B, T, n_dir * 3 * H) * 0.5).cuda()
+    w = (torch.randn(n_dir, 3 * H, H) * 0.2).cuda()
+    bhh = (torch.randn(n_dir, 3 * H) * 0.1).cuda()
+    h0 = (torch.randn(n_dir, B, H) * 0.5).cuda()
+    dO = torch.randn(B, T, n_dir * H).cuda()
+    dH = torch.randn(n_dir, B, H).cuda()
+
+    gi1 = gi.clone().requires_grad_(True)
+    w1 = w.clone().requires_grad_(True)
+    b1 = bhh.clone().requires_grad_(True)
+    h01 = h0.clone().requires_grad_(True)
+    out1, hl1 = gru_directions(gi1, w1, b1, h01)
+    (out1 * dO).sum().add_((hl1 * dH).sum()).backward()
+
+    gi2 = gi.clone().requires_grad_(True)
+    w2 = w.clone().requires_grad_(True)
+    b2 = bhh.clone().requires_grad_(True)
+    h02 = h0.clone().requires_grad_(True)
+    out2, hl2 = _gru_ref_from_gi_autograd(gi2, w2, b2, h02)
+    (out2 * dO).sum().add_((hl2 * dH).sum()).backward()
+
+    assert torch.allclose(out1, out2, atol=1e-5)
+    for nm, a, b in [("dgi", gi1.grad, gi2.grad), ("dw", w1.grad, w2.grad),
+                     ("dbhh", b1.grad, b2.grad), ("dh0", h01.grad, h02.grad)]:
+        scale = b.abs().max().clamp(min=1.0)
+        assert ((a - b).abs().max() / scale) < 1e-4, (nm, (a - b).abs().max())
+
+
+def test_bigru_model_h0_gpu_matches_cpu():
+    """Model-level nn.GRU-signature parity: forward(x, hidden) with a
+    nonzero hidden on the HIP engine vs the CPU ATen oracle, plus gradient
+    flow back to the hidden tensor."""
+    from fmda_amd.models import BiGRU
+    torch.manual_seed(15)
+    m = BiGRU(32, 24, 4, n_layers=2, spatial_dropout=False, dropout=0.0)
+    x = torch.randn(6, 15, 24)
+    h0 = torch.randn(2 * 2, 6, 32) * 0.5
+
+    h0_cpu = h0.clone().requires_grad_(True)
+    logits_cpu = m(x, h0_cpu)
+    logits_cpu.sum().backward()
+
+    mg = BiGRU(32, 24, 4, n_layers=2, spatial_dropout=False, dropout=0.0)
+    mg.load_state_dict(m.state_dict())
+    mg = mg.cuda()
+    h0_gpu = h0.clone().cuda().requires_grad_(True)
+    logits_gpu = mg(x.cuda(), h0_gpu)
+    logits_gpu.sum().backward()
+
+    assert (logits_gpu.cpu() - logits_cpu).abs().max() < 1e-3
+    assert (h0_gpu.grad.cpu() - h0_cpu.grad).abs().max() < 1e-3
