@@ -91,10 +91,18 @@ class BiGRU(nn.Module):
         input_length = input_seq.size(1)
 
         if self.spatial_dropout:
-            # Dropout2d over (B, F, T): zeroes whole feature channels.
-            input_seq = input_seq.permute(0, 2, 1)
-            input_seq = self.spatial_dropout1d(input_seq)
-            input_seq = input_seq.permute(0, 2, 1)
+            if (self.training and input_seq.is_cuda
+                    and input_seq.dtype == torch.bfloat16
+                    and self.dropout_p > 0):
+                # channel-mask HIP kernel: Dropout2d semantics without the
+                # permute(0,2,1) round trips (biGRU_model.py:87-94)
+                from ..ops.interface import fused_spatial_dropout
+                input_seq = fused_spatial_dropout(input_seq, self.dropout_p)
+            else:
+                # Dropout2d over (B, F, T): zeroes whole feature channels.
+                input_seq = input_seq.permute(0, 2, 1)
+                input_seq = self.spatial_dropout1d(input_seq)
+                input_seq = input_seq.permute(0, 2, 1)
         elif (self.training and input_seq.is_cuda
               and input_seq.dtype == torch.bfloat16 and self.dropout_p > 0):
             from ..ops.interface import fused_dropout

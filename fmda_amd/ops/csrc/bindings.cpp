@@ -43,6 +43,10 @@ extern "C" int fmda_pool_bwd_launch(int is_bf16, const float* dmax,
 extern "C" int fmda_dropout_launch(int is_bf16, const void* x, void* y,
                                    long n, float p, unsigned long long seed,
                                    hipStream_t stream);
+extern "C" int fmda_spatial_dropout_launch(int is_bf16, const void* x,
+                                           void* y, long B, long Tlen, long F,
+                                           float p, unsigned long long seed,
+                                           hipStream_t stream);
 extern "C" int fmda_head_fwd_launch(int is_bf16, const void* x, const void* W,
                                     const void* bias, const float* y,
                                     const float* wgt, const float* pw,
@@ -355,6 +359,19 @@ torch::Tensor dropout_fused(torch::Tensor x, double p, int64_t seed) {
     return y;
 }
 
+torch::Tensor spatial_dropout_fused(torch::Tensor x, double p,
+                                    int64_t seed) {
+    TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3 &&
+                x.scalar_type() == torch::kBFloat16);
+    auto y = torch::empty_like(x);
+    auto stream = at::hip::getCurrentHIPStream();
+    int rc = fmda_spatial_dropout_launch(
+        1, x.data_ptr(), y.data_ptr(), x.size(0), x.size(1), x.size(2),
+        (float)p, (unsigned long long)seed, stream.stream());
+    TORCH_CHECK(rc == 0, "spatial dropout launch failed");
+    return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gru_fwd", &gru_fwd, "fused biGRU recurrence forward (HIP/CDNA4)",
           py::arg("gi"), py::arg("w"), py::arg("bhh"),
@@ -366,6 +383,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("mfma_selftest", &mfma_selftest, "mfma fragment layout self-test");
     m.def("dropout_fused", &dropout_fused,
           "counter-based dropout (mask recomputed in backward)");
+    m.def("spatial_dropout_fused", &spatial_dropout_fused,
+          "channel (Dropout2d) dropout over (B, T, F) without permutes");
     m.def("pool_fwd", &pool_fwd, "fused dirsum+max/avg pooling forward");
     m.def("pool_bwd", &pool_bwd, "fused pooling backward (d_out assembly)");
     m.def("head_loss_fwd", &head_loss_fwd, "fused head GEMM + BCE loss");

@@ -2452,6 +2452,58 @@ extern "C" int fmda_dropout_launch(int is_bf16, const void* x, void* y,
     return hipGetLastError() == hipSuccess ? 0 : -1;
 }
 
+// Spatial (channel) dropout: the reference's Dropout2d call
+// (biGRU_model.py:50-52,87-94) zeroes whole FEATURE channels — the mask is
+// a function of (b, f) only and is shared by every timestep. Counter-based
+// like dropout_kernel (one splitmix64 draw per channel octet, recomputed in
+// backward), but WITHOUT the permute(0,2,1) round trips the reference needs:
+// the mask is evaluated per element of the natural (B, T, F) layout.
+template <typename T>
+__global__ void spatial_dropout_kernel(const T* __restrict__ x,
+                                       T* __restrict__ y, long n, long TF,
+                                       int F, unsigned int thr, float scale,
+                                       unsigned long long seed) {
+    const long o = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (o >= n) return;
+    const int cnt = (int)((o + 8 <= n) ? 8 : (n - o));
+    T v[8];
+    if (cnt == 8)
+        *(chunk16*)v = *(const chunk16*)(x + o);
+    else
+        for (int k = 0; k < cnt; ++k) v[k] = x[o + k];
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+        if (k >= cnt) break;
+        const long i = o + k;
+        const long c = (i / TF) * F + (i % F);   // channel id (b, f)
+        const unsigned long long r =
+            mix64(seed ^ (unsigned long long)(c >> 3));
+        const unsigned int u = (unsigned int)(r >> (8 * ((int)c & 7))) & 0xFF;
+        v[k] = (u < thr) ? from_f32<T>(0.0f)
+                         : from_f32<T>(to_f32<T>(v[k]) * scale);
+    }
+    if (cnt == 8)
+        *(chunk16*)(y + o) = *(const chunk16*)v;
+    else
+        for (int k = 0; k < cnt; ++k) y[o + k] = v[k];
+}
+
+extern "C" int fmda_spatial_dropout_launch(int is_bf16, const void* x,
+                                           void* y, long B, long Tlen, long F,
+                                           float p, unsigned long long seed,
+                                           hipStream_t stream) {
+    if (!is_bf16) return -2;
+    const float scale = 1.0f / (1.0f - p);
+    const unsigned int thr = (unsigned int)(p * 256.0f);
+    const long n = B * Tlen * F;
+    const long threads = (n + 7) / 8;
+    const dim3 grid((threads + 255) / 256);
+    spatial_dropout_kernel<__hip_bfloat16><<<grid, 256, 0, stream>>>(
+        (const __hip_bfloat16*)x, (__hip_bfloat16*)y, n, Tlen * F, (int)F,
+        thr, scale, seed);
+    return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
 // ===========================================================================
 // Fused direction-sum + temporal max/avg pooling (biGRU_model.py:108-133
 // semantics: gru_out = fwd_dir + bwd_dir; max over T with argmax; sum/T).

@@ -179,6 +179,37 @@ def fused_dropout(x: torch.Tensor, p: float) -> torch.Tensor:
     return _FusedDropout.apply(x, p, seed)
 
 
+class _FusedSpatialDropout(torch.autograd.Function):
+    """Channel (Dropout2d) dropout on (B, T, F): the mask is per (b, f) and
+    shared across timesteps (reference biGRU_model.py:50-52,87-94 semantics
+    without the permute(0,2,1) round trips). Counter-based: backward
+    recomputes the identical mask instead of saving it."""
+
+    @staticmethod
+    def forward(ctx, x, p, seed):
+        ext = load_extension()
+        ctx.p = p
+        ctx.seed = seed
+        return ext.spatial_dropout_fused(x.contiguous(), p, seed)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_extension()
+        return (ext.spatial_dropout_fused(dy.contiguous(), ctx.p, ctx.seed),
+                None, None)
+
+
+def fused_spatial_dropout(x: torch.Tensor, p: float) -> torch.Tensor:
+    """Training-mode spatial dropout on the HIP engine; torch Dropout2d
+    fallback for other dtypes/devices."""
+    if not (x.is_cuda and x.dtype == torch.bfloat16 and 0.0 < p < 1.0):
+        y = torch.nn.functional.dropout2d(
+            x.permute(0, 2, 1).unsqueeze(-1), p=p, training=True)
+        return y.squeeze(-1).permute(0, 2, 1)
+    seed = int(torch.empty((), dtype=torch.int64).random_())
+    return _FusedSpatialDropout.apply(x, p, seed)
+
+
 class _DirSumPool(torch.autograd.Function):
     """Fused direction-sum + max/avg temporal pooling on the HIP engine
     (biGRU_model.py:108-133 semantics). Returns (max (B,H), avg (B,H)) in

@@ -624,3 +624,61 @@ def test_bigru_model_h0_gpu_matches_cpu():
 
     assert (logits_gpu.cpu() - logits_cpu).abs().max() < 1e-3
     assert (h0_gpu.grad.cpu() - h0_cpu.grad).abs().max() < 1e-3
+
+
+# ---------------------------------------------------------------------------
+# Spatial (channel) dropout — reference Dropout2d semantics on the engine
+# (biGRU_model.py:50-52,87-94; the reference API default spatial_dropout=True)
+# ---------------------------------------------------------------------------
+
+def test_spatial_dropout_channel_mask_and_backward():
+    ext = _ext()
+    B, T, F = 8, 33, 108
+    p = 0.4
+    x = torch.ones(B, T, F, device="cuda", dtype=torch.bfloat16)
+    y = ext.spatial_dropout_fused(x, p, 1234)
+    yf = y.float()
+    # channel mask: for each (b, f) the value is identical across ALL t
+    assert (yf == yf[:, :1, :]).all(), "mask varies along time"
+    # kept channels carry exactly the 1/(1-p) scale
+    kept = yf[:, 0, :][yf[:, 0, :] > 0]
+    assert torch.allclose(kept, torch.full_like(kept, 1.0 / (1.0 - p)),
+                          atol=1e-2)
+    # drop fraction near p over B*F channels
+    frac = (yf[:, 0, :] == 0).float().mean().item()
+    assert abs(frac - p) < 0.08, frac
+    # backward recomputes the identical mask: grad of sum(y) w.r.t. x is
+    # the mask * scale
+    from fmda_amd.ops.interface import _FusedSpatialDropout
+    x2 = torch.ones(B, T, F, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y2 = _FusedSpatialDropout.apply(x2, p, 1234)
+    y2.sum().backward()
+    assert (y2.float() == x2.grad.float()).all()  # x==1 -> y == mask*scale
+    # different seeds give different masks
+    y3 = ext.spatial_dropout_fused(x, p, 99)
+    assert not (y3.float() == yf).all()
+
+
+def test_spatial_dropout_model_train_step():
+    """The reference API default (spatial_dropout=True) runs the engine
+    path end-to-end on GPU: forward + backward + fused optimizer."""
+    from fmda_amd.models import BiGRU
+    from fmda_amd.optim import FusedClipAdam
+    torch.manual_seed(21)
+    m = BiGRU(128, 96, 4, n_layers=2, spatial_dropout=True,
+              dropout=0.3).cuda()
+    opt = FusedClipAdam(m.parameters(), lr=1e-3, clip=50.0)
+    x = (torch.randn(16, 30, 96, device="cuda") * 0.5).bfloat16()
+    y = (torch.rand(16, 4, device="cuda") < 0.3).float()
+    m.train()
+    for _ in range(2):
+        opt.zero_grad(set_to_none=True)
+        logits = m(x)
+        loss = torch.nn.functional.binary_cross_entropy_with_logits(
+            logits.float(), y)
+        loss.backward()
+        opt.step()
+    torch.cuda.synchronize()
+    for prm in m.parameters():
+        assert torch.isfinite(prm).all()
