@@ -7,7 +7,10 @@
 extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const float* bhh, void* out,
                                    float* hlast, int B, int Tseq, int n_dir,
-                                   const float* h0, hipStream_t stream);
+                                   const float* h0, void* out_drop,
+                                   unsigned int drop_thr, float drop_scale,
+                                   unsigned long long drop_seed,
+                                   hipStream_t stream);
 extern "C" int fmda_gru_fwd_cs_launch(const void* gi, const void* w,
                                       const float* bhh, void* out,
                                       float* hlast, void* hpub,
@@ -101,7 +104,8 @@ void check_common(const torch::Tensor& gi, const torch::Tensor& w,
 
 std::vector<torch::Tensor> gru_fwd(torch::Tensor gi, torch::Tensor w,
                                    torch::Tensor bhh,
-                                   c10::optional<torch::Tensor> h0_opt) {
+                                   c10::optional<torch::Tensor> h0_opt,
+                                   double drop_p, int64_t drop_seed) {
     int B, T, n_dir, Hp;
     bool is_bf16;
     check_common(gi, w, bhh, B, T, n_dir, Hp, is_bf16);
@@ -117,6 +121,21 @@ std::vector<torch::Tensor> gru_fwd(torch::Tensor gi, torch::Tensor w,
     auto out = torch::empty({B, T, (int64_t)n_dir * Hp}, gi.options());
     auto hlast = torch::empty({n_dir, B, Hp},
                               gi.options().dtype(torch::kFloat32));
+    // fused forward dropout: emit out AND the dropped copy (the layer
+    // above's input) from the store epilogue, replacing the separate
+    // full-tensor dropout pass. Quantized exactly like dropout_kernel.
+    const unsigned int drop_thr =
+        (drop_p > 0.0) ? (unsigned int)((float)drop_p * 256.0f) : 0u;
+    const float drop_scale =
+        (drop_p > 0.0) ? (float)(1.0 / (1.0 - drop_p)) : 1.0f;
+    torch::Tensor out_drop;
+    void* out_drop_ptr = nullptr;
+    if (drop_thr != 0u) {
+        TORCH_CHECK(is_bf16 && Hp == 128,
+                    "fused fwd dropout requires bf16 Hp=128");
+        out_drop = torch::empty_like(out);
+        out_drop_ptr = out_drop.data_ptr();
+    }
     auto stream = at::hip::getCurrentHIPStream();
     int rc;
     // the column-split Hp=512 kernel pre-publishes h_{-1}=0; with a real
@@ -137,9 +156,14 @@ std::vector<torch::Tensor> gru_fwd(torch::Tensor gi, torch::Tensor w,
         rc = fmda_gru_fwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
                                  w.data_ptr(), bhh.data_ptr<float>(),
                                  out.data_ptr(), hlast.data_ptr<float>(), B, T,
-                                 n_dir, h0, stream.stream());
+                                 n_dir, h0, out_drop_ptr, drop_thr,
+                                 drop_scale,
+                                 (unsigned long long)drop_seed,
+                                 stream.stream());
     }
     TORCH_CHECK(rc == 0, "fmda gru_fwd launch failed rc=", rc, " Hp=", Hp);
+    if (out_drop_ptr != nullptr)
+        return {out, hlast, out_drop};
     return {out, hlast};
 }
 
@@ -375,7 +399,8 @@ torch::Tensor spatial_dropout_fused(torch::Tensor x, double p,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gru_fwd", &gru_fwd, "fused biGRU recurrence forward (HIP/CDNA4)",
           py::arg("gi"), py::arg("w"), py::arg("bhh"),
-          py::arg("h0") = py::none());
+          py::arg("h0") = py::none(), py::arg("drop_p") = 0.0,
+          py::arg("drop_seed") = 0);
     m.def("gru_bwd", &gru_bwd, "fused biGRU recurrence backward (HIP/CDNA4)",
           py::arg("gi"), py::arg("w"), py::arg("bhh"), py::arg("out"),
           py::arg("dout"), py::arg("dhT"), py::arg("drop_p") = 0.0,

@@ -896,7 +896,10 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
                        const float* __restrict__ bhh,
                        __hip_bfloat16* __restrict__ out,
                        float* __restrict__ hlast, int B, int Tseq,
-                       int n_dir, const float* __restrict__ h0) {
+                       int n_dir, const float* __restrict__ h0,
+                       __hip_bfloat16* __restrict__ out_drop,
+                       unsigned int drop_thr, float drop_scale,
+                       unsigned long long drop_seed) {
     constexpr int NW = NT / 64;
     constexpr int MT = BT / 16;
     constexpr int NCT = Hp / 16;
@@ -935,6 +938,39 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
     const long out_row = (long)Tseq * n_dir * Hp;
     const __hip_bfloat16* gi_b = gi + (long)b0 * gi_row + (long)dir * 3 * Hp;
     __hip_bfloat16* out_b = out + (long)b0 * out_row + (long)dir * Hp;
+    __hip_bfloat16* outd_b =
+        (out_drop != nullptr)
+            ? out_drop + (long)b0 * out_row + (long)dir * Hp : nullptr;
+
+    // Fused forward half of the inter-layer dropout (the backward half
+    // lives in gru_bwd_v3_kernel's d_out read): the same counter-based
+    // octet mask as dropout_kernel over the (B, T, n_dir*Hp) `out` layout,
+    // applied while the data is still in LDS — the separate full-tensor
+    // dropout pass (1 read + 1 write of ~0.5 GB at the flagship config)
+    // becomes one extra store stream here.
+    auto store_dropped = [&](const __hip_bfloat16* hb, int ttp) {
+        constexpr int CPR = (Hp * 2) / 16;    // 16-byte chunks per row
+        __hip_bfloat16* dst = outd_b + (long)ttp * n_dir * Hp;
+        for (int c = tid; c < BT * CPR; c += NT) {
+            const int r = c / CPR, jc = c % CPR;
+            if (r >= rows_valid) continue;
+            bf16x8_t v = *(const bf16x8_t*)((const char*)hb +
+                                            (long)r * WP * 2 + jc * 16);
+            const long o = ((long)(b0 + r) * Tseq + ttp) * n_dir * Hp +
+                           (long)dir * Hp + jc * 8;   // octet-aligned
+            const unsigned long long rnd =
+                mix64(drop_seed ^ (unsigned long long)(o >> 3));
+#pragma unroll
+            for (int k = 0; k < 8; ++k) {
+                const unsigned int u = (unsigned int)(rnd >> (8 * k)) & 0xFF;
+                const float f =
+                    (u < drop_thr)
+                        ? 0.0f : (float)((__bf16*)&v)[k] * drop_scale;
+                ((__bf16*)&v)[k] = (__bf16)__float2bfloat16(f);
+            }
+            *(bf16x8_t*)((char*)dst + (long)r * out_row * 2 + jc * 16) = v;
+        }
+    };
 
     for (int c = tid; c < 3 * Hp; c += NT)
         bhh_s[c] = bhh[(long)dir * 3 * Hp + c];
@@ -1011,6 +1047,7 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
             store_tile<__hip_bfloat16, BT, NT>(
                 hb_buf(pb), out_b + (long)ttp * n_dir * Hp, Hp, WP, out_row,
                 rows_valid, tid);
+            if (outd_b != nullptr) store_dropped(hb_buf(pb), ttp);
         }
 
         f32x4_t acc[CPW][3][MT];
@@ -1076,12 +1113,27 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
             }
         }
         // rendezvous: drain this step's glds (tile u+1) across all waves,
-        // leaving the out[u-1] store in flight (it was issued after the
-        // glds, so the counted wait never stalls on store completion).
-        if (u > 0)
-            asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
-        else
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        // leaving the out[u-1] store (and, when fused dropout is on, the
+        // out_drop[u-1] store — one more instruction per thread) in
+        // flight: both were issued after the glds, so the counted wait
+        // never stalls on store completion. The leave-count is PER-THREAD
+        // exact: a batch-tail thread whose row is dead issued no store, so
+        // leaving 1 outstanding there would leave a gi glds un-drained.
+        {
+            // vmcnt is per-WAVE: each store pass issues one instruction per
+            // wave iff any of the wave's rows is live (first row suffices:
+            // rows are wave-contiguous).
+            constexpr int RPW = 64 / ((Hp * 2) / 16);   // rows per wave
+            const bool wave_stored = (u > 0) && (wave * RPW < rows_valid);
+            const int leave =
+                wave_stored ? (outd_b != nullptr ? 2 : 1) : 0;
+            if (leave == 2)
+                asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+            else if (leave == 1)
+                asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
     }
@@ -1092,6 +1144,7 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
         store_tile<__hip_bfloat16, BT, NT>(
             hb_buf(pl), out_b + (long)ttl * n_dir * Hp, Hp, WP, out_row,
             rows_valid, tid);
+        if (outd_b != nullptr) store_dropped(hb_buf(pl), ttl);
         float* hl = hlast + ((long)dir * B + b0) * Hp;
 #pragma unroll
         for (int i = 0; i < CPW; ++i) {
@@ -2897,7 +2950,9 @@ using bf16_t = __hip_bfloat16;
 // v3 launch helpers (bf16 Hp=128).
 static int launch_fwd_v3_128(const void* gi, const void* w, const float* bhh,
                              void* out, float* hlast, int B, int Tseq,
-                             int n_dir, const float* h0,
+                             int n_dir, const float* h0, void* out_drop,
+                             unsigned int drop_thr, float drop_scale,
+                             unsigned long long drop_seed,
                              hipStream_t stream) {
     static const bool big = getenv("FMDA_FWD_BT32") != nullptr;
     if (big) {   // A/B: one 8-wave block per CU instead of two 4-wave
@@ -2911,7 +2966,8 @@ static int launch_fwd_v3_128(const void* gi, const void* w, const float* bhh,
         k2<<<grid, NT, lds, stream>>>((const __hip_bfloat16*)gi,
                                       (const __hip_bfloat16*)w, bhh,
                                       (__hip_bfloat16*)out, hlast, B, Tseq,
-                                      n_dir, h0);
+                                      n_dir, h0, (__hip_bfloat16*)out_drop,
+                                      drop_thr, drop_scale, drop_seed);
         return 0;
     }
     constexpr int BT = 16, Hp = 128, NT = 256;
@@ -2924,7 +2980,8 @@ static int launch_fwd_v3_128(const void* gi, const void* w, const float* bhh,
     k<<<grid, NT, lds, stream>>>((const __hip_bfloat16*)gi,
                                  (const __hip_bfloat16*)w, bhh,
                                  (__hip_bfloat16*)out, hlast, B, Tseq, n_dir,
-                                 h0);
+                                 h0, (__hip_bfloat16*)out_drop, drop_thr,
+                                 drop_scale, drop_seed);
     return 0;
 }
 
@@ -2975,7 +3032,12 @@ extern "C" int fmda_gru_fwd_b1_launch(const void* gi, const void* w,
 extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const float* bhh, void* out,
                                    float* hlast, int B, int Tseq, int n_dir,
-                                   const float* h0, hipStream_t stream) {
+                                   const float* h0, void* out_drop,
+                                   unsigned int drop_thr, float drop_scale,
+                                   unsigned long long drop_seed,
+                                   hipStream_t stream) {
+    if (out_drop != nullptr && !(is_bf16 && Hp == 128))
+        return -6;   // fused fwd dropout only on the bf16 Hp=128 path
     const LaunchCfg c = fwd_cfg(is_bf16, Hp);
     const size_t lds = fwd_lds_bytes(is_bf16, Hp, c.bt, c.wlds);
     if (lds > 160 * 1024) return -2;
@@ -2988,12 +3050,13 @@ extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
             case 32: F(bf16_t, 32, 32, true, 256, false); break;
             case 64: F(bf16_t, 32, 64, true, 256, false); break;
             case 128:
-                if (B == 1 &&
+                if (B == 1 && out_drop == nullptr &&
                     fmda_gru_fwd_b1_launch(gi, w, bhh, out, hlast, Tseq,
                                            n_dir, h0, stream) == 0)
                     break;   // LDS-resident batch-1 kernel took it
                 launch_fwd_v3_128(gi, w, bhh, out, hlast, B, Tseq, n_dir,
-                                  h0, stream);
+                                  h0, out_drop, drop_thr, drop_scale,
+                                  drop_seed, stream);
                 break;
             case 256: F(bf16_t, 32, 256, false, 512, false); break;
             case 512: F(bf16_t, 16, 512, false, 512, false); break;

@@ -320,18 +320,38 @@ class _BiGRULayer(torch.autograd.Function):
         x2d = x.reshape(B * T, F)
         gi = torch.addmm(b_ih_cat, x2d, w_ih_cat.t()).view(B, T, -1)
         h0c = h0.detach().contiguous() if h0 is not None else None
-        out, h_last = ext.gru_fwd(gi, w_hh_cat, b_hh_cat, h0c)
+        # out_drop_p > 0: the fwd kernel emits out AND the dropped copy
+        # (the next layer's input) from its store epilogue — no separate
+        # dropout pass; the mask is recomputed in the BPTT kernel's d_out
+        # read (backward half of the fusion).
+        res = ext.gru_fwd(gi, w_hh_cat, b_hh_cat, h0c, out_drop_p,
+                          out_drop_seed)
+        out, h_last = res[0], res[1]
         ctx.save_for_backward(x2d, w_ih_cat, w_hh_cat, b_hh_cat, gi, out)
         ctx.h0 = h0c
         ctx.meta = (D, H, Hp, out_drop_p, out_drop_seed)
+        ctx.set_materialize_grads(False)
+        if len(res) > 2:
+            return out, h_last, res[2]
         return out, h_last
 
     @staticmethod
-    def backward(ctx, d_out, d_hlast):
+    def backward(ctx, d_out, d_hlast, d_outdrop=None):
         ext = load_extension()
         x2d, w_ih_cat, w_hh_cat, b_hh_cat, gi, out = ctx.saved_tensors
         D, H, Hp, drop_p, drop_seed = ctx.meta
         need_dx = ctx.needs_input_grad[0]
+        if drop_p > 0:
+            # the raw `out` output feeds nothing downstream when the fused
+            # dropout path is active (h_n comes from h_last; the layer
+            # above consumes the DROPPED output) — its grad must be empty
+            assert d_out is None, "unexpected grad on raw out with fused dropout"
+            d_out = d_outdrop
+        if d_out is None:
+            d_out = torch.zeros_like(out)
+        if d_hlast is None:
+            d_hlast = out.new_zeros(
+                (w_hh_cat.shape[0], out.shape[0], Hp), dtype=torch.float32)
         d_out = d_out.contiguous().to(gi.dtype)
         d_hlast = d_hlast.contiguous().float()
         h0c = ctx.h0
@@ -456,6 +476,7 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
         dp = (dropout_p, seed) if defer else (0.0, 0)
 
         h0_l = h0_layers[layer]
+        dropped_pad = None
         if not training and not torch.is_grad_enabled():
             # inference fast path: cached packed weights, direct kernel call
             ext = load_extension()
@@ -466,12 +487,14 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
                              w_ih_cat.t()).view(B, T, -1)
             out_pad, h_last = ext.gru_fwd(gi, w_hh_cat, b_hh_cat, h0_l)
         elif D == 2:
-            out_pad, h_last = _BiGRULayer.apply(inp, Hp, *p[0], *p[1], *dp,
-                                                h0_l)
+            res = _BiGRULayer.apply(inp, Hp, *p[0], *p[1], *dp, h0_l)
+            out_pad, h_last = res[0], res[1]
+            dropped_pad = res[2] if len(res) > 2 else None
         else:
-            out_pad, h_last = _BiGRULayer.apply(inp, Hp, *p[0],
-                                                None, None, None, None, *dp,
-                                                h0_l)
+            res = _BiGRULayer.apply(inp, Hp, *p[0],
+                                    None, None, None, None, *dp, h0_l)
+            out_pad, h_last = res[0], res[1]
+            dropped_pad = res[2] if len(res) > 2 else None
 
         if Hp == H:
             out = out_pad
@@ -485,7 +508,8 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
         inp = out
         if drop_here:
             if defer:
-                inp = _DeferredDropout.apply(inp, dropout_p, seed)
+                # fused: the fwd kernel already emitted the dropped copy
+                inp = dropped_pad
             else:
                 inp = fused_dropout(inp, dropout_p)
 

@@ -428,8 +428,10 @@ def test_deferred_dropout_backward_matches_explicit():
     def run(deferred):
         ms = [m.detach().clone().requires_grad_(True) for m in masters]
         if deferred:
-            out, hl = _BiGRULayer.apply(x, H, *ms[:4], *ms[4:], p_drop, seed)
-            dropped = _DeferredDropout.apply(out, p_drop, seed)
+            # fully fused: fwd kernel emits the dropped copy, BPTT kernel
+            # recomputes the mask at its d_out read
+            out, hl, dropped = _BiGRULayer.apply(x, H, *ms[:4], *ms[4:],
+                                                 p_drop, seed)
         else:
             out, hl = _BiGRULayer.apply(x, H, *ms[:4], *ms[4:], 0.0, 0)
             dropped = _FusedDropout.apply(out, p_drop, seed)
