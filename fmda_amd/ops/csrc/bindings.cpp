@@ -50,6 +50,17 @@ extern "C" int fmda_spatial_dropout_launch(int is_bf16, const void* x,
                                            void* y, long B, long Tlen, long F,
                                            float p, unsigned long long seed,
                                            hipStream_t stream);
+extern "C" int fmda_ingest_row_launch(void* ring, const float* row,
+                                      const float* xmin, const float* xrng,
+                                      int Tseq, int F, hipStream_t stream);
+extern "C" int fmda_pool_concat_launch(int is_bf16, const void* out,
+                                       const float* hlast, void* feat, int B,
+                                       int Tseq, int H, int Hp, int n_dir,
+                                       hipStream_t stream);
+extern "C" int fmda_head_sigmoid_launch(int is_bf16, const void* x,
+                                        const void* W, const void* bias,
+                                        float* probs, int B, int K, int C,
+                                        hipStream_t stream);
 extern "C" int fmda_head_fwd_launch(int is_bf16, const void* x, const void* W,
                                     const void* bias, const float* y,
                                     const float* wgt, const float* pw,
@@ -396,6 +407,58 @@ torch::Tensor spatial_dropout_fused(torch::Tensor x, double p,
     return y;
 }
 
+void ingest_row(torch::Tensor ring, torch::Tensor row, torch::Tensor xmin,
+                torch::Tensor xrng) {
+    TORCH_CHECK(ring.is_cuda() && ring.is_contiguous() && ring.dim() == 2 &&
+                ring.scalar_type() == torch::kBFloat16,
+                "ring must be (T, F) bf16 on GPU");
+    TORCH_CHECK(row.is_cuda() && row.is_contiguous() &&
+                row.scalar_type() == torch::kFloat32 &&
+                row.numel() == ring.size(1));
+    auto stream = at::hip::getCurrentHIPStream();
+    int rc = fmda_ingest_row_launch(
+        ring.data_ptr(), row.data_ptr<float>(), xmin.data_ptr<float>(),
+        xrng.data_ptr<float>(), ring.size(0), ring.size(1), stream.stream());
+    TORCH_CHECK(rc == 0, "ingest_row launch failed rc=", rc);
+}
+
+torch::Tensor pool_concat_infer(torch::Tensor out, torch::Tensor hlast,
+                                int64_t H) {
+    TORCH_CHECK(out.is_cuda() && out.is_contiguous() && out.dim() == 3);
+    TORCH_CHECK(hlast.is_contiguous() &&
+                hlast.scalar_type() == torch::kFloat32);
+    const bool bf16 = out.scalar_type() == torch::kBFloat16;
+    const int B = out.size(0), T = out.size(1);
+    const int n_dir = hlast.size(0);
+    const int Hp = hlast.size(2);
+    auto feat = torch::empty({B, 3 * H}, out.options());
+    auto stream = at::hip::getCurrentHIPStream();
+    int rc = fmda_pool_concat_launch(bf16 ? 1 : 0, out.data_ptr(),
+                                     hlast.data_ptr<float>(), feat.data_ptr(),
+                                     B, T, (int)H, Hp, n_dir,
+                                     stream.stream());
+    TORCH_CHECK(rc == 0, "pool_concat launch failed");
+    return feat;
+}
+
+torch::Tensor head_sigmoid(torch::Tensor x, torch::Tensor W,
+                           torch::Tensor b) {
+    TORCH_CHECK(x.is_cuda() && x.is_contiguous() && W.is_contiguous() &&
+                b.is_contiguous());
+    TORCH_CHECK(x.scalar_type() == W.scalar_type() &&
+                x.scalar_type() == b.scalar_type());
+    const bool bf16 = x.scalar_type() == torch::kBFloat16;
+    const int B = x.size(0), K = x.size(1), C = W.size(0);
+    auto probs = torch::empty({B, C}, x.options().dtype(torch::kFloat32));
+    auto stream = at::hip::getCurrentHIPStream();
+    int rc = fmda_head_sigmoid_launch(bf16 ? 1 : 0, x.data_ptr(),
+                                      W.data_ptr(), b.data_ptr(),
+                                      probs.data_ptr<float>(), B, K, C,
+                                      stream.stream());
+    TORCH_CHECK(rc == 0, "head_sigmoid launch failed");
+    return probs;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gru_fwd", &gru_fwd, "fused biGRU recurrence forward (HIP/CDNA4)",
           py::arg("gi"), py::arg("w"), py::arg("bhh"),
@@ -410,6 +473,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "counter-based dropout (mask recomputed in backward)");
     m.def("spatial_dropout_fused", &spatial_dropout_fused,
           "channel (Dropout2d) dropout over (B, T, F) without permutes");
+    m.def("ingest_row", &ingest_row,
+          "streaming ingest: shift GPU window ring + normalize new row");
+    m.def("pool_concat_infer", &pool_concat_infer,
+          "fused [dirsum(h_last) | max | avg] concat for inference");
+    m.def("head_sigmoid", &head_sigmoid,
+          "linear head + sigmoid emitting probabilities");
     m.def("pool_fwd", &pool_fwd, "fused dirsum+max/avg pooling forward");
     m.def("pool_bwd", &pool_bwd, "fused pooling backward (d_out assembly)");
     m.def("head_loss_fwd", &head_loss_fwd, "fused head GEMM + BCE loss");

@@ -1663,9 +1663,12 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
         }
         // single per-step barrier: publishes dgh_s[q] for phase C AND
         // rendezvouses this step's DMA (tiles for step u-1, read at
-        // B(u-1)), leaving the trailing dGi/dGh stores in flight.
+        // B(u-1)), leaving the trailing dGi/dGh stores in flight. The
+        // counted form assumes every thread issued its full store quota;
+        // in the (single) batch-tail block dead rows skip stores, so that
+        // block full-drains instead — exactness over speed there.
         if (have_next) {
-            if (have_prev)
+            if (have_prev && rows_valid == BT)
                 asm volatile("s_waitcnt vmcnt(%0)" ::"i"(NST) : "memory");
             else
                 asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -2718,6 +2721,160 @@ extern "C" int fmda_pool_bwd_launch(int is_bf16, const float* dmax,
 }
 
 // ===========================================================================
+// Streaming-inference fused kernels (the predict.py path, batch = 1;
+// reference predict.py:124-197). The window ring lives ON the GPU: per
+// tick the host uploads one raw fp32 feature row (384 B) and the captured
+// graph runs [ingest -> gi GEMM -> b1 recurrence] x L -> pool_concat ->
+// head_sigmoid — no CPU normalize, no full-window H2D, no eager glue.
+// ===========================================================================
+
+// Shift the GPU-resident window one row up and append the normalized new
+// row: ring[t] = ring[t+1]; ring[T-1][f] = (row[f] - xmin[f]) / xrng[f]
+// (predict.py:175 normalize + the SQL last-N-rows fetch semantics).
+// One workgroup; read-everything -> barrier -> write-everything makes the
+// in-place shift race-free. 48 chunks/thread x 1024 threads bounds T*F.
+__global__ void ingest_row_kernel(__hip_bfloat16* __restrict__ ring,
+                                  const float* __restrict__ row,
+                                  const float* __restrict__ xmin,
+                                  const float* __restrict__ xrng,
+                                  int Tseq, int F) {
+    const int n = (Tseq - 1) * F;
+    const int tid = threadIdx.x;
+    __hip_bfloat16 keep[48];
+    int cnt = 0;
+    for (int i = tid; i < n; i += blockDim.x) keep[cnt++] = ring[i + F];
+    __syncthreads();
+    cnt = 0;
+    for (int i = tid; i < n; i += blockDim.x) ring[i] = keep[cnt++];
+    for (int f = tid; f < F; f += blockDim.x)
+        ring[n + f] = __float2bfloat16((row[f] - xmin[f]) / xrng[f]);
+}
+
+extern "C" int fmda_ingest_row_launch(void* ring, const float* row,
+                                      const float* xmin, const float* xrng,
+                                      int Tseq, int F, hipStream_t stream) {
+    const int n = (Tseq - 1) * F;
+    if (n > 48 * 1024) return -5;   // window too large for one workgroup
+    ingest_row_kernel<<<1, 1024, 0, stream>>>(
+        (__hip_bfloat16*)ring, row, xmin, xrng, Tseq, F);
+    return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
+// Fused 3-way pooling head for inference: feat = concat[dirsum(h_last),
+// maxpool_T, avgpool_T] (biGRU_model.py:108-133) in ONE kernel — replaces
+// pool + hidden view/sum + cat (+ their launch latency) inside the
+// captured predict graph. Wave per (b, h-pair), lanes strided over T.
+template <typename T>
+__global__ void pool_concat_infer_kernel(const T* __restrict__ out,
+                                         const float* __restrict__ hlast,
+                                         T* __restrict__ feat, int B,
+                                         int Tseq, int H, int Hp, int n_dir) {
+    const int HP2 = H / 2;
+    const int wave = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const int idx = blockIdx.x * (blockDim.x >> 6) + wave;
+    if (idx >= B * HP2) return;
+    const int b = idx / HP2;
+    const int h2 = (idx % HP2) * 2;
+    const int HD = n_dir * H;
+    const T* p = out + (long)b * Tseq * HD + h2;
+    float mx0 = -3.4e38f, mx1 = -3.4e38f, s0 = 0.0f, s1 = 0.0f;
+    for (int t = lane; t < Tseq; t += 64) {
+        const T* pt = p + (long)t * HD;
+        struct alignas(2 * sizeof(T)) Pair { T x, y; };
+        const Pair v2 = *(const Pair*)pt;
+        float v0 = to_f32<T>(v2.x), v1 = to_f32<T>(v2.y);
+        if (n_dir == 2) {
+            const Pair w2 = *(const Pair*)(pt + H);
+            v0 += to_f32<T>(w2.x);
+            v1 += to_f32<T>(w2.y);
+        }
+        s0 += v0; s1 += v1;
+        if (v0 > mx0) mx0 = v0;
+        if (v1 > mx1) mx1 = v1;
+    }
+#pragma unroll
+    for (int d = 1; d < 64; d <<= 1) {
+        const float omx0 = __shfl_xor(mx0, d), omx1 = __shfl_xor(mx1, d);
+        s0 += __shfl_xor(s0, d);
+        s1 += __shfl_xor(s1, d);
+        if (omx0 > mx0) mx0 = omx0;
+        if (omx1 > mx1) mx1 = omx1;
+    }
+    if (lane == 0) {
+        float hl0 = hlast[(long)b * Hp + h2];
+        float hl1 = hlast[(long)b * Hp + h2 + 1];
+        if (n_dir == 2) {
+            hl0 += hlast[((long)B + b) * Hp + h2];
+            hl1 += hlast[((long)B + b) * Hp + h2 + 1];
+        }
+        T* fr = feat + (long)b * 3 * H;
+        fr[h2] = from_f32<T>(hl0);
+        fr[h2 + 1] = from_f32<T>(hl1);
+        fr[H + h2] = from_f32<T>(mx0);
+        fr[H + h2 + 1] = from_f32<T>(mx1);
+        fr[2 * H + h2] = from_f32<T>(s0 / (float)Tseq);
+        fr[2 * H + h2 + 1] = from_f32<T>(s1 / (float)Tseq);
+    }
+}
+
+extern "C" int fmda_pool_concat_launch(int is_bf16, const void* out,
+                                       const float* hlast, void* feat, int B,
+                                       int Tseq, int H, int Hp, int n_dir,
+                                       hipStream_t stream) {
+    const int n = B * (H / 2);
+    const dim3 grid((n + 3) / 4);
+    if (is_bf16)
+        pool_concat_infer_kernel<__hip_bfloat16><<<grid, 256, 0, stream>>>(
+            (const __hip_bfloat16*)out, hlast, (__hip_bfloat16*)feat, B,
+            Tseq, H, Hp, n_dir);
+    else
+        pool_concat_infer_kernel<float><<<grid, 256, 0, stream>>>(
+            (const float*)out, hlast, (float*)feat, B, Tseq, H, Hp, n_dir);
+    return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
+// logits = x @ W^T + b -> sigmoid, emitted directly as probabilities
+// (predict.py:178-186 forward + sigmoid). Wave per (b, c), lanes over K.
+template <typename T>
+__global__ void head_sigmoid_kernel(const T* __restrict__ x,
+                                    const T* __restrict__ W,
+                                    const T* __restrict__ bias,
+                                    float* __restrict__ probs, int B, int K,
+                                    int C) {
+    const int wave = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const int idx = blockIdx.x * (blockDim.x >> 6) + wave;
+    if (idx >= B * C) return;
+    const int b = idx / C, c = idx % C;
+    const T* xr = x + (long)b * K;
+    const T* wr = W + (long)c * K;
+    float acc = 0.0f;
+    for (int k = lane; k < K; k += 64)
+        acc += to_f32<T>(xr[k]) * to_f32<T>(wr[k]);
+#pragma unroll
+    for (int d = 1; d < 64; d <<= 1) acc += __shfl_xor(acc, d);
+    if (lane == 0) probs[idx] = sigmoidf(acc + to_f32<T>(bias[c]));
+}
+
+extern "C" int fmda_head_sigmoid_launch(int is_bf16, const void* x,
+                                        const void* W, const void* bias,
+                                        float* probs, int B, int K, int C,
+                                        hipStream_t stream) {
+    const int n = B * C;
+    const dim3 grid((n + 3) / 4);
+    if (is_bf16)
+        head_sigmoid_kernel<__hip_bfloat16><<<grid, 256, 0, stream>>>(
+            (const __hip_bfloat16*)x, (const __hip_bfloat16*)W,
+            (const __hip_bfloat16*)bias, probs, B, K, C);
+    else
+        head_sigmoid_kernel<float><<<grid, 256, 0, stream>>>(
+            (const float*)x, (const float*)W, (const float*)bias, probs, B,
+            K, C);
+    return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
+// ===========================================================================
 // Fused classifier head + BCEWithLogitsLoss (weight, pos_weight).
 // Replaces nn.Linear (biGRU_model.py:137) + the notebook's
 // BCEWithLogitsLoss(weight, pos_weight) (cell 29) forward/backward chain
@@ -2985,16 +3142,18 @@ static int launch_fwd_v3_128(const void* gi, const void* w, const float* bhh,
     return 0;
 }
 
-static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
-                             const float* bhh,
-                             const void* out, const void* dout,
-                             const float* dhT, void* dgi, void* dgh,
-                             float* dh0, float* dbhh, int B, int Tseq,
-                             int n_dir, unsigned int drop_thr,
-                             float drop_scale, unsigned long long drop_seed,
-                             const float* h0, void* dgh0,
-                             hipStream_t stream) {
-    constexpr int BT = 32, Hp = 128, NT = 512;
+template <int BT, int NT>
+static int launch_bwd_v3_128_t(const void* gi, const void* w, const void* wt,
+                               const float* bhh,
+                               const void* out, const void* dout,
+                               const float* dhT, void* dgi, void* dgh,
+                               float* dh0, float* dbhh, int B, int Tseq,
+                               int n_dir, unsigned int drop_thr,
+                               float drop_scale,
+                               unsigned long long drop_seed,
+                               const float* h0, void* dgh0,
+                               hipStream_t stream) {
+    constexpr int Hp = 128;
     const size_t lds = 3 * 2 * BT * 3 * Hp + 2 * 2 * BT * Hp +
                        2 * 2 * BT * Hp + 2 * 2 * BT * (3 * Hp + 8) +
                        4 * 3 * Hp;
@@ -3010,6 +3169,29 @@ static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
         n_dir, drop_thr, drop_scale, drop_seed, h0,
         (__hip_bfloat16*)dgh0);
     return 0;
+}
+
+static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
+                             const float* bhh,
+                             const void* out, const void* dout,
+                             const float* dhT, void* dgi, void* dgh,
+                             float* dh0, float* dbhh, int B, int Tseq,
+                             int n_dir, unsigned int drop_thr,
+                             float drop_scale, unsigned long long drop_seed,
+                             const float* h0, void* dgh0,
+                             hipStream_t stream) {
+    // A/B: BT=16/NT=256 runs TWO blocks per CU (like the forward) — two
+    // independent serial chains hide each other's latencies; MT=1 halves
+    // the per-lane accumulator footprint. ~80 KB dynamic LDS per block,
+    // 2x just fits the 160 KB CU budget.
+    static const bool small = getenv("FMDA_BWD_BT16") != nullptr;
+    if (small)
+        return launch_bwd_v3_128_t<16, 256>(
+            gi, w, wt, bhh, out, dout, dhT, dgi, dgh, dh0, dbhh, B, Tseq,
+            n_dir, drop_thr, drop_scale, drop_seed, h0, dgh0, stream);
+    return launch_bwd_v3_128_t<32, 512>(
+        gi, w, wt, bhh, out, dout, dhT, dgi, dgh, dh0, dbhh, B, Tseq,
+        n_dir, drop_thr, drop_scale, drop_seed, h0, dgh0, stream);
 }
 
 extern "C" int fmda_gru_fwd_b1_launch(const void* gi, const void* w,

@@ -11,9 +11,13 @@ Semantics of predict.py:124-197, MI355X-native:
 - forward -> sigmoid -> threshold 0.5 -> publish labels to the `prediction`
   topic (predict.py:177-197).
 
-On GPU the batch-1 step (normalize -> biGRU -> head -> sigmoid) is
-hipGraph-captured (torch.cuda.CUDAGraph == hipGraph on ROCm): one graph
-replay per tick instead of tens of small kernel launches.
+On GPU the whole tick is GPU-resident (torch.cuda.CUDAGraph == hipGraph on
+ROCm). Fast path (bf16, unpadded hidden sizes): the window ring LIVES on
+the GPU — push_row uploads one raw 384-byte fp32 row and a fused ingest
+kernel shifts + normalizes + casts in place; the captured graph is then
+just [gi GEMM -> persistent b1 recurrence] x L -> fused pool/concat ->
+fused head+sigmoid (~6 kernels, no eager glue). Other configs fall back to
+a captured model.forward with a host-side ring.
 """
 import time
 from typing import Dict, List, Optional
@@ -70,44 +74,109 @@ class StreamingPredictor:
         self._static_out = None
         self.n_predictions = 0
 
+        # GPU-resident fast path: ring on device, fused ingest kernel,
+        # dedicated captured graph. Requires bf16 + an unpadded hidden size
+        # (Hp == H, so the direction-concat layout needs no unpad slicing).
+        self._gpu_fast = False
+        if self.device.type == "cuda" and dtype == torch.bfloat16:
+            from ..ops.interface import _pad_h
+            if (_pad_h(model.hidden_size) == model.hidden_size
+                    and (window - 1) * self.n_features <= 48 * 1024):
+                self._gpu_fast = True
+                self._ring_gpu = torch.zeros(window, self.n_features,
+                                             device=self.device,
+                                             dtype=torch.bfloat16)
+                self._row_staging = torch.zeros(self.n_features,
+                                                device=self.device)
+                self._xmin_gpu = self.x_min.float().to(self.device)
+                self._xrng_gpu = self.x_rng.float().to(self.device)
+
     # ---------------- feature ingestion ----------------
 
     def push_row(self, row: torch.Tensor) -> None:
         """Ingest one raw (unnormalized) 108-feature row."""
+        if self._gpu_fast:
+            from ..ops import load_extension
+            ext = load_extension()
+            self._row_staging.copy_(row.float())
+            ext.ingest_row(self._ring_gpu, self._row_staging,
+                           self._xmin_gpu, self._xrng_gpu)
+            self.ring.count += 1
+            return
         self.ring.push(row)
 
     # ---------------- the batch-1 step ----------------
 
+    def _fast_forward(self):
+        """Direct-kernel forward over the GPU ring: gi GEMMs + persistent
+        b1 recurrence per layer, fused pool/concat, fused head+sigmoid."""
+        from ..ops import load_extension
+        from ..ops.interface import _packed_inference_weights, _pad_h
+        ext = load_extension()
+        m = self.model
+        H = m.hidden_size
+        Hp = _pad_h(H)
+        D = m.n_directions
+        T = self.window
+        inp = self._ring_gpu          # (T, F) normalized bf16
+        out = hl = None
+        for layer in range(m.n_layers):
+            params = []
+            for d, sfx_d in enumerate(["", "_reverse"][:D]):
+                sfx = f"l{layer}{sfx_d}"
+                params.append((getattr(m.gru, f"weight_ih_{sfx}"),
+                               getattr(m.gru, f"weight_hh_{sfx}"),
+                               getattr(m.gru, f"bias_ih_{sfx}"),
+                               getattr(m.gru, f"bias_hh_{sfx}")))
+            w_ih, b_ih, w_hh, b_hh = _packed_inference_weights(
+                m.gru, layer, params, Hp, H, self.dtype)
+            gi = torch.addmm(b_ih, inp, w_ih.t()).view(1, T, -1)
+            out, hl = ext.gru_fwd(gi, w_hh, b_hh)
+            inp = out.view(T, -1)
+        feat = ext.pool_concat_infer(out, hl, H)
+        if not hasattr(self, "_head_wb"):
+            self._head_wb = (m.linear.weight.to(self.dtype).contiguous(),
+                             m.linear.bias.to(self.dtype).contiguous())
+        return ext.head_sigmoid(feat, *self._head_wb)
+
     def _capture_graph(self):
         g = torch.cuda.CUDAGraph()
+        fwd = (self._fast_forward if self._gpu_fast else
+               lambda: torch.sigmoid(self.model(self._static_in)))
         # warmup on a side stream (required before capture)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s), torch.no_grad():
             for _ in range(3):
-                out = torch.sigmoid(self.model(self._static_in))
+                out = fwd()
         torch.cuda.current_stream().wait_stream(s)
         with torch.cuda.graph(g), torch.no_grad():
-            self._static_out = torch.sigmoid(self.model(self._static_in))
+            self._static_out = fwd()
         self._graph = g
 
-    def _forward_probs(self, x_norm: torch.Tensor) -> torch.Tensor:
+    def _forward_probs(self, x_norm: Optional[torch.Tensor]) -> torch.Tensor:
         if self._use_graph:
             if self._graph is None:
                 self._capture_graph()
-            self._static_in.copy_(x_norm.to(self.device, self.dtype))
+            if not self._gpu_fast:
+                self._static_in.copy_(x_norm.to(self.device, self.dtype))
             self._graph.replay()
             return self._static_out.float()
         with torch.no_grad():
+            if self._gpu_fast:
+                return self._fast_forward().float()
             x = x_norm.to(self.device, self.dtype)
             return torch.sigmoid(self.model(x)).float()
 
     def predict_window(self) -> Dict:
         """Run inference on the current window; returns the prediction dict
         (shape of predict.py:193-194)."""
-        x = self.ring.buf.unsqueeze(0)  # (1, window, F)
-        x_norm = (x - self.x_min) / self.x_rng
-        probs = self._forward_probs(x_norm).squeeze(0).cpu()
+        if self._gpu_fast:
+            probs = self._forward_probs(None).squeeze(0).cpu()
+        else:
+            x = self.ring.buf.unsqueeze(0)  # (1, window, F)
+            x_norm = (x - self.x_min) / self.x_rng
+            probs = self._forward_probs(x_norm).squeeze(0).cpu()
         idx = (probs > self.prob_threshold).nonzero().flatten().tolist()
         labels = [self.y_fields[i] for i in idx]
         self.n_predictions += 1

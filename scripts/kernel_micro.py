@@ -40,8 +40,26 @@ def main():
     def run_bwd():
         return ext.gru_bwd(gi, w, bhh, out, dout, dhT)
 
-    for name, fn in [("fwd", run_fwd), ("bwd", run_bwd)]:
-        if args.what not in ("both", name):
+    cases = [("fwd", run_fwd), ("bwd", run_bwd)]
+
+    if args.what == "outer":
+        # dW reduction GEMM policy sweep (chunked_outer split-K chunks)
+        from fmda_amd.ops.blas import chunked_outer, enable_tunableop
+        enable_tunableop()
+        M = B * T
+        dgh2 = (torch.randn(M, n_dir * 3 * Hp, device="cuda") * 0.1).bfloat16()
+        out2 = out.reshape(M, -1)
+        cases = []
+        for c in (16, 32, 64, 128, 256):
+            cases.append((f"dW_hh outer c={c}",
+                          lambda c=c: chunked_outer(dgh2, out2, chunks=c)))
+        x2d = (torch.randn(M, 2 * Hp, device="cuda") * 0.1).bfloat16()
+        for c in (32, 64, 128):
+            cases.append((f"dW_ih outer c={c}",
+                          lambda c=c: chunked_outer(dgh2, x2d, chunks=c)))
+
+    for name, fn in cases:
+        if args.what not in ("both", "outer", name):
             continue
         for _ in range(5):
             fn()

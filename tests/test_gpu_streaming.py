@@ -28,11 +28,55 @@ def test_hipgraph_predict_matches_eager():
         b = torch.tensor(pe.predict_window()["probabilities"])
         assert torch.allclose(a, b, atol=1e-3), (a, b)
     # replay with changed input actually changes the output
-    pg.push_row(torch.rand(96, generator=g) * 0.5)
-    pe.push_row(pg.ring.buf[-1].clone())
+    r = torch.rand(96, generator=g) * 0.5
+    pg.push_row(r.clone())
+    pe.push_row(r.clone())
     a = torch.tensor(pg.predict_window()["probabilities"])
     b = torch.tensor(pe.predict_window()["probabilities"])
     assert torch.allclose(a, b, atol=1e-3)
+
+
+def test_gpu_fast_path_matches_model_forward():
+    """The GPU-resident ring + fused ingest/pool/head path must reproduce
+    the plain model forward + sigmoid on the same normalized window."""
+    from fmda_amd.models import BiGRU
+    from fmda_amd.runtime import StreamingPredictor
+    torch.manual_seed(4)
+    m = BiGRU(128, 96, 4, n_layers=2, spatial_dropout=False).to(torch.bfloat16)
+    x_min = torch.rand(96) * 0.1
+    x_max = x_min + 0.5 + torch.rand(96)
+    p = StreamingPredictor(m, x_min, x_max, window=32, device="cuda:0",
+                           dtype=torch.bfloat16, use_graph=True)
+    assert p._gpu_fast
+    g = torch.Generator().manual_seed(11)
+    rows = [torch.rand(96, generator=g) for _ in range(40)]
+    for r in rows:
+        p.push_row(r)
+    probs = torch.tensor(p.predict_window()["probabilities"])
+
+    win = torch.stack(rows[-32:])
+    x_norm = ((win - x_min) / (x_max - x_min)).unsqueeze(0)
+    with torch.no_grad():
+        ref = torch.sigmoid(
+            m.cuda()(x_norm.to("cuda", torch.bfloat16))).float().cpu()
+    assert torch.allclose(probs, ref.squeeze(0), atol=2e-2), (probs, ref)
+
+
+def test_gpu_padded_hidden_falls_back_to_legacy_graph():
+    """H=8 (reference predict config) pads to Hp=16, so the fast path must
+    decline and the legacy captured-model path must still work."""
+    from fmda_amd.models import BiGRU
+    from fmda_amd.runtime import StreamingPredictor
+    torch.manual_seed(5)
+    m = BiGRU(8, 96, 4, n_layers=1, spatial_dropout=False).to(torch.bfloat16)
+    p = StreamingPredictor(m, torch.zeros(96), torch.ones(96), window=5,
+                           device="cuda:0", dtype=torch.bfloat16,
+                           use_graph=True)
+    assert not p._gpu_fast
+    for _ in range(6):
+        p.push_row(torch.rand(96))
+    out = p.predict_window()
+    assert len(out["probabilities"]) == 4
 
 
 def test_graph_capture_happens_once():
