@@ -3198,14 +3198,18 @@ extern "C" int fmda_gru_fwd_b1_launch(const void* gi, const void* w,
                                       const float* bhh, void* out,
                                       float* hlast, int Tseq, int n_dir,
                                       const float* h0, hipStream_t stream) {
-    constexpr int Hp = 128, NT = 256;
+    constexpr int Hp = 128;
     const size_t lds = 2 * (size_t)Tseq * (3 * Hp + 8) + 2 * 2 * (Hp + 8) +
                        4 * 3 * Hp;
     if (lds > 150 * 1024) return -5;   // sequence too long for LDS residency
-    auto k = gru_fwd_b1_kernel<Hp, NT>;
+    // A/B note: a 2-wave NT=128 variant (halved barrier width, CPW 4) was
+    // measured 1.75x SLOWER — the doubled wA fragment hoist (192 VGPRs)
+    // blows the register budget and the spill reloads dominate. 4 waves
+    // with CPW=2 stays the right split.
+    auto k = gru_fwd_b1_kernel<Hp, 256>;
     (void)hipFuncSetAttribute((const void*)k,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
-    k<<<dim3(n_dir), NT, lds, stream>>>(
+    k<<<dim3(n_dir), 256, lds, stream>>>(
         (const __hip_bfloat16*)gi, (const __hip_bfloat16*)w, bhh,
         (__hip_bfloat16*)out, hlast, Tseq, n_dir, h0);
     return hipGetLastError() == hipSuccess ? 0 : -4;

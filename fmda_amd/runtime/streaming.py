@@ -140,7 +140,6 @@ class StreamingPredictor:
         return ext.head_sigmoid(feat, *self._head_wb)
 
     def _capture_graph(self):
-        g = torch.cuda.CUDAGraph()
         fwd = (self._fast_forward if self._gpu_fast else
                lambda: torch.sigmoid(self.model(self._static_in)))
         # warmup on a side stream (required before capture)
@@ -150,6 +149,25 @@ class StreamingPredictor:
             for _ in range(3):
                 out = fwd()
         torch.cuda.current_stream().wait_stream(s)
+        self._pinned_out = None
+        if self._gpu_fast:
+            # capture the D2H probs copy INTO the graph (pinned dst): the
+            # replay then delivers host-readable results with no separate
+            # copy enqueue — one replay + one stream sync per tick
+            pin = torch.zeros(self.model.output_size, dtype=torch.float32,
+                              pin_memory=True)
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g), torch.no_grad():
+                    self._static_out = fwd()
+                    pin.copy_(self._static_out.reshape(-1),
+                              non_blocking=True)
+                self._graph = g
+                self._pinned_out = pin
+                return
+            except RuntimeError:
+                pass  # D2H capture unsupported: plain graph below
+        g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g), torch.no_grad():
             self._static_out = fwd()
         self._graph = g
@@ -171,16 +189,25 @@ class StreamingPredictor:
     def predict_window(self) -> Dict:
         """Run inference on the current window; returns the prediction dict
         (shape of predict.py:193-194)."""
-        if self._gpu_fast:
-            probs = self._forward_probs(None).squeeze(0).cpu()
+        if self._gpu_fast and self._use_graph:
+            if self._graph is None:
+                self._capture_graph()
+            self._graph.replay()
+            torch.cuda.current_stream().synchronize()
+            if self._pinned_out is not None:
+                pl = self._pinned_out.tolist()
+            else:
+                pl = self._static_out.reshape(-1).cpu().tolist()
+        elif self._gpu_fast:
+            pl = self._fast_forward().reshape(-1).cpu().tolist()
         else:
             x = self.ring.buf.unsqueeze(0)  # (1, window, F)
             x_norm = (x - self.x_min) / self.x_rng
-            probs = self._forward_probs(x_norm).squeeze(0).cpu()
-        idx = (probs > self.prob_threshold).nonzero().flatten().tolist()
+            pl = self._forward_probs(x_norm).reshape(-1).cpu().tolist()
+        idx = [i for i, v in enumerate(pl) if v > self.prob_threshold]
         labels = [self.y_fields[i] for i in idx]
         self.n_predictions += 1
-        return {"probabilities": probs.tolist(),
+        return {"probabilities": pl,
                 "prob_threshold": self.prob_threshold,
                 "pred_indices": idx, "pred_labels": labels}
 
