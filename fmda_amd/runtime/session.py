@@ -23,17 +23,58 @@ from .streaming import StreamingPredictor
 
 @dataclass
 class MarketCalendar:
-    """Synthetic market calendar (reference get_market_calendar,
-    getMarketData.py:251, gated at producer.py:218-243)."""
+    """Synthetic market calendar with the reference's full gating
+    semantics (get_market_calendar, getMarketData.py:251; gated at
+    producer.py:218-254):
+
+    - per-day status: weekday gate + HOLIDAY table (the Tradier calendar's
+      'closed' days);
+    - intraday market hours: each open day trades bars [0, close); an
+      EARLY-CLOSE entry shortens `close` (the reference reads per-day
+      open/close times from the calendar entry and bounds the poll loop);
+    - instrument-conditioned FOREX fallback (producer.py:239-243): a
+      non-IEX instrument trades the continuous FOREX week — no intraday
+      bounds and no holidays, dark only on Saturday (the Sunday-17:00 ->
+      Friday-16:00 ET week window, compressed to the synthetic day grid).
+
+    Days are absolute indices from the session start (day 0 = Monday);
+    bars within a day index the 5-minute grid.
+    """
     bars_per_day: int = 78          # 6.5 h of 5-min bars
-    open_days: tuple = (0, 1, 2, 3, 4)  # Mon-Fri
-    forex_fallback: bool = False    # 24h session when True
+    open_days: tuple = (0, 1, 2, 3, 4)  # Mon-Fri weekdays
+    holidays: tuple = ()            # absolute day indices with status closed
+    early_close: Optional[dict] = None  # day index -> bars traded that day
+    forex_fallback: bool = False    # non-IEX instrument: FOREX week window
+
+    def day_index(self, bar_index: int) -> int:
+        return bar_index // self.bars_per_day
+
+    def status(self, day: int) -> str:
+        """'open' | 'early' | 'closed' for an absolute day index."""
+        if self.forex_fallback:
+            return "closed" if day % 7 == 5 else "open"   # Saturday only
+        if (day % 7) not in self.open_days or day in self.holidays:
+            return "closed"
+        if self.early_close and day in self.early_close:
+            return "early"
+        return "open"
+
+    def market_hours(self, day: int):
+        """(open_bar, close_bar) within the day's bar grid; (0, 0) when
+        closed. The FOREX week has no intraday bounds."""
+        st = self.status(day)
+        if st == "closed":
+            return (0, 0)
+        if self.forex_fallback:
+            return (0, self.bars_per_day)
+        if st == "early":
+            return (0, int(self.early_close[day]))
+        return (0, self.bars_per_day)
 
     def is_open(self, bar_index: int) -> bool:
-        if self.forex_fallback:
-            return True
-        day = (bar_index // self.bars_per_day) % 7
-        return day in self.open_days
+        day = self.day_index(bar_index)
+        lo, hi = self.market_hours(day)
+        return lo <= (bar_index % self.bars_per_day) < hi
 
 
 class MarketSession:

@@ -92,7 +92,41 @@ def test_calendar_gate():
     cal = MarketCalendar()
     assert cal.is_open(0)            # Monday
     assert not cal.is_open(5 * 78)   # Saturday
-    assert MarketCalendar(forex_fallback=True).is_open(5 * 78)
+    # FOREX week window (producer.py:239-243): continuous Sun->Fri, dark
+    # only on Saturday; no holiday or intraday gating
+    fx = MarketCalendar(forex_fallback=True, holidays=(0,),
+                        early_close={1: 10})
+    assert not fx.is_open(5 * 78)    # Saturday stays dark
+    assert fx.is_open(6 * 78)        # Sunday session
+    assert fx.is_open(0)             # holiday ignored for FOREX
+    assert fx.is_open(1 * 78 + 50)   # early close ignored for FOREX
+
+
+def test_calendar_holiday_and_early_close():
+    """Holiday-skip day and an early close (reference market calendar's
+    per-day status + open/close hours, producer.py:218-243)."""
+    cal = MarketCalendar(holidays=(2,), early_close={3: 39})
+    assert cal.status(0) == "open"
+    assert cal.status(2) == "closed" and not cal.is_open(2 * 78)
+    assert cal.status(3) == "early"
+    assert cal.market_hours(3) == (0, 39)
+    assert cal.is_open(3 * 78 + 38)       # last bar before the early close
+    assert not cal.is_open(3 * 78 + 39)   # first bar after
+    assert cal.market_hours(2) == (0, 0)
+    assert cal.market_hours(5) == (0, 0)  # weekend
+
+
+def test_session_skips_holiday_and_early_close_bars():
+    from fmda_amd.data.generator import SyntheticMarket
+    from fmda_amd.runtime import MarketSession, MessageBus
+
+    market = SyntheticMarket(78 * 7, seed=5)
+    cal = MarketCalendar(holidays=(1,), early_close={2: 39})
+    bus = MessageBus()
+    s = MarketSession(market, bus=bus, calendar=cal)
+    s.run()
+    # Mon full + Tue holiday (0) + Wed half + Thu/Fri full, weekend closed
+    assert s.published == 78 + 0 + 39 + 78 + 78
 
 
 def test_bus_threaded_producer_consumer():
