@@ -190,4 +190,4 @@ def test_market_calendar_gates_bars():
     cal24 = MarketCalendar(forex_fallback=True)
     s2 = MarketSession(market, bus=MessageBus(), calendar=cal24)
     s2.run()
-    assert s2.published == 78 * 7         # nothing skipped
+    assert s2.published == 78 * 6         # FOREX week: only Saturday dark
