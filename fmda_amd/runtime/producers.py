@@ -172,3 +172,12 @@ class FeedProducers:
         for i in range(n):
             self.publish_bar(i)
         return n
+
+    def run_range(self, start: int, stop: int) -> int:
+        """Publish bars [start, stop) — out-of-order/late-arrival test
+        harness (the reference's delayed-data scenarios,
+        getMarketData.py:208-218)."""
+        stop = min(stop, self.market.n_rows)
+        for i in range(start, stop):
+            self.publish_bar(i)
+        return max(0, stop - start)

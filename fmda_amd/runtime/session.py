@@ -107,7 +107,7 @@ class MarketSession:
         row = self.market.X[i]
         self.bus.publish("deep", {"Timestamp": ts, "row": row})
         if self.predictor is not None:
-            self.predictor.push_row(row)
+            self.predictor.push_row(row, ts=ts)
         self.bus.publish("predict_timestamp", {"Timestamp": ts})
         self.published += 1
         return True

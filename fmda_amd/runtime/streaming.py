@@ -53,13 +53,20 @@ class StreamingPredictor:
                  window: int, bus: Optional[MessageBus] = None,
                  prob_threshold: float = 0.5, stale_after: float = 240.0,
                  device: str = "cpu", dtype: torch.dtype = torch.float32,
-                 use_graph: Optional[bool] = None):
+                 use_graph: Optional[bool] = None,
+                 settle_delay: float = 0.0):
         self.model = model.eval().to(device)
         self.window = window
         self.n_features = model.n_features
         self.bus = bus or MessageBus()
         self.prob_threshold = prob_threshold
         self.stale_after = stale_after
+        # delayed-data tolerance (predict.py:141-157): wait `settle_delay`
+        # for the bar's feature row to land, retry ONCE, then give up
+        self.settle_delay = settle_delay
+        self.last_row_ts: Optional[float] = None
+        self.n_retries = 0
+        self.n_dropped_missing = 0
         self.device = torch.device(device)
         self.dtype = dtype
         self.ring = FeatureRing(window, self.n_features)
@@ -93,8 +100,11 @@ class StreamingPredictor:
 
     # ---------------- feature ingestion ----------------
 
-    def push_row(self, row: torch.Tensor) -> None:
+    def push_row(self, row: torch.Tensor,
+                 ts: Optional[float] = None) -> None:
         """Ingest one raw (unnormalized) 108-feature row."""
+        if ts is not None:
+            self.last_row_ts = ts
         if self._gpu_fast:
             from ..ops import load_extension
             ext = load_extension()
@@ -220,6 +230,16 @@ class StreamingPredictor:
         now = time.time() if now is None else now
         if ts <= now - self.stale_after:    # stale filter (predict.py:135)
             return None
+        # delayed-data tolerance (predict.py:141-157): if the bar's
+        # feature row hasn't landed yet (producer lag), wait for the
+        # settle delay and retry ONCE; drop the message if still missing
+        if self.last_row_ts is not None and self.last_row_ts < ts:
+            self.n_retries += 1
+            if self.settle_delay > 0:
+                time.sleep(self.settle_delay)
+            if self.last_row_ts < ts:
+                self.n_dropped_missing += 1
+                return None
         if not self.ring.full:
             return None
         pred = self.predict_window()

@@ -152,3 +152,34 @@ def test_assembler_tolerates_missing_optional_topics():
     assert torch.all(X[:, col["sd.Asset_long_pos"]] == 0)
     assert torch.equal(X[:, col["sd.4_close"]],
                        market.X[:, col["sd.4_close"]])
+
+
+def test_late_arrival_after_table_read():
+    """A bar arriving AFTER table() has already been read must appear at
+    its correct time position on the next read, with the windowed features
+    (MAs, price_change) recomputed to include it — the in-process
+    equivalent of the reference's delayed-data acceptance
+    (getMarketData.py:208-218): late rows are incorporated, not dropped."""
+    market = SyntheticMarket(60, seed=9)
+    bus = MessageBus()
+    asm = StreamAssembler(bus, emit_signal=False)
+    prod = FeedProducers(market, bus)
+    # publish bars 0..49 except bar 30 (delayed upstream)
+    prod.run_range(0, 30)
+    prod.run_range(31, 50)
+    X1 = asm.table()
+    assert X1.shape[0] == 49           # the missing bar is simply absent
+
+    # the late bar lands after the first read
+    prod.run_range(30, 31)
+    X2 = asm.table()
+    assert X2.shape[0] == 50
+    col = {n: i for i, n in enumerate(FEATURE_NAMES)}
+    # the late row sits at its correct position with its exact raw values
+    torch.testing.assert_close(X2[30, col["sd.4_close"]],
+                               market.X[30, col["sd.4_close"]])
+    # downstream windowed features now include it (recomputed, not stale):
+    # price_change at bar 31 pairs with the true bar-30 close again
+    torch.testing.assert_close(X2[:50, col["pc.price_change"]],
+                               market.X[:50, col["pc.price_change"]],
+                               rtol=2e-4, atol=2e-4)

@@ -191,3 +191,46 @@ def test_market_calendar_gates_bars():
     s2 = MarketSession(market, bus=MessageBus(), calendar=cal24)
     s2.run()
     assert s2.published == 78 * 6         # FOREX week: only Saturday dark
+
+
+def test_delayed_data_retry_and_drop():
+    """Delayed-data tolerance (predict.py:141-157): a timestamp message
+    whose feature row hasn't arrived is retried once after the settle
+    delay, and dropped if the row is still missing."""
+    import threading
+
+    from fmda_amd.models import BiGRU
+    from fmda_amd.runtime import StreamingPredictor
+
+    torch.manual_seed(0)
+    model = BiGRU(8, 16, 4, spatial_dropout=False)
+    now = 1_000_000.0
+
+    # row never arrives -> retry once, then drop
+    p = StreamingPredictor(model, torch.zeros(16), torch.ones(16), window=3,
+                           use_graph=False, settle_delay=0.0)
+    for i in range(3):
+        p.push_row(torch.rand(16), ts=now + i)
+    assert p.handle_timestamp({"Timestamp": now + 10}, now=now + 10) is None
+    assert p.n_retries == 1 and p.n_dropped_missing == 1
+
+    # row lands during the settle sleep (producer lag) -> prediction runs
+    p2 = StreamingPredictor(model, torch.zeros(16), torch.ones(16), window=3,
+                            use_graph=False, settle_delay=0.2)
+    for i in range(3):
+        p2.push_row(torch.rand(16), ts=now + i)
+    t = threading.Timer(0.05, lambda: p2.push_row(torch.rand(16),
+                                                  ts=now + 10))
+    t.start()
+    out = p2.handle_timestamp({"Timestamp": now + 10}, now=now + 10)
+    t.join()
+    assert out is not None and p2.n_retries == 1
+    assert p2.n_dropped_missing == 0
+
+    # on-time message needs no retry
+    p3 = StreamingPredictor(model, torch.zeros(16), torch.ones(16), window=3,
+                            use_graph=False)
+    for i in range(3):
+        p3.push_row(torch.rand(16), ts=now + i)
+    assert p3.handle_timestamp({"Timestamp": now + 2}, now=now + 2) is not None
+    assert p3.n_retries == 0
