@@ -58,17 +58,21 @@ def chunked_colsum(dg: torch.Tensor, chunks: int = 48) -> torch.Tensor:
 
 
 def chunked_outer(dg: torch.Tensor, x: torch.Tensor,
-                  chunks: int = 64) -> torch.Tensor:
+                  chunks: int = 64, out_fp32: bool = False) -> torch.Tensor:
     """(M, N)^T @ (M, K) -> (N, K) via hand split-K: C row-chunks, batched
     GEMM, fp32 chunk sum. Beats the library's unsplit fat-K GEMM by ~7x on
-    MI355X for the training-step dW shapes."""
+    MI355X for the training-step dW shapes. out_fp32=True returns the fp32
+    chunk sum directly — callers that want fp32 master-grads then skip a
+    round trip through the compute dtype (two elementwise kernels each)."""
     M, N = dg.shape
     K = x.shape[1]
     c = chunks
     while c > 1 and M % c != 0:
         c //= 2
     if c <= 1:
-        return torch.matmul(dg.t(), x)
+        r = torch.matmul(dg.t(), x)
+        return r.float() if out_fp32 else r
     parts = torch.bmm(dg.view(c, M // c, N).transpose(1, 2),
                       x.view(c, M // c, K))
-    return parts.float().sum(dim=0).to(dg.dtype)
+    s = parts.float().sum(dim=0)
+    return s if out_fp32 else s.to(dg.dtype)

@@ -362,15 +362,16 @@ class _BiGRULayer(torch.autograd.Function):
         M = dgi.shape[0] * dgi.shape[1]
 
         # dW_hh via the time-shifted dGh and one split-K reduction
-        cross = chunked_outer(dgh.reshape(M, -1), out.reshape(M, -1))
+        # (fp32 out: the grads feed fp32 masters, skip the bf16 round trip)
+        cross = chunked_outer(dgh.reshape(M, -1), out.reshape(M, -1),
+                              out_fp32=True)
         if dgh0 is not None:
             # t=0 term of dW_hh: dGh_0 (x) h0 (boundary slot holds zeros)
             for d in range(D):
                 cross[d * 3 * Hp:(d + 1) * 3 * Hp,
-                      d * Hp:(d + 1) * Hp] += \
-                    (dgh0[d].float().t() @ h0c[d]).to(cross.dtype)
+                      d * Hp:(d + 1) * Hp] += dgh0[d].float().t() @ h0c[d]
         # dW_ih for both directions in one split-K reduction
-        dwih_cat = chunked_outer(dgi.reshape(M, -1), x2d)
+        dwih_cat = chunked_outer(dgi.reshape(M, -1), x2d, out_fp32=True)
         dx = None
         if need_dx:
             dx = torch.matmul(dgi.reshape(M, -1), w_ih_cat)
@@ -378,9 +379,11 @@ class _BiGRULayer(torch.autograd.Function):
 
         grads = []
         for d in range(D):
-            dwih = dwih_cat[d * 3 * Hp:(d + 1) * 3 * Hp].float()
+            dwih = dwih_cat[d * 3 * Hp:(d + 1) * 3 * Hp]
+            # the column-block slice is non-contiguous; the fused optimizer
+            # (and DDP flattening) need contiguous grads
             dwhh = cross[d * 3 * Hp:(d + 1) * 3 * Hp,
-                         d * Hp:(d + 1) * Hp].float()
+                         d * Hp:(d + 1) * Hp].contiguous()
             grads.append((
                 _unpad_gate_rows(dwih, H, Hp),
                 _unpad_gate_rows(dwhh, H, Hp)[:, :H],
