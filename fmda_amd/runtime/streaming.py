@@ -19,6 +19,7 @@ just [gi GEMM -> persistent b1 recurrence] x L -> fused pool/concat ->
 fused head+sigmoid (~6 kernels, no eager glue). Other configs fall back to
 a captured model.forward with a host-side ring.
 """
+import threading
 import time
 from typing import Dict, List, Optional
 
@@ -70,6 +71,11 @@ class StreamingPredictor:
         self.device = torch.device(device)
         self.dtype = dtype
         self.ring = FeatureRing(window, self.n_features)
+        # serve.py runs handlers on a threadpool: ingestion and prediction
+        # must not interleave on the ring / staging buffers / graph.
+        # handle_timestamp's settle-delay sleep stays OUTSIDE the lock so a
+        # concurrent push_row can satisfy the retry.
+        self._lock = threading.Lock()
         self.x_min = x_min.clone()
         self.x_rng = (x_max - x_min).clone()
         self.y_fields: List[str] = list(TARGET_NAMES)
@@ -103,17 +109,20 @@ class StreamingPredictor:
     def push_row(self, row: torch.Tensor,
                  ts: Optional[float] = None) -> None:
         """Ingest one raw (unnormalized) 108-feature row."""
-        if ts is not None:
-            self.last_row_ts = ts
-        if self._gpu_fast:
-            from ..ops import load_extension
-            ext = load_extension()
-            self._row_staging.copy_(row.float())
-            ext.ingest_row(self._ring_gpu, self._row_staging,
-                           self._xmin_gpu, self._xrng_gpu)
-            self.ring.count += 1
-            return
-        self.ring.push(row)
+        with self._lock:
+            if self._gpu_fast:
+                from ..ops import load_extension
+                ext = load_extension()
+                self._row_staging.copy_(row.float())
+                ext.ingest_row(self._ring_gpu, self._row_staging,
+                               self._xmin_gpu, self._xrng_gpu)
+                self.ring.count += 1
+            else:
+                self.ring.push(row)
+            if ts is not None:
+                # published AFTER the row is in the ring: a reader seeing
+                # last_row_ts >= ts may predict on the window immediately
+                self.last_row_ts = ts
 
     # ---------------- the batch-1 step ----------------
 
@@ -199,6 +208,10 @@ class StreamingPredictor:
     def predict_window(self) -> Dict:
         """Run inference on the current window; returns the prediction dict
         (shape of predict.py:193-194)."""
+        with self._lock:
+            return self._predict_window_locked()
+
+    def _predict_window_locked(self) -> Dict:
         if self._gpu_fast and self._use_graph:
             if self._graph is None:
                 self._capture_graph()

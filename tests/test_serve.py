@@ -64,3 +64,47 @@ def test_metrics_exposition(client):
     text = r.text
     assert "fmda_rows_ingested_total" in text
     assert "fmda_predict_latency_seconds" in text
+
+
+def test_concurrent_ingest_and_timestamp():
+    """serve.py handlers run on a threadpool: hammer ingest and timestamp
+    concurrently; the predictor lock must keep every response well-formed
+    and the ring consistent."""
+    import threading
+
+    import torch
+
+    from fmda_amd.models import BiGRU
+    from fmda_amd.runtime import StreamingPredictor
+
+    torch.manual_seed(0)
+    m = BiGRU(8, 16, 4, spatial_dropout=False)
+    p = StreamingPredictor(m, torch.zeros(16), torch.ones(16), window=4,
+                           use_graph=False, stale_after=1e12)
+    now = 1_000_000.0
+    errs = []
+
+    def ingest(k):
+        try:
+            for i in range(50):
+                p.push_row(torch.rand(16), ts=now + k * 50 + i)
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    def predict():
+        try:
+            for i in range(50):
+                p.handle_timestamp({"Timestamp": now}, now=now)
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    threads = [threading.Thread(target=ingest, args=(k,)) for k in range(2)]
+    threads += [threading.Thread(target=predict) for _ in range(2)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    assert not errs
+    assert p.ring.count == 100
+    out = p.predict_window()
+    assert len(out["probabilities"]) == 4
