@@ -415,6 +415,12 @@ void ingest_row(torch::Tensor ring, torch::Tensor row, torch::Tensor xmin,
     TORCH_CHECK(row.is_cuda() && row.is_contiguous() &&
                 row.scalar_type() == torch::kFloat32 &&
                 row.numel() == ring.size(1));
+    TORCH_CHECK(xmin.is_cuda() && xmin.is_contiguous() &&
+                xmin.scalar_type() == torch::kFloat32 &&
+                xmin.numel() == ring.size(1), "xmin must be fp32 (F) on GPU");
+    TORCH_CHECK(xrng.is_cuda() && xrng.is_contiguous() &&
+                xrng.scalar_type() == torch::kFloat32 &&
+                xrng.numel() == ring.size(1), "xrng must be fp32 (F) on GPU");
     auto stream = at::hip::getCurrentHIPStream();
     int rc = fmda_ingest_row_launch(
         ring.data_ptr(), row.data_ptr<float>(), xmin.data_ptr<float>(),

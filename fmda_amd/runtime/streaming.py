@@ -222,7 +222,8 @@ class StreamingPredictor:
             else:
                 pl = self._static_out.reshape(-1).cpu().tolist()
         elif self._gpu_fast:
-            pl = self._fast_forward().reshape(-1).cpu().tolist()
+            with torch.no_grad():
+                pl = self._fast_forward().reshape(-1).cpu().tolist()
         else:
             x = self.ring.buf.unsqueeze(0)  # (1, window, F)
             x_norm = (x - self.x_min) / self.x_rng
