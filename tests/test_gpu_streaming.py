@@ -107,3 +107,37 @@ def test_end_to_end_demo_on_gpu(tmp_path):
     assert out.returncode == 0, (out.stdout[-800:], out.stderr[-1200:])
     assert "predictions" in out.stdout
     assert (tmp_path / "model_params.pt").exists()
+
+
+def test_predict_tick_latency_bound():
+    """Latency guard for the hipGraph streaming tick: p50 must stay in the
+    captured-graph regime (measured 0.21 ms; bound set loose enough for
+    box variance but far below the ~0.5 ms eager / ~1 ms uncaptured
+    regimes, so a silent fallback or a lost fusion trips it)."""
+    import time as _t
+
+    import numpy as np
+
+    from fmda_amd.models import BiGRU
+    from fmda_amd.runtime import StreamingPredictor
+
+    torch.manual_seed(2)
+    m = BiGRU(128, 96, 4, n_layers=2, spatial_dropout=False).to(torch.bfloat16)
+    p = StreamingPredictor(m, torch.zeros(96), torch.ones(96), window=120,
+                           device="cuda:0", dtype=torch.bfloat16,
+                           use_graph=True)
+    assert p._gpu_fast
+    g = torch.Generator().manual_seed(1)
+    for _ in range(120):
+        p.push_row(torch.rand(96, generator=g))
+    for _ in range(10):
+        p.predict_window()
+    torch.cuda.synchronize()
+    lats = []
+    for _ in range(100):
+        p.push_row(torch.rand(96, generator=g))
+        t0 = _t.perf_counter()
+        p.predict_window()
+        lats.append((_t.perf_counter() - t0) * 1000.0)
+    p50 = float(np.percentile(lats, 50))
+    assert p50 < 0.40, f"predict tick p50 regressed: {p50:.3f} ms"
