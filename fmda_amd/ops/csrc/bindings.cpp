@@ -15,14 +15,21 @@ extern "C" int fmda_gru_fwd_cs_launch(const void* gi, const void* w,
                                       const float* bhh, void* out,
                                       float* hlast, void* hpub,
                                       unsigned int* cnt, int B, int Tseq,
-                                      int n_dir, hipStream_t stream);
+                                      int n_dir, void* out_drop,
+                                      unsigned int drop_thr,
+                                      float drop_scale,
+                                      unsigned long long drop_seed,
+                                      hipStream_t stream);
 extern "C" int fmda_gru_bwd_cs_launch(const void* gi, const void* w,
                                       const void* wt, const float* bhh,
                                       const void* out, const void* dout,
                                       const float* dhT, void* dgi, void* dgh,
                                       float* dh0, float* dbhh, void* gpub,
                                       unsigned int* cnt, int B, int Tseq,
-                                      int n_dir, hipStream_t stream);
+                                      int n_dir, unsigned int drop_thr,
+                                      float drop_scale,
+                                      unsigned long long drop_seed,
+                                      hipStream_t stream);
 extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                    const void* w, const void* wt,
                                    const float* bhh,
@@ -142,8 +149,10 @@ std::vector<torch::Tensor> gru_fwd(torch::Tensor gi, torch::Tensor w,
     torch::Tensor out_drop;
     void* out_drop_ptr = nullptr;
     if (drop_thr != 0u) {
-        TORCH_CHECK(is_bf16 && Hp == 128,
-                    "fused fwd dropout requires bf16 Hp=128");
+        TORCH_CHECK(is_bf16 && (Hp == 128 || Hp == 512),
+                    "fused fwd dropout requires bf16 Hp=128/512");
+        TORCH_CHECK(Hp == 128 || h0 == nullptr,
+                    "fused fwd dropout with h0 requires Hp=128");
         out_drop = torch::empty_like(out);
         out_drop_ptr = out_drop.data_ptr();
     }
@@ -162,7 +171,10 @@ std::vector<torch::Tensor> gru_fwd(torch::Tensor gi, torch::Tensor w,
                                     bhh.data_ptr<float>(), out.data_ptr(),
                                     hlast.data_ptr<float>(), hpub.data_ptr(),
                                     (unsigned int*)cnt.data_ptr(), B, T,
-                                    n_dir, stream.stream());
+                                    n_dir, out_drop_ptr, drop_thr,
+                                    drop_scale,
+                                    (unsigned long long)drop_seed,
+                                    stream.stream());
     } else {
         rc = fmda_gru_fwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
                                  w.data_ptr(), bhh.data_ptr<float>(),
@@ -227,6 +239,10 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
     auto stream = at::hip::getCurrentHIPStream();
     int rc;
     if (is_bf16 && Hp == 512 && h0 == nullptr) {
+        const unsigned int cs_thr =
+            (drop_p > 0.0) ? (unsigned int)((float)drop_p * 256.0f) : 0u;
+        const float cs_scale =
+            (drop_p > 0.0) ? (float)(1.0 / (1.0 - drop_p)) : 1.0f;
         const int BR = 256;
         const int GB = (B + BR - 1) / BR;
         const int G = GB * n_dir;
@@ -237,7 +253,8 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
             out.data_ptr(), dout.data_ptr(), dhT.data_ptr<float>(),
             dgi.data_ptr(), dgh.data_ptr(), dh0.data_ptr<float>(),
             dbsum.data_ptr<float>(), gpub.data_ptr(),
-            (unsigned int*)cnt.data_ptr(), B, T, n_dir, stream.stream());
+            (unsigned int*)cnt.data_ptr(), B, T, n_dir, cs_thr, cs_scale,
+            (unsigned long long)drop_seed, stream.stream());
     } else {
         // fused inter-layer dropout backward (d_out is w.r.t. the DROPPED
         // activations; the kernel recomputes the counter-based mask at the
@@ -250,7 +267,8 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
         const float drop_scale =
             (drop_p > 0.0) ? (float)(1.0 / (1.0 - drop_p)) : 1.0f;
         TORCH_CHECK(drop_thr == 0u || (is_bf16 && Hp == 128),
-                    "fused dropout-backward requires bf16 Hp=128");
+                    "fused dropout-backward on this path requires bf16 "
+                    "Hp=128");
         rc = fmda_gru_bwd_launch(is_bf16 ? 1 : 0, Hp, gi.data_ptr(),
                                  w.data_ptr(), wt_ptr, bhh.data_ptr<float>(),
                                  out.data_ptr(), dout.data_ptr(),

@@ -1804,7 +1804,10 @@ void gru_fwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
                        float* __restrict__ hlast,
                        __hip_bfloat16* __restrict__ hpub,  // (2,G,BR,Hp)
                        unsigned int* __restrict__ cnt,     // (G)
-                       int B, int Tseq, int n_dir, int GB) {
+                       int B, int Tseq, int n_dir, int GB,
+                       __hip_bfloat16* __restrict__ out_drop,
+                       unsigned int drop_thr, float drop_scale,
+                       unsigned long long drop_seed) {
     constexpr int CT = Hp / CS;          // column-blocks per group
     constexpr int NW = NT / 64;          // waves
     constexpr int MT = BR / 16 / NW;     // m-tiles per wave
@@ -2008,6 +2011,35 @@ void gru_fwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
         store_tile<__hip_bfloat16, BR, NT>(
             hst, out_b + (long)tt * n_dir * Hp + ct * CS, CS, CS, out_row,
             rows_valid, tid);
+        if (out_drop != nullptr) {
+            // fused forward dropout (same octet mask as dropout_kernel
+            // over the (B, T, n_dir*Hp) out layout)
+            __hip_bfloat16* dst = out_drop + (long)b0 * out_row +
+                                  (long)dir * Hp + (long)tt * n_dir * Hp +
+                                  ct * CS;
+            constexpr int CPR = CS * 2 / 16;
+            for (int c = tid; c < BR * CPR; c += NT) {
+                const int r = c / CPR, jc = c % CPR;
+                if (r >= rows_valid) continue;
+                bf16x8_t v = *(const bf16x8_t*)((const char*)hst +
+                                                (long)r * CS * 2 + jc * 16);
+                const long o = ((long)(b0 + r) * Tseq + tt) * n_dir * Hp +
+                               (long)dir * Hp + ct * CS + jc * 8;
+                const unsigned long long rnd =
+                    mix64(drop_seed ^ (unsigned long long)(o >> 3));
+#pragma unroll
+                for (int k = 0; k < 8; ++k) {
+                    const unsigned int u =
+                        (unsigned int)(rnd >> (8 * k)) & 0xFF;
+                    const float f =
+                        (u < drop_thr)
+                            ? 0.0f : (float)((__bf16*)&v)[k] * drop_scale;
+                    ((__bf16*)&v)[k] = (__bf16)__float2bfloat16(f);
+                }
+                *(bf16x8_t*)((char*)dst + (long)r * out_row * 2 + jc * 16) =
+                    v;
+            }
+        }
         {
             __hip_bfloat16* dst = pub_base + (long)((u + 1) & 1) * ring_stride;
             constexpr int CPR = CS * 2 / 16;
@@ -2053,7 +2085,11 @@ extern "C" int fmda_gru_fwd_cs_launch(const void* gi, const void* w,
                                       const float* bhh, void* out,
                                       float* hlast, void* hpub,
                                       unsigned int* cnt, int B, int Tseq,
-                                      int n_dir, hipStream_t stream) {
+                                      int n_dir, void* out_drop,
+                                      unsigned int drop_thr,
+                                      float drop_scale,
+                                      unsigned long long drop_seed,
+                                      hipStream_t stream) {
     constexpr int BR = 256, Hp = 512, CS = 32, NT = 512;
     const int GB = (B + BR - 1) / BR;
     const size_t lds = 2 * 3 * CS * (Hp + 8) + 2 * (size_t)BR * 3 * CS +
@@ -2066,7 +2102,8 @@ extern "C" int fmda_gru_fwd_cs_launch(const void* gi, const void* w,
                                  (const __hip_bfloat16*)w, bhh,
                                  (__hip_bfloat16*)out, hlast,
                                  (__hip_bfloat16*)hpub, cnt, B, Tseq, n_dir,
-                                 GB);
+                                 GB, (__hip_bfloat16*)out_drop, drop_thr,
+                                 drop_scale, drop_seed);
     return hipGetLastError() == hipSuccess ? 0 : -1;
 }
 
@@ -2102,7 +2139,8 @@ void gru_bwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
                        float* __restrict__ dh0, float* __restrict__ dbhh,
                        __hip_bfloat16* __restrict__ gpub,  // (2,G,BR,3Hp)
                        unsigned int* __restrict__ cnt, int B, int Tseq,
-                       int n_dir, int GB) {
+                       int n_dir, int GB, unsigned int drop_thr,
+                       float drop_scale, unsigned long long drop_seed) {
     constexpr int CT = Hp / CS;
     constexpr int NW = NT / 64;
     constexpr int MT = BR / 16 / NW;
@@ -2326,9 +2364,22 @@ void gru_bwd_cs_kernel(const __hip_bfloat16* __restrict__ gi,
                         (live && haveh)
                             ? (float)hsrc[(long)b * out_row + ct * CS + jj]
                             : 0.0f;
-                    const float dht =
-                        dhcar[j][m][e] +
-                        (live ? (float)do_t[(long)b * out_row + jj] : 0.0f);
+                    float doval =
+                        live ? (float)do_t[(long)b * out_row + jj] : 0.0f;
+                    if (drop_thr != 0u) {
+                        // fused inter-layer dropout backward: recompute
+                        // the counter-based mask at the d_out read (the
+                        // v3 kernel's scheme at the cs shapes)
+                        const long o = ((long)(b0 + b) * Tseq + tt) *
+                                           n_dir * Hp +
+                                       (long)dir * Hp + ct * CS + jj;
+                        const unsigned long long rnd =
+                            mix64(drop_seed ^ (unsigned long long)(o >> 3));
+                        const unsigned int u =
+                            (unsigned int)(rnd >> (8 * ((int)o & 7))) & 0xFF;
+                        doval = (u < drop_thr) ? 0.0f : doval * drop_scale;
+                    }
+                    const float dht = dhcar[j][m][e] + doval;
                     const float dz_pre =
                         live ? dht * (hprev - n) * z * (1.0f - z) : 0.0f;
                     const float dn_pre =
@@ -3355,7 +3406,10 @@ extern "C" int fmda_gru_bwd_cs_launch(const void* gi, const void* w,
                                       const float* dhT, void* dgi, void* dgh,
                                       float* dh0, float* dbhh, void* gpub,
                                       unsigned int* cnt, int B, int Tseq,
-                                      int n_dir, hipStream_t stream) {
+                                      int n_dir, unsigned int drop_thr,
+                                      float drop_scale,
+                                      unsigned long long drop_seed,
+                                      hipStream_t stream) {
     constexpr int BR = 256, Hp = 512, CS = 32, NT = 512;
     const int GB = (B + BR - 1) / BR;
     const size_t lds = 2 * 3 * CS * (Hp + 8) + 2 * (size_t)BR * 3 * CS +
@@ -3369,7 +3423,7 @@ extern "C" int fmda_gru_bwd_cs_launch(const void* gi, const void* w,
         (const __hip_bfloat16*)wt, bhh, (const __hip_bfloat16*)out,
         (const __hip_bfloat16*)dout, dhT, (__hip_bfloat16*)dgi,
         (__hip_bfloat16*)dgh, dh0, dbhh, (__hip_bfloat16*)gpub, cnt, B,
-        Tseq, n_dir, GB);
+        Tseq, n_dir, GB, drop_thr, drop_scale, drop_seed);
     return hipGetLastError() == hipSuccess ? 0 : -1;
 }
 

@@ -472,8 +472,14 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
         # seed NOW so the producing layer's BPTT kernel can apply the same
         # counter-based mask at its d_out read (backward pass fused away)
         drop_here = (training and dropout_p > 0 and layer < n_layers - 1)
-        defer = (drop_here and Hp == H == 128
-                 and x.dtype == torch.bfloat16 and x.is_cuda)
+        # fully fused inter-layer dropout: v3 (Hp=128) and column-split
+        # (Hp=512, zero-h0) kernels emit the dropped copy in forward and
+        # recompute the mask at the backward d_out read
+        import os as _os
+        defer = (drop_here and Hp == H and Hp in (128, 512)
+                 and x.dtype == torch.bfloat16 and x.is_cuda
+                 and (Hp == 128 or (h0_layers[layer] is None
+                      and _os.environ.get("FMDA_CS_DROP", "1") != "0")))
         seed = (int(torch.empty((), dtype=torch.int64).random_())
                 if defer else 0)
         dp = (dropout_p, seed) if defer else (0.0, 0)

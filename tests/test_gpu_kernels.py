@@ -406,14 +406,15 @@ def test_unidirectional_model_on_gpu():
     assert torch.allclose(got, ref, atol=5e-3), (got - ref).abs().max()
 
 
-def test_deferred_dropout_backward_matches_explicit():
+@pytest.mark.parametrize("B,T,F,H", [(48, 24, 64, 128), (16, 8, 64, 512)])
+def test_deferred_dropout_backward_matches_explicit(B, T, F, H):
     """The dropout mask applied inside gru_bwd's d_out read (deferred
     path) must produce the same layer gradients as the explicit
-    _FusedDropout backward pass (same seed, same counter-based mask)."""
+    _FusedDropout backward pass (same seed, same counter-based mask) —
+    on the v3 (H=128) and column-split (H=512) kernels."""
     from fmda_amd.ops.interface import (_BiGRULayer, _DeferredDropout,
                                         _FusedDropout)
     torch.manual_seed(3)
-    B, T, F, H = 48, 24, 64, 128
     p_drop, seed = 0.3, 987654321
     masters = []
     for _ in range(2):
@@ -442,9 +443,10 @@ def test_deferred_dropout_backward_matches_explicit():
     ga, da = run(True)
     gb, db = run(False)
     assert torch.equal(da, db)  # forward mask identical by construction
+    tol = 2e-2 if H <= 128 else 8e-2   # bf16 compounding at H=512
     for i, (a, b) in enumerate(zip(ga, gb)):
         rel = (a - b).norm() / b.norm().clamp(min=1e-12)
-        assert rel < 2e-2, (i, float(rel))
+        assert rel < tol, (i, float(rel))
 
 
 def test_padded_hidden_h8_model_matches_cpu():
