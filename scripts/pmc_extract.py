@@ -16,15 +16,17 @@ vi = cols.index("value")
 agg = collections.defaultdict(float)
 for r in pdb.execute("SELECT * FROM counters_collection"):
     agg[(r[ki].split("(")[0][:50], r[ci])] += r[vi]
+counters = sorted({c for _, c in agg})
 with open(out_path, "w") as f:
     for k in sorted({a for a, _ in agg}):
         busy = agg.get((k, "SQ_BUSY_CYCLES"), 0.0)
-
-        def ratio(c):
-            return agg.get((k, c), 0.0) / busy / 16 if busy else 0.0
-
-        line = (f"{k}: active={ratio('SQ_ACTIVE_INST_ANY'):.3f} "
-                f"mfma={ratio('SQ_VALU_MFMA_BUSY_CYCLES'):.3f} "
-                f"wait={ratio('SQ_WAIT_ANY'):.3f}")
+        parts = [k]
+        for c in counters:
+            if c == "SQ_BUSY_CYCLES":
+                continue
+            v = agg.get((k, c), 0.0)
+            r = v / busy / 16 if busy else 0.0
+            parts.append(f"  {c}: {v:.3e} (ratio {r:.3f})")
+        line = "\n".join(parts) + f"\n  SQ_BUSY_CYCLES: {busy:.3e}"
         print(line)
         f.write(line + "\n")
