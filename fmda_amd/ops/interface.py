@@ -153,23 +153,6 @@ class _FusedDropout(torch.autograd.Function):
         return ext.dropout_fused(dy.contiguous(), ctx.p, ctx.seed), None, None
 
 
-class _DeferredDropout(torch.autograd.Function):
-    """Inter-layer dropout whose BACKWARD is fused into the producing
-    layer's BPTT kernel: forward applies the counter-based mask as usual,
-    backward passes the gradient through untouched — the (p, seed) pair is
-    handed to the upstream _BiGRULayer, whose gru_bwd applies the same
-    mask at its d_out read (saving a full-tensor read+write pass)."""
-
-    @staticmethod
-    def forward(ctx, x, p, seed):
-        ext = load_extension()
-        return ext.dropout_fused(x.contiguous(), p, seed)
-
-    @staticmethod
-    def backward(ctx, dy):
-        return dy, None, None
-
-
 def fused_dropout(x: torch.Tensor, p: float) -> torch.Tensor:
     """Training-mode dropout on the HIP engine (bf16 CUDA tensors);
     falls back to torch for other dtypes/devices."""

@@ -412,8 +412,7 @@ def test_deferred_dropout_backward_matches_explicit(B, T, F, H):
     path) must produce the same layer gradients as the explicit
     _FusedDropout backward pass (same seed, same counter-based mask) —
     on the v3 (H=128) and column-split (H=512) kernels."""
-    from fmda_amd.ops.interface import (_BiGRULayer, _DeferredDropout,
-                                        _FusedDropout)
+    from fmda_amd.ops.interface import _BiGRULayer, _FusedDropout
     torch.manual_seed(3)
     p_drop, seed = 0.3, 987654321
     masters = []
