@@ -116,6 +116,11 @@ void check_common(const torch::Tensor& gi, const torch::Tensor& w,
     is_bf16 = gi.scalar_type() == torch::kBFloat16;
     TORCH_CHECK(is_bf16 || gi.scalar_type() == torch::kFloat32,
                 "fmda gru: dtype must be bf16 or fp32");
+    // the bf16 recurrence tiles MFMA K=32: Hp=16 would silently drop the
+    // recurrent term (KK = Hp/32 = 0) — pad to Hp >= 32 instead
+    TORCH_CHECK(!(is_bf16 && Hp < 32),
+                "fmda gru: bf16 requires Hp >= 32 (MFMA K=32); pad the "
+                "hidden size");
 }
 
 }  // namespace

@@ -3369,7 +3369,7 @@ extern "C" int fmda_gru_fwd_launch(int is_bf16, int Hp, const void* gi,
                                           n_dir, h0, lds, stream)
     if (is_bf16) {
         switch (Hp) {
-            case 16: F(bf16_t, 32, 16, true, 256, false); break;
+            case 16: return -7;   // KK = Hp/32 = 0: zero-trip MFMA loop
             case 32: F(bf16_t, 32, 32, true, 256, false); break;
             case 64: F(bf16_t, 32, 64, true, 256, false); break;
             case 128:
@@ -3450,7 +3450,7 @@ extern "C" int fmda_gru_bwd_launch(int is_bf16, int Hp, const void* gi,
                                           h0, dgh0, lds, stream)
     if (is_bf16) {
         switch (Hp) {
-            case 16: G(bf16_t, 32, 16, true, 256, false); break;
+            case 16: return -7;   // KK = Hp/32 = 0: zero-trip MFMA loop
             case 32: G(bf16_t, 32, 32, true, 256, false); break;
             case 64: G(bf16_t, 32, 64, true, 256, false); break;
             case 128:

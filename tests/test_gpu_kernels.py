@@ -727,3 +727,52 @@ def test_gru_backward_h0_bf16_v3_kernel():
         bhh.clone().requires_grad_(True), None)
     ((out3 * dO).sum() + (hl3 * dH).sum()).backward()
     assert (w1.grad.float() - w3.grad).norm() / w3.grad.norm() > 1e-3
+
+
+def test_gru_randomized_shape_sweep():
+    """Seeded randomized sweep over (H, T, B, n_dir, dtype, h0?) — odd
+    batch/sequence sizes and padded hidden sizes that the fixed
+    parametrizations miss. Forward vs the golden recurrence; backward via
+    gru_directions for the fp32 draws."""
+    from fmda_amd.ops.interface import gru_directions
+    import random
+    ext = _ext()
+    rng = random.Random(20240914)
+    for trial in range(12):
+        T = rng.randint(2, 40)
+        B = rng.choice([1, 2, 3, 7, 17, 33, 65, 130])
+        n_dir = rng.choice([1, 2])
+        bf16 = rng.random() < 0.5
+        # bf16 needs Hp >= 32 (MFMA K=32; Hp=16 is rejected at the binding)
+        H = rng.choice([32, 64, 128, 256] if bf16 else [16, 32, 64, 128, 256])
+        use_h0 = rng.random() < 0.5
+        torch.manual_seed(1000 + trial)
+        gi = (torch.randn(B, T, n_dir * 3 * H) * 0.5).cuda()
+        w = (torch.randn(n_dir, 3 * H, H) * 0.2).cuda()
+        bhh = (torch.randn(n_dir, 3 * H) * 0.1).cuda()
+        h0 = ((torch.randn(n_dir, B, H) * 0.5).cuda().contiguous()
+              if use_h0 else None)
+        if bf16:
+            out, hlast = ext.gru_fwd(gi.bfloat16(), w.bfloat16(), bhh, h0)
+            out_ref, hl_ref = _gru_ref_from_gi(gi, w, bhh, h0)
+            ctx = (trial, H, T, B, n_dir, "bf16", use_h0)
+            assert (out.float() - out_ref).abs().max() < 0.06, ctx
+            assert (hlast - hl_ref).abs().max() < 0.06, ctx
+        else:
+            gi1 = gi.clone().requires_grad_(True)
+            w1 = w.clone().requires_grad_(True)
+            b1 = bhh.clone().requires_grad_(True)
+            out, hlast = gru_directions(gi1, w1, b1, h0)
+            dO = torch.randn_like(out)
+            (out * dO).sum().backward()
+            gi2 = gi.clone().requires_grad_(True)
+            w2 = w.clone().requires_grad_(True)
+            b2 = bhh.clone().requires_grad_(True)
+            out2, _ = _gru_ref_from_gi_autograd(gi2, w2, b2, h0)
+            (out2 * dO).sum().backward()
+            ctx = (trial, H, T, B, n_dir, "fp32", use_h0)
+            assert torch.allclose(out, out2, atol=2e-4), ctx
+            for a, b in [(gi1.grad, gi2.grad), (w1.grad, w2.grad),
+                         (b1.grad, b2.grad)]:
+                scale = b.abs().max().clamp(min=1.0)
+                assert ((a - b).abs().max() / scale) < 2e-4, ctx
