@@ -286,7 +286,7 @@ def test_fused_head_loss_matches_eager(dtype):
 
 
 @pytest.mark.parametrize("B,H", [(64, 128), (33, 128), (48, 512), (300, 512),
-                                 (33, 64)])
+                                 (33, 64), (16, 256), (24, 32)])
 def test_bigru_layer_bf16_all_grads_vs_fp32_reference(B, H):
     """Full single-layer node (pack + projection + recurrence) in bf16 vs
     an fp32 autograd reference: checks out/h_last AND every parameter
@@ -776,3 +776,44 @@ def test_gru_randomized_shape_sweep():
                          (b1.grad, b2.grad)]:
                 scale = b.abs().max().clamp(min=1.0)
                 assert ((a - b).abs().max() / scale) < 2e-4, ctx
+
+
+@pytest.mark.parametrize("dtype", ["fp32", "bf16"])
+@pytest.mark.parametrize("n_dir", [1, 2])
+def test_gru_single_timestep_boundary(dtype, n_dir):
+    """T=1: every boundary special-case (prologue==epilogue, dGh slot
+    always out of range, h0 staging at the only step) collapses into one
+    iteration — forward and full backward vs the golden recurrence."""
+    from fmda_amd.ops.interface import gru_directions
+    torch.manual_seed(31)
+    H, B, T = 128, 33, 1
+    gi = (torch.randn(B, T, n_dir * 3 * H) * 0.5).cuda()
+    w = (torch.randn(n_dir, 3 * H, H) * 0.2).cuda()
+    bhh = (torch.randn(n_dir, 3 * H) * 0.1).cuda()
+    h0 = (torch.randn(n_dir, B, H) * 0.5).cuda().contiguous()
+    cast = ((lambda t: t.bfloat16()) if dtype == "bf16"
+            else (lambda t: t.clone()))
+
+    gi1 = cast(gi).requires_grad_(True)
+    w1 = cast(w).requires_grad_(True)
+    b1 = bhh.clone().requires_grad_(True)
+    h01 = h0.clone().requires_grad_(True)
+    out1, hl1 = gru_directions(gi1, w1, b1, h01)
+    dO = torch.randn(B, T, n_dir * H).cuda()
+    (out1.float() * dO).sum().backward()
+
+    gi2 = gi.clone().requires_grad_(True)
+    w2 = w.clone().requires_grad_(True)
+    b2 = bhh.clone().requires_grad_(True)
+    h02 = h0.clone().requires_grad_(True)
+    out2, hl2 = _gru_ref_from_gi_autograd(gi2, w2, b2, h02)
+    (out2 * dO).sum().backward()
+
+    tol = 5e-2 if dtype == "bf16" else 2e-4
+    assert (out1.float() - out2).abs().max() < tol
+    for nm, a, b in [("dgi", gi1.grad.float(), gi2.grad),
+                     ("dw", w1.grad.float(), w2.grad),
+                     ("dbhh", b1.grad, b2.grad),
+                     ("dh0", h01.grad, h02.grad)]:
+        scale = b.abs().max().clamp(min=1e-2)
+        assert ((a - b).abs().max() / scale) < tol, (nm, dtype, n_dir)
