@@ -120,8 +120,9 @@ class BiGRU(nn.Module):
                             self.hidden_size)
         last_hidden = hidden_v[-1].sum(dim=0)
 
-        if input_seq.is_cuda:
-            # fused direction-sum + max/avg pooling HIP kernel
+        if input_seq.is_cuda and self.hidden_size % 2 == 0:
+            # fused direction-sum + max/avg pooling HIP kernel (pairs
+            # columns; odd H takes the eager path below)
             from ..ops.interface import dirsum_pool
             max_pool, avg_pool = dirsum_pool(gru_out, self.n_directions)
         else:

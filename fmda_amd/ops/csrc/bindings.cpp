@@ -313,6 +313,8 @@ std::vector<torch::Tensor> pool_fwd(torch::Tensor out, int64_t n_dir) {
     TORCH_CHECK(bf16 || out.scalar_type() == torch::kFloat32);
     const int B = out.size(0), T = out.size(1);
     const int H = out.size(2) / n_dir;
+    TORCH_CHECK(H % 2 == 0, "pool_fwd pairs columns: H must be even "
+                "(odd H uses the eager path)");
     auto f32 = out.options().dtype(torch::kFloat32);
     auto maxv = torch::empty({B, H}, f32);
     auto avgv = torch::empty({B, H}, f32);

@@ -817,3 +817,27 @@ def test_gru_single_timestep_boundary(dtype, n_dir):
                      ("dh0", h01.grad, h02.grad)]:
         scale = b.abs().max().clamp(min=1e-2)
         assert ((a - b).abs().max() / scale) < tol, (nm, dtype, n_dir)
+
+
+def test_odd_hidden_size_model_matches_cpu():
+    """Odd H (25): the paired pooling kernels cannot apply, so the model
+    must take the eager pooling path on GPU and still match the CPU
+    oracle end to end (fwd + grads)."""
+    from fmda_amd.models import BiGRU
+    torch.manual_seed(44)
+    m = BiGRU(25, 20, 4, n_layers=1, spatial_dropout=False, dropout=0.0)
+    x = torch.randn(9, 11, 20)
+    logits_cpu = m(x)
+    logits_cpu.sum().backward()
+    gref = {n: p.grad.clone() for n, p in m.named_parameters()}
+
+    mg = BiGRU(25, 20, 4, n_layers=1, spatial_dropout=False, dropout=0.0)
+    mg.load_state_dict(m.state_dict())
+    mg = mg.cuda()
+    logits_gpu = mg(x.cuda())
+    logits_gpu.sum().backward()
+    assert (logits_gpu.cpu() - logits_cpu).abs().max() < 1e-3
+    for n, p in mg.named_parameters():
+        rel = (p.grad.cpu() - gref[n]).abs().max() / \
+            gref[n].abs().max().clamp(min=1e-3)
+        assert rel < 1e-3, n
