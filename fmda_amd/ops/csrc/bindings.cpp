@@ -219,6 +219,11 @@ std::vector<torch::Tensor> gru_bwd(torch::Tensor gi, torch::Tensor w,
     TORCH_CHECK(out.is_contiguous() && dout.is_contiguous() &&
                 dhT.is_contiguous(), "fmda gru_bwd: tensors must be contiguous");
     TORCH_CHECK(dhT.scalar_type() == torch::kFloat32, "dhT must be fp32");
+    // fp32 BPTT at Hp=512 needs ~264 KB of LDS staging — architecturally
+    // out of budget (160 KB/CU); bf16 (the training dtype) covers H>256.
+    TORCH_CHECK(is_bf16 || Hp <= 256,
+                "fp32 backward unsupported for Hp > 256 (LDS budget); "
+                "train in bf16 for hidden sizes above 256");
     torch::Tensor dgi, dgh;
     if (getenv("FMDA_ZERO_GRADS")) {   // debug: expose unwritten positions
         dgi = torch::zeros_like(gi);
