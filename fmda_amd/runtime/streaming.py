@@ -77,7 +77,10 @@ class StreamingPredictor:
         # concurrent push_row can satisfy the retry.
         self._lock = threading.Lock()
         self.x_min = x_min.clone()
-        self.x_rng = (x_max - x_min).clone()
+        # MIN==MAX guard mirroring the chunk loader's epsilon fix
+        # (sql_pytorch_dataloader.py:107-113): a degenerate range must not
+        # produce inf/NaN in the normalize (CPU or the ingest kernel)
+        self.x_rng = (x_max - x_min).clamp(min=1e-6)
         self.y_fields: List[str] = list(TARGET_NAMES)
         self._graph = None
         self._use_graph = (self.device.type == "cuda"
