@@ -219,6 +219,8 @@ class StreamingPredictor:
             if self._graph is None:
                 self._capture_graph()
             self._graph.replay()
+            # (an event-query spin measured identical p50 — the tick is
+            # GPU-pipeline-bound, not sync-latency-bound)
             torch.cuda.current_stream().synchronize()
             if self._pinned_out is not None:
                 pl = self._pinned_out.tolist()
