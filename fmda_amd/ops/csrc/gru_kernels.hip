@@ -1327,7 +1327,11 @@ void gru_fwd_b1_kernel(const __hip_bfloat16* __restrict__ gi,
     }
 }
 
-template <int BT, int Hp, int NT, int WPE>
+// PHASES: diagnostic bitmask (1 = recompute GEMM, 2 = gate grads,
+// 4 = carry GEMM). Production uses 7; other masks exist only so
+// scripts/kernel_micro.py can time each phase's contribution to the
+// serial chain by differencing (outputs are garbage for masks != 7).
+template <int BT, int Hp, int NT, int WPE, int PHASES = 7>
 __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
                           amdgpu_waves_per_eu(WPE, WPE)))
 void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
@@ -1576,7 +1580,7 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 #pragma unroll
                 for (int m = 0; m < MT; ++m) acc[i][g][m] = f32x4_t{0.f};
 #pragma unroll
-        for (int i = 0; i < CPW; ++i) {
+        for (int i = 0; PHASES & 1 && i < CPW; ++i) {
             const int arow = lane & 15;
 #pragma unroll
             for (int kk = 0; kk < KK; ++kk) {
@@ -1597,7 +1601,7 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
             }
         }
 #pragma unroll
-        for (int i = 0; i < CPW; ++i) {
+        for (int i = 0; PHASES & 2 && i < CPW; ++i) {
             const int ct = wave + NW * i;
             const int j = ct * 16 + (lane & 15);
             __bf16* grow = (__bf16*)gi_slot(slot);
@@ -1679,7 +1683,7 @@ void gru_bwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
         __builtin_amdgcn_s_barrier();
 
         // ---- phase C: carry GEMM dh += dGh W — registers + LDS only ----
-        {
+        if (PHASES & 4) {
             f32x4_t acc2[CPW][MT];
 #pragma unroll
             for (int i = 0; i < CPW; ++i)
@@ -3279,7 +3283,7 @@ static int launch_fwd_v3_128(const void* gi, const void* w, const float* bhh,
     return 0;
 }
 
-template <int BT, int NT>
+template <int BT, int NT, int PHASES = 7>
 static int launch_bwd_v3_128_t(const void* gi, const void* w, const void* wt,
                                const float* bhh,
                                const void* out, const void* dout,
@@ -3294,7 +3298,7 @@ static int launch_bwd_v3_128_t(const void* gi, const void* w, const void* wt,
     const size_t lds = 3 * 2 * BT * 3 * Hp + 2 * 2 * BT * Hp +
                        2 * 2 * BT * Hp + 2 * 2 * BT * (3 * Hp + 8) +
                        4 * 3 * Hp;
-    auto k = gru_bwd_v3_kernel<BT, Hp, NT, 2>;
+    auto k = gru_bwd_v3_kernel<BT, Hp, NT, 2, PHASES>;
     (void)hipFuncSetAttribute((const void*)k,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     const dim3 grid((B + BT - 1) / BT, n_dir);
@@ -3326,6 +3330,29 @@ static int launch_bwd_v3_128(const void* gi, const void* w, const void* wt,
         return launch_bwd_v3_128_t<16, 256>(
             gi, w, wt, bhh, out, dout, dhT, dgi, dgh, dh0, dbhh, B, Tseq,
             n_dir, drop_thr, drop_scale, drop_seed, h0, dgh0, stream);
+    // diagnostic phase masks (outputs garbage; timing-only — see the
+    // kernel's PHASES comment)
+    static const char* pm = getenv("FMDA_BWD_PHASES");
+    if (pm) {
+        switch (atoi(pm)) {
+            case 6: return launch_bwd_v3_128_t<32, 512, 6>(
+                gi, w, wt, bhh, out, dout, dhT, dgi, dgh, dh0, dbhh, B,
+                Tseq, n_dir, drop_thr, drop_scale, drop_seed, h0, dgh0,
+                stream);
+            case 5: return launch_bwd_v3_128_t<32, 512, 5>(
+                gi, w, wt, bhh, out, dout, dhT, dgi, dgh, dh0, dbhh, B,
+                Tseq, n_dir, drop_thr, drop_scale, drop_seed, h0, dgh0,
+                stream);
+            case 3: return launch_bwd_v3_128_t<32, 512, 3>(
+                gi, w, wt, bhh, out, dout, dhT, dgi, dgh, dh0, dbhh, B,
+                Tseq, n_dir, drop_thr, drop_scale, drop_seed, h0, dgh0,
+                stream);
+            case 0: return launch_bwd_v3_128_t<32, 512, 0>(
+                gi, w, wt, bhh, out, dout, dhT, dgi, dgh, dh0, dbhh, B,
+                Tseq, n_dir, drop_thr, drop_scale, drop_seed, h0, dgh0,
+                stream);
+        }
+    }
     return launch_bwd_v3_128_t<32, 512>(
         gi, w, wt, bhh, out, dout, dhT, dgi, dgh, dh0, dbhh, B, Tseq,
         n_dir, drop_thr, drop_scale, drop_seed, h0, dgh0, stream);
