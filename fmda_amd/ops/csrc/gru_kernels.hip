@@ -888,7 +888,9 @@ FMDA_DEV void glds16(const void* gsrc, void* ldst) {
         (__attribute__((address_space(3))) unsigned int*)ldst, 16, 0, 0);
 }
 
-template <int BT, int Hp, int NT, int WPE>
+// PHASES diagnostic mask like the backward's: 1 = recurrent GEMM,
+// 2 = gates. Production uses 3; other masks are timing-only.
+template <int BT, int Hp, int NT, int WPE, int PHASES = 3>
 __global__ __attribute__((amdgpu_flat_work_group_size(NT, NT),
                           amdgpu_waves_per_eu(WPE, WPE)))
 void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
@@ -1058,7 +1060,7 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 #pragma unroll
                 for (int m = 0; m < MT; ++m) acc[i][g][m] = f32x4_t{0.f};
 #pragma unroll
-        for (int i = 0; i < CPW; ++i) {
+        for (int i = 0; PHASES & 1 && i < CPW; ++i) {
             const int arow = lane & 15;
             const int koff = 8 * (lane >> 4);
 #pragma unroll
@@ -1086,7 +1088,7 @@ void gru_fwd_v3_kernel(const __hip_bfloat16* __restrict__ gi,
 
         // ---- phase B: fused gates; h stays in registers ----
 #pragma unroll
-        for (int i = 0; i < CPW; ++i) {
+        for (int i = 0; PHASES & 2 && i < CPW; ++i) {
             const int ct = wave + NW * i;
             const int j = ct * 16 + (lane & 15);
             const __bf16* gprow = (const __bf16*)gi_buf(pb);
@@ -3271,6 +3273,31 @@ static int launch_fwd_v3_128(const void* gi, const void* w, const float* bhh,
     constexpr int BT = 16, Hp = 128, NT = 256;
     const size_t lds = 2 * 2 * BT * 3 * Hp + 2 * 2 * BT * (Hp + 8) +
                        4 * 3 * Hp;
+    static const char* pm = getenv("FMDA_FWD_PHASES");
+    if (pm && atoi(pm) == 0) {   // timing-only skeleton (garbage output)
+        auto k0 = gru_fwd_v3_kernel<BT, Hp, NT, 2, 0>;
+        (void)hipFuncSetAttribute((const void*)k0,
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+        const dim3 grid((B + BT - 1) / BT, n_dir);
+        k0<<<grid, NT, lds, stream>>>((const __hip_bfloat16*)gi,
+                                      (const __hip_bfloat16*)w, bhh,
+                                      (__hip_bfloat16*)out, hlast, B, Tseq,
+                                      n_dir, h0, (__hip_bfloat16*)out_drop,
+                                      drop_thr, drop_scale, drop_seed);
+        return 0;
+    }
+    if (pm && atoi(pm) == 1) {   // GEMM only, no gates
+        auto k1 = gru_fwd_v3_kernel<BT, Hp, NT, 2, 1>;
+        (void)hipFuncSetAttribute((const void*)k1,
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+        const dim3 grid((B + BT - 1) / BT, n_dir);
+        k1<<<grid, NT, lds, stream>>>((const __hip_bfloat16*)gi,
+                                      (const __hip_bfloat16*)w, bhh,
+                                      (__hip_bfloat16*)out, hlast, B, Tseq,
+                                      n_dir, h0, (__hip_bfloat16*)out_drop,
+                                      drop_thr, drop_scale, drop_seed);
+        return 0;
+    }
     auto k = gru_fwd_v3_kernel<BT, Hp, NT, 2>;
     (void)hipFuncSetAttribute((const void*)k,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
