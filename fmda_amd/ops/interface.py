@@ -11,6 +11,7 @@ compose naturally.
 Replaces the reference's `self.gru(input_seq)` cuDNN RNN call site
 (biGRU_model.py:102).
 """
+import os
 from typing import List, Optional, Tuple
 
 import torch
@@ -461,11 +462,10 @@ def bigru_stack(x: torch.Tensor, gru_module: torch.nn.GRU, n_layers: int,
         # fully fused inter-layer dropout: v3 (Hp=128) and column-split
         # (Hp=512, zero-h0) kernels emit the dropped copy in forward and
         # recompute the mask at the backward d_out read
-        import os as _os
         defer = (drop_here and Hp == H and Hp in (128, 512)
                  and x.dtype == torch.bfloat16 and x.is_cuda
                  and (Hp == 128 or (h0_layers[layer] is None
-                      and _os.environ.get("FMDA_CS_DROP", "1") != "0")))
+                      and os.environ.get("FMDA_CS_DROP", "1") != "0")))
         seed = (int(torch.empty((), dtype=torch.int64).random_())
                 if defer else 0)
         dp = (dropout_p, seed) if defer else (0.0, 0)
